@@ -1,0 +1,550 @@
+// ============================================================================
+// ethrex_mi355 C-ABI implementation — MI355X-native BN254 MSM/NTT core.
+//
+// See include/ethrex_mi355.h for the boundary contract (mirrors the in-repo
+// ZisK accelerator FFI convention, crates/guest-program/src/crypto/zisk.rs:71-137)
+// and DESIGN.md for the kernel design.  Threading: the backend is called
+// from ONE actor on a blocking thread (crates/prover/src/prover.rs:241-251),
+// so plans are not internally locked.
+//
+// NO CPU FALLBACK: every compute entry point requires a visible GPU and
+// returns EM_ERR_HIP otherwise.
+// ============================================================================
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+#include <cstring>
+#include <string>
+
+#include <rocprim/rocprim.hpp>
+
+#include "../../include/ethrex_mi355.h"
+#include "gpu_field.h"
+#include "gpu_g1.h"
+#include "msm_kernels.h"
+#include "ntt_kernels.h"
+
+using namespace em;
+
+static thread_local std::string g_last_err;
+
+static int hip_fail(hipError_t e, const char *where) {
+    g_last_err = std::string(where) + ": " + hipGetErrorString(e);
+    return EM_ERR_HIP;
+}
+
+#define HIP_TRY(call)                                    \
+    do {                                                 \
+        hipError_t _e = (call);                          \
+        if (_e != hipSuccess) return hip_fail(_e, #call); \
+    } while (0)
+
+extern "C" const char *ethrex_mi355_version(void) { return "0.1.0-gfx950"; }
+
+extern "C" const char *ethrex_mi355_last_error(void) { return g_last_err.c_str(); }
+
+extern "C" int ethrex_mi355_device_count(int *count) {
+    if (!count) return EM_ERR_INPUT;
+    hipError_t e = hipGetDeviceCount(count);
+    if (e != hipSuccess) {
+        *count = 0;
+        return hip_fail(e, "hipGetDeviceCount");
+    }
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_set_device(int device) {
+    HIP_TRY(hipSetDevice(device));
+    return EM_OK;
+}
+
+static int require_gpu() {
+    int n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    if (e != hipSuccess || n == 0) {
+        g_last_err = "no HIP device visible (MI355X required; no CPU fallback)";
+        return EM_ERR_HIP;
+    }
+    return EM_OK;
+}
+
+static inline uint32_t blocks_for(size_t n, int bs) {
+    return (uint32_t)((n + bs - 1) / bs);
+}
+
+// ============================ MSM plan ============================
+
+struct em_msm_plan {
+    size_t n;
+    g1a *d_pts = nullptr;
+    uint8_t *d_inf = nullptr;
+    fe4 *d_scalars = nullptr;
+    uint8_t *d_scratch = nullptr;     // 64n bytes: point/scalar byte staging
+    uint32_t *d_keys = nullptr;       // 16n
+    uint32_t *d_vals = nullptr;
+    uint32_t *d_keys_out = nullptr;
+    uint32_t *d_vals_out = nullptr;
+    void *d_sort_tmp = nullptr;
+    size_t sort_tmp_bytes = 0;
+    uint32_t *d_offsets = nullptr;    // NBUCKET_TOTAL + 1
+    g1j *d_buckets = nullptr;         // NBUCKET_TOTAL
+    g1j *d_seg_sum = nullptr;         // NWIN*NSEG
+    g1j *d_seg_wsum = nullptr;
+    g1j *d_windows = nullptr;         // NWIN
+    uint8_t *d_out = nullptr;         // 96 B
+    uint32_t *d_err = nullptr;
+    bool have_scalars = false;
+    bool have_points = false;
+    hipEvent_t ev[6];
+    double last_ms[5] = {0, 0, 0, 0, 0};
+};
+
+extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
+    if (!plan || n == 0) return EM_ERR_INPUT;
+    int rc = require_gpu();
+    if (rc) return rc;
+    em_msm_plan *p = new em_msm_plan();
+    p->n = n;
+    size_t total = n * MSM_NWIN;
+    hipError_t e = hipSuccess;
+    auto mal = [&](void **ptr, size_t bytes) {
+        if (e == hipSuccess) e = hipMalloc(ptr, bytes);
+    };
+    mal((void **)&p->d_pts, n * sizeof(g1a));
+    mal((void **)&p->d_inf, n);
+    mal((void **)&p->d_scalars, n * sizeof(fe4));
+    mal((void **)&p->d_scratch, n * 64);
+    mal((void **)&p->d_keys, total * 4);
+    mal((void **)&p->d_vals, total * 4);
+    mal((void **)&p->d_keys_out, total * 4);
+    mal((void **)&p->d_vals_out, total * 4);
+    mal((void **)&p->d_offsets, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
+    mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1j));
+    mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1j));
+    mal((void **)&p->d_seg_wsum, MSM_NWIN * MSM_NSEG * sizeof(g1j));
+    mal((void **)&p->d_windows, MSM_NWIN * sizeof(g1j));
+    mal((void **)&p->d_out, 96);
+    mal((void **)&p->d_err, 4);
+    if (e == hipSuccess) {
+        // rocPRIM temp-storage size query
+        e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
+                                      p->d_keys_out, p->d_vals, p->d_vals_out,
+                                      total, 0, 20);
+        if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
+    }
+    for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
+    if (e != hipSuccess) {
+        ethrex_mi355_msm_plan_destroy(p);
+        return hip_fail(e, "msm_plan_create");
+    }
+    *plan = p;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
+    if (!p) return EM_ERR_INPUT;
+    hipFree(p->d_pts);
+    hipFree(p->d_inf);
+    hipFree(p->d_scalars);
+    hipFree(p->d_scratch);
+    hipFree(p->d_keys);
+    hipFree(p->d_vals);
+    hipFree(p->d_keys_out);
+    hipFree(p->d_vals_out);
+    hipFree(p->d_sort_tmp);
+    hipFree(p->d_offsets);
+    hipFree(p->d_buckets);
+    hipFree(p->d_seg_sum);
+    hipFree(p->d_seg_wsum);
+    hipFree(p->d_windows);
+    hipFree(p->d_out);
+    hipFree(p->d_err);
+    delete p;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_msm_upload_points(em_msm_plan *p,
+                                              const uint8_t *points64) {
+    if (!p || !points64) return EM_ERR_INPUT;
+    HIP_TRY(hipMemset(p->d_err, 0, 4));
+    HIP_TRY(hipMemcpy(p->d_scratch, points64, p->n * 64, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_parse_points, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_scratch, p->d_pts, p->d_inf, p->n, p->d_err);
+    uint32_t err = 0;
+    HIP_TRY(hipMemcpy(&err, p->d_err, 4, hipMemcpyDeviceToHost));
+    if (err) return EM_ERR_POINT;
+    p->have_points = true;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_msm_gen_points(em_msm_plan *p, uint64_t start) {
+    if (!p) return EM_ERR_INPUT;
+    hipLaunchKernelGGL(k_gen_points, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_pts, p->d_inf, p->n, start);
+    HIP_TRY(hipDeviceSynchronize());
+    p->have_points = true;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_msm_download_points(em_msm_plan *p, uint8_t *out64) {
+    if (!p || !out64 || !p->have_points) return EM_ERR_INPUT;
+    hipLaunchKernelGGL(k_points_to_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_pts, p->d_inf, p->d_scratch, p->n);
+    HIP_TRY(hipMemcpy(out64, p->d_scratch, p->n * 64, hipMemcpyDeviceToHost));
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_msm_upload_scalars(em_msm_plan *p,
+                                               const uint8_t *scalars32) {
+    if (!p || !scalars32) return EM_ERR_INPUT;
+    HIP_TRY(hipMemcpy(p->d_scratch, scalars32, p->n * 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_parse_scalars, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_scratch, p->d_scalars, p->n);
+    HIP_TRY(hipDeviceSynchronize());
+    p->have_scalars = true;
+    return EM_OK;
+}
+
+static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
+    if (!p || !out) return EM_ERR_INPUT;
+    if (!p->have_points || !p->have_scalars) {
+        g_last_err = "msm_run: points/scalars not uploaded";
+        return EM_ERR_INPUT;
+    }
+    size_t total = p->n * MSM_NWIN;
+    HIP_TRY(hipEventRecord(p->ev[0], 0));
+    // digits
+    hipLaunchKernelGGL(k_digits, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_scalars, p->d_inf, p->d_keys, p->d_vals, p->n);
+    // sort on 20 key bits
+    size_t tmp = p->sort_tmp_bytes;
+    hipError_t e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
+                                             p->d_keys_out, p->d_vals,
+                                             p->d_vals_out, total, 0, 20);
+    if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs");
+    // offsets
+    hipLaunchKernelGGL(k_offsets,
+                       dim3(blocks_for((size_t)MSM_NBUCKET_TOTAL + 1, 256)),
+                       dim3(256), 0, 0, p->d_keys_out, total, p->d_offsets);
+    HIP_TRY(hipEventRecord(p->ev[1], 0));
+    // bucket accumulation (hot)
+    hipLaunchKernelGGL(k_bucket_acc, dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)),
+                       dim3(256), 0, 0, p->d_pts, p->d_vals_out, p->d_offsets,
+                       p->d_buckets);
+    HIP_TRY(hipEventRecord(p->ev[2], 0));
+    // reductions
+    hipLaunchKernelGGL(k_segment_reduce,
+                       dim3(blocks_for(MSM_NWIN * MSM_NSEG, 256)), dim3(256), 0, 0,
+                       p->d_buckets, p->d_seg_sum, p->d_seg_wsum);
+    hipLaunchKernelGGL(k_window_reduce, dim3(1), dim3(64), 0, 0, p->d_seg_sum,
+                       p->d_seg_wsum, p->d_windows);
+    HIP_TRY(hipEventRecord(p->ev[3], 0));
+    hipLaunchKernelGGL(k_final_combine, dim3(1), dim3(64), 0, 0, p->d_windows,
+                       p->d_out, out_mode);
+    HIP_TRY(hipEventRecord(p->ev[4], 0));
+    HIP_TRY(hipMemcpy(out, p->d_out, out_mode == 0 ? 64 : 96,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipDeviceSynchronize());
+    float ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[1]));
+    p->last_ms[0] = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[1], p->ev[2]));
+    p->last_ms[1] = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[2], p->ev[3]));
+    p->last_ms[2] = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[3], p->ev[4]));
+    p->last_ms[3] = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[4]));
+    p->last_ms[4] = ms;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_msm_run(em_msm_plan *p, uint8_t out[64]) {
+    return msm_run_inner(p, out, 0);
+}
+
+extern "C" int ethrex_mi355_msm_run_partial(em_msm_plan *p, uint8_t out[96]) {
+    return msm_run_inner(p, out, 1);
+}
+
+extern "C" int ethrex_mi355_msm_last_times(em_msm_plan *p, double times_ms[5]) {
+    if (!p || !times_ms) return EM_ERR_INPUT;
+    memcpy(times_ms, p->last_ms, sizeof p->last_ms);
+    return EM_OK;
+}
+
+// ============================ one-shot MSM ============================
+
+extern "C" int ethrex_mi355_bn254_g1_msm(const uint8_t *points64,
+                                         const uint8_t *scalars32, size_t n,
+                                         uint8_t out[64]) {
+    if (!points64 || !scalars32 || !out || n == 0) return EM_ERR_INPUT;
+    em_msm_plan *p = nullptr;
+    int rc = ethrex_mi355_msm_plan_create(n, &p);
+    if (rc) return rc;
+    rc = ethrex_mi355_msm_upload_points(p, points64);
+    if (!rc) rc = ethrex_mi355_msm_upload_scalars(p, scalars32);
+    if (!rc) rc = ethrex_mi355_msm_run(p, out);
+    ethrex_mi355_msm_plan_destroy(p);
+    return rc;
+}
+
+// ============================ single ops ============================
+
+extern "C" int ethrex_mi355_bn254_g1_add(const uint8_t p1[64],
+                                         const uint8_t p2[64], uint8_t out[64]) {
+    if (!p1 || !p2 || !out) return EM_ERR_INPUT;
+    int rc = require_gpu();
+    if (rc) return rc;
+    uint8_t *d_in, *d_out;
+    uint32_t *d_err;
+    HIP_TRY(hipMalloc(&d_in, 128));
+    HIP_TRY(hipMalloc(&d_out, 64));
+    HIP_TRY(hipMalloc(&d_err, 4));
+    HIP_TRY(hipMemset(d_err, 0, 4));
+    HIP_TRY(hipMemcpy(d_in, p1, 64, hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_in + 64, p2, 64, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_g1_add_single, dim3(1), dim3(64), 0, 0, d_in, d_out, d_err);
+    uint32_t err;
+    HIP_TRY(hipMemcpy(&err, d_err, 4, hipMemcpyDeviceToHost));
+    if (!err) HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
+    hipFree(d_in);
+    hipFree(d_out);
+    hipFree(d_err);
+    return err ? EM_ERR_POINT : EM_OK;
+}
+
+extern "C" int ethrex_mi355_bn254_g1_mul(const uint8_t point[64],
+                                         const uint8_t scalar[32],
+                                         uint8_t out[64]) {
+    if (!point || !scalar || !out) return EM_ERR_INPUT;
+    int rc = require_gpu();
+    if (rc) return rc;
+    uint8_t *d_in, *d_out;
+    uint32_t *d_err;
+    HIP_TRY(hipMalloc(&d_in, 96));
+    HIP_TRY(hipMalloc(&d_out, 64));
+    HIP_TRY(hipMalloc(&d_err, 4));
+    HIP_TRY(hipMemset(d_err, 0, 4));
+    HIP_TRY(hipMemcpy(d_in, point, 64, hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_in + 64, scalar, 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_g1_mul_single, dim3(1), dim3(64), 0, 0, d_in, d_out, d_err);
+    uint32_t err;
+    HIP_TRY(hipMemcpy(&err, d_err, 4, hipMemcpyDeviceToHost));
+    if (!err) HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
+    hipFree(d_in);
+    hipFree(d_out);
+    hipFree(d_err);
+    return err ? EM_ERR_POINT : EM_OK;
+}
+
+extern "C" int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96,
+                                             size_t count, uint8_t out[64]) {
+    if (!jacobians96 || !out || count == 0) return EM_ERR_INPUT;
+    int rc = require_gpu();
+    if (rc) return rc;
+    uint8_t *d_in, *d_out;
+    HIP_TRY(hipMalloc(&d_in, 96 * count));
+    HIP_TRY(hipMalloc(&d_out, 64));
+    HIP_TRY(hipMemcpy(d_in, jacobians96, 96 * count, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_g1_combine, dim3(1), dim3(64), 0, 0, d_in, count, d_out);
+    HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
+    hipFree(d_in);
+    hipFree(d_out);
+    return EM_OK;
+}
+
+// ============================ NTT plan ============================
+
+struct em_ntt_plan {
+    size_t n;
+    int logn;
+    fe4 *d_data = nullptr;
+    uint8_t *d_bytes = nullptr;
+    fe4 *d_tw = nullptr;      // forward twiddles, n/2
+    fe4 *d_tw_inv = nullptr;  // inverse twiddles, n/2
+    uint32_t *d_err = nullptr;
+    hipEvent_t ev[4];
+    double last_ms[3] = {0, 0, 0};
+};
+
+extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
+    if (!plan || n == 0 || (n & (n - 1))) return EM_ERR_INPUT;
+    int logn = 0;
+    while (((size_t)1 << logn) < n) logn++;
+    if (logn > bn254::FR_TWO_ADICITY) return EM_ERR_INPUT;
+    int rc = require_gpu();
+    if (rc) return rc;
+    em_ntt_plan *p = new em_ntt_plan();
+    p->n = n;
+    p->logn = logn;
+    size_t half = n > 1 ? n / 2 : 1;
+    hipError_t e = hipSuccess;
+    auto mal = [&](void **ptr, size_t bytes) {
+        if (e == hipSuccess) e = hipMalloc(ptr, bytes);
+    };
+    mal((void **)&p->d_data, n * sizeof(fe4));
+    mal((void **)&p->d_bytes, n * 32);
+    mal((void **)&p->d_tw, half * sizeof(fe4));
+    mal((void **)&p->d_tw_inv, half * sizeof(fe4));
+    mal((void **)&p->d_err, 4);
+    for (int i = 0; i < 4 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
+    if (e != hipSuccess) {
+        ethrex_mi355_ntt_plan_destroy(p);
+        return hip_fail(e, "ntt_plan_create");
+    }
+    // host-side twiddle seed powers: w = W28^(2^(28-logn)); w2k[k] = w^(2^k)
+    for (int dir = 0; dir < 2; dir++) {
+        fe4 w = dir == 0
+                    ? fe4{{bn254::FR_W28_MONT[0], bn254::FR_W28_MONT[1],
+                           bn254::FR_W28_MONT[2], bn254::FR_W28_MONT[3]}}
+                    : fe4{{bn254::FR_W28_INV_MONT[0], bn254::FR_W28_INV_MONT[1],
+                           bn254::FR_W28_INV_MONT[2], bn254::FR_W28_INV_MONT[3]}};
+        for (int k = bn254::FR_TWO_ADICITY; k > logn; k--) w = mont_sqr<Fr>(w);
+        fe4 w2k[32];
+        w2k[0] = w;
+        for (int k = 1; k < logn; k++) w2k[k] = mont_sqr<Fr>(w2k[k - 1]);
+        if (logn == 0) w2k[0] = fe_one_mont<Fr>();
+        fe4 *d_w2k;
+        HIP_TRY(hipMalloc(&d_w2k, sizeof(w2k)));
+        HIP_TRY(hipMemcpy(d_w2k, w2k, sizeof(w2k), hipMemcpyHostToDevice));
+        hipLaunchKernelGGL(k_gen_twiddles, dim3(blocks_for(half, 256)), dim3(256),
+                           0, 0, dir == 0 ? p->d_tw : p->d_tw_inv, half, d_w2k,
+                           logn);
+        HIP_TRY(hipDeviceSynchronize());
+        hipFree(d_w2k);
+    }
+    *plan = p;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_plan_destroy(em_ntt_plan *p) {
+    if (!p) return EM_ERR_INPUT;
+    hipFree(p->d_data);
+    hipFree(p->d_bytes);
+    hipFree(p->d_tw);
+    hipFree(p->d_tw_inv);
+    hipFree(p->d_err);
+    delete p;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_upload(em_ntt_plan *p, const uint8_t *elems32) {
+    if (!p || !elems32) return EM_ERR_INPUT;
+    HIP_TRY(hipMemset(p->d_err, 0, 4));
+    HIP_TRY(hipMemcpy(p->d_bytes, elems32, p->n * 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_fr_from_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_bytes, p->d_data, p->n, p->d_err);
+    uint32_t err;
+    HIP_TRY(hipMemcpy(&err, p->d_err, 4, hipMemcpyDeviceToHost));
+    return err ? EM_ERR_INPUT : EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
+    if (!p) return EM_ERR_INPUT;
+    size_t n = p->n;
+    HIP_TRY(hipEventRecord(p->ev[0], 0));
+    if (n > 1) {
+        hipLaunchKernelGGL(k_bit_reverse, dim3(blocks_for(n, 256)), dim3(256), 0, 0,
+                           p->d_data, n, p->logn);
+    }
+    HIP_TRY(hipEventRecord(p->ev[1], 0));
+    const fe4 *tw = inverse ? p->d_tw_inv : p->d_tw;
+    for (int s = 1; s <= p->logn; s++) {
+        hipLaunchKernelGGL(k_ntt_stage, dim3(blocks_for(n / 2, 256)), dim3(256), 0,
+                           0, p->d_data, tw, n, p->logn, s);
+    }
+    if (inverse) {
+        hipLaunchKernelGGL(k_ntt_scale, dim3(blocks_for(n, 256)), dim3(256), 0, 0,
+                           p->d_data, n, p->logn);
+    }
+    HIP_TRY(hipEventRecord(p->ev[2], 0));
+    HIP_TRY(hipDeviceSynchronize());
+    float ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[1]));
+    p->last_ms[0] = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[1], p->ev[2]));
+    p->last_ms[1] = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[2]));
+    p->last_ms[2] = ms;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_download(em_ntt_plan *p, uint8_t *elems32) {
+    if (!p || !elems32) return EM_ERR_INPUT;
+    hipLaunchKernelGGL(k_fr_to_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_data, p->d_bytes, p->n);
+    HIP_TRY(hipMemcpy(elems32, p->d_bytes, p->n * 32, hipMemcpyDeviceToHost));
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_last_times(em_ntt_plan *p, double times_ms[3]) {
+    if (!p || !times_ms) return EM_ERR_INPUT;
+    memcpy(times_ms, p->last_ms, sizeof p->last_ms);
+    return EM_OK;
+}
+
+// ============================ one-shot NTT ============================
+
+extern "C" int ethrex_mi355_bn254_fr_ntt(uint8_t *elems32, size_t n, int inverse) {
+    if (!elems32 || n == 0) return EM_ERR_INPUT;
+    em_ntt_plan *p = nullptr;
+    int rc = ethrex_mi355_ntt_plan_create(n, &p);
+    if (rc) return rc;
+    rc = ethrex_mi355_ntt_upload(p, elems32);
+    if (!rc) rc = ethrex_mi355_ntt_run(p, inverse);
+    if (!rc) rc = ethrex_mi355_ntt_download(p, elems32);
+    ethrex_mi355_ntt_plan_destroy(p);
+    return rc;
+}
+
+// ============================ input generation ============================
+// Product-side restatement of BASELINE.md's deterministic input scheme
+// (splitmix64-seeded xoshiro256++, rejection to [0, r), 254-bit mask).
+// Parity-tested against the oracle's independent restatement.
+
+namespace {
+
+struct Xosh {
+    uint64_t s[4];
+};
+
+uint64_t splitmix64_next(uint64_t &x) {
+    uint64_t z = (x += 0x9e3779b97f4a7c15ull);
+    z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+    z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+    return z ^ (z >> 31);
+}
+
+uint64_t rotl64(uint64_t x, int k) { return (x << k) | (x >> (64 - k)); }
+
+uint64_t xosh_next(Xosh &g) {
+    uint64_t r = rotl64(g.s[0] + g.s[3], 23) + g.s[0];
+    uint64_t t = g.s[1] << 17;
+    g.s[2] ^= g.s[0];
+    g.s[3] ^= g.s[1];
+    g.s[1] ^= g.s[2];
+    g.s[0] ^= g.s[3];
+    g.s[2] ^= t;
+    g.s[3] = rotl64(g.s[3], 45);
+    return r;
+}
+
+}  // namespace
+
+extern "C" void ethrex_mi355_gen_fr(uint64_t seed, size_t n, uint8_t *out) {
+    Xosh g;
+    uint64_t sm = seed;
+    for (int i = 0; i < 4; i++) g.s[i] = splitmix64_next(sm);
+    const fe4 rmod{{Fr::MOD[0], Fr::MOD[1], Fr::MOD[2], Fr::MOD[3]}};
+    for (size_t i = 0; i < n; i++) {
+        fe4 s;
+        do {
+            s.v[0] = xosh_next(g);
+            s.v[1] = xosh_next(g);
+            s.v[2] = xosh_next(g);
+            s.v[3] = xosh_next(g) & 0x3fffffffffffffffull;
+        } while (fe_geq(s, rmod));
+        fe_to_be(out + 32 * i, s);
+    }
+}

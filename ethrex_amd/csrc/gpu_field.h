@@ -1,0 +1,231 @@
+// ============================================================================
+// Device 4x64-limb Montgomery field arithmetic for BN254 Fq / Fr — gfx950.
+//
+// Written for CDNA4: big-integer modular mul is VALU work (no MFMA — it is
+// not a dense fp contraction).  64x64->128 products lower to v_mad_u64_u32
+// chains on amdgcn; everything is fully unrolled so limbs live in VGPRs and
+// the compiler can schedule the carry chains.
+//
+// Semantics restated from the reference's ark-bn254 path
+// (crates/common/crypto/provider.rs:247-318): 4x64 little-endian limbs,
+// Montgomery R = 2^256, inputs to/from big-endian canonical bytes.
+// Implemented independently from oracle/bn254_oracle.c (CIOS here, SOS
+// there) so the parity comparison does not share bugs.
+// ============================================================================
+#pragma once
+#include <hip/hip_runtime.h>
+#include "bn254_constants_dev.h"
+
+namespace em {
+
+using u64 = uint64_t;
+using u128 = unsigned __int128;
+
+struct fe4 {
+    u64 v[4];
+};
+
+__device__ __host__ __forceinline__ bool fe_is_zero(const fe4 &a) {
+    return (a.v[0] | a.v[1] | a.v[2] | a.v[3]) == 0;
+}
+
+__device__ __host__ __forceinline__ bool fe_eq(const fe4 &a, const fe4 &b) {
+    return ((a.v[0] ^ b.v[0]) | (a.v[1] ^ b.v[1]) | (a.v[2] ^ b.v[2]) |
+            (a.v[3] ^ b.v[3])) == 0;
+}
+
+// a >= b ?
+__device__ __host__ __forceinline__ bool fe_geq(const fe4 &a, const fe4 &b) {
+#pragma unroll
+    for (int i = 3; i >= 0; i--) {
+        if (a.v[i] != b.v[i]) return a.v[i] > b.v[i];
+    }
+    return true;
+}
+
+template <typename F>
+__device__ __host__ __forceinline__ fe4 fe_load_const(const u64 (&c)[4]) {
+    return fe4{{c[0], c[1], c[2], c[3]}};
+}
+
+// ---- modular add/sub (F = bn254::Fq or bn254::Fr) ----
+
+template <typename F>
+__device__ __host__ __forceinline__ fe4 mod_add(const fe4 &a, const fe4 &b) {
+    u64 r[4];
+    u128 c = 0;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+        c += (u128)a.v[i] + b.v[i];
+        r[i] = (u64)c;
+        c >>= 64;
+    }
+    // subtract MOD if carry or r >= MOD
+    fe4 out{{r[0], r[1], r[2], r[3]}};
+    bool ge = (bool)c ||
+              fe_geq(out, fe4{{F::MOD[0], F::MOD[1], F::MOD[2], F::MOD[3]}});
+    if (ge) {
+        u128 bor = 0;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+            u128 t = (u128)out.v[i] - F::MOD[i] - bor;
+            out.v[i] = (u64)t;
+            bor = (t >> 64) & 1;
+        }
+    }
+    return out;
+}
+
+template <typename F>
+__device__ __host__ __forceinline__ fe4 mod_sub(const fe4 &a, const fe4 &b) {
+    u64 r[4];
+    u128 bor = 0;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+        u128 t = (u128)a.v[i] - b.v[i] - bor;
+        r[i] = (u64)t;
+        bor = (t >> 64) & 1;
+    }
+    fe4 out{{r[0], r[1], r[2], r[3]}};
+    if (bor) {
+        u128 c = 0;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+            c += (u128)out.v[i] + F::MOD[i];
+            out.v[i] = (u64)c;
+            c >>= 64;
+        }
+    }
+    return out;
+}
+
+template <typename F>
+__device__ __host__ __forceinline__ fe4 mod_dbl(const fe4 &a) {
+    return mod_add<F>(a, a);
+}
+
+// ---- CIOS Montgomery multiplication ----
+// t has 6 limbs (4 + 2 guard); after each a_i row we fold in m_i * MOD.
+// Inputs < MOD  =>  output < MOD.
+
+template <typename F>
+__device__ __host__ __forceinline__ fe4 mont_mul(const fe4 &a, const fe4 &b) {
+    u64 t0 = 0, t1 = 0, t2 = 0, t3 = 0, t4 = 0, t5 = 0;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+        // t += a[i] * b
+        u128 c = (u128)a.v[i] * b.v[0] + t0;
+        t0 = (u64)c;
+        c = (c >> 64) + (u128)a.v[i] * b.v[1] + t1;
+        t1 = (u64)c;
+        c = (c >> 64) + (u128)a.v[i] * b.v[2] + t2;
+        t2 = (u64)c;
+        c = (c >> 64) + (u128)a.v[i] * b.v[3] + t3;
+        t3 = (u64)c;
+        c = (c >> 64) + t4;
+        t4 = (u64)c;
+        t5 = (u64)(c >> 64);
+
+        // m = t0 * n0inv;  t += m * MOD;  t >>= 64
+        u64 m = t0 * F::N0INV;
+        c = (u128)m * F::MOD[0] + t0;
+        c >>= 64;
+        c += (u128)m * F::MOD[1] + t1;
+        t0 = (u64)c;
+        c = (c >> 64) + (u128)m * F::MOD[2] + t2;
+        t1 = (u64)c;
+        c = (c >> 64) + (u128)m * F::MOD[3] + t3;
+        t2 = (u64)c;
+        c = (c >> 64) + t4;
+        t3 = (u64)c;
+        t4 = t5 + (u64)(c >> 64);
+        t5 = 0;
+    }
+    fe4 out{{t0, t1, t2, t3}};
+    // final conditional subtraction (t4 is 0 or 1)
+    bool ge = t4 != 0;
+    if (!ge) ge = fe_geq(out, fe4{{F::MOD[0], F::MOD[1], F::MOD[2], F::MOD[3]}});
+    if (ge) {
+        u128 bor = 0;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+            u128 t = (u128)out.v[i] - F::MOD[i] - bor;
+            out.v[i] = (u64)t;
+            bor = (t >> 64) & 1;
+        }
+    }
+    return out;
+}
+
+template <typename F>
+__device__ __host__ __forceinline__ fe4 mont_sqr(const fe4 &a) {
+    return mont_mul<F>(a, a);
+}
+
+// to Montgomery form; input may be ANY 256-bit value — CIOS with b = R2 < MOD
+// and a < 2^256 keeps the accumulator within the guard limbs and fully
+// reduces, mirroring ark's from_be_bytes_mod_order for 32-byte inputs.
+template <typename F>
+__device__ __host__ __forceinline__ fe4 to_mont(const fe4 &a) {
+    return mont_mul<F>(a, fe4{{F::R2[0], F::R2[1], F::R2[2], F::R2[3]}});
+}
+
+template <typename F>
+__device__ __host__ __forceinline__ fe4 from_mont(const fe4 &a) {
+    return mont_mul<F>(a, fe4{{1, 0, 0, 0}});
+}
+
+template <typename F>
+__device__ __host__ __forceinline__ fe4 fe_one_mont() {
+    return fe4{{F::R[0], F::R[1], F::R[2], F::R[3]}};
+}
+
+// x^e (Montgomery in/out), e canonical 4x64
+template <typename F>
+__device__ __forceinline__ fe4 mont_pow(const fe4 &x, const fe4 &e) {
+    fe4 acc = fe_one_mont<F>();
+    fe4 base = x;
+    for (int i = 255; i >= 0; i--) {
+        acc = mont_sqr<F>(acc);
+        if ((e.v[i >> 6] >> (i & 63)) & 1) acc = mont_mul<F>(acc, base);
+    }
+    return acc;
+}
+
+// inverse via Fermat: x^(MOD-2)
+template <typename F>
+__device__ __forceinline__ fe4 mont_inv(const fe4 &x) {
+    fe4 e{{F::MOD[0] - 2, F::MOD[1], F::MOD[2], F::MOD[3]}};  // MOD odd => no borrow
+    return mont_pow<F>(x, e);
+}
+
+// ---- big-endian byte conversion ----
+
+__device__ __host__ __forceinline__ u64 bswap64(u64 x) {
+#ifdef __HIP_DEVICE_COMPILE__
+    return __builtin_bswap64(x);
+#else
+    return __builtin_bswap64(x);
+#endif
+}
+
+// load 32 big-endian bytes -> 4x64 LE limbs (no reduction)
+__device__ __host__ __forceinline__ fe4 fe_from_be(const uint8_t *b) {
+    const u64 *w = (const u64 *)b;  // byte buffers are 8-aligned by ABI contract
+    fe4 r;
+    r.v[3] = bswap64(w[0]);
+    r.v[2] = bswap64(w[1]);
+    r.v[1] = bswap64(w[2]);
+    r.v[0] = bswap64(w[3]);
+    return r;
+}
+
+__device__ __host__ __forceinline__ void fe_to_be(uint8_t *b, const fe4 &x) {
+    u64 *w = (u64 *)b;
+    w[0] = bswap64(x.v[3]);
+    w[1] = bswap64(x.v[2]);
+    w[2] = bswap64(x.v[1]);
+    w[3] = bswap64(x.v[0]);
+}
+
+}  // namespace em

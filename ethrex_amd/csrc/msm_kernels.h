@@ -1,0 +1,283 @@
+// ============================================================================
+// Pippenger bucket MSM over BN254 G1 — gfx950 kernels.
+//
+// Window size c = 16 => 16 windows over the 254-bit scalar, 65535 live
+// buckets per window.  Pipeline per run:
+//   1. parse/reduce scalars (ark from_be_bytes_mod_order semantics)
+//   2. digit extraction -> (key = window<<16 | digit, value = point index)
+//   3. device radix sort of the 16n pairs on 20 key bits (rocPRIM)
+//   4. per-bucket segment offsets by binary search in the sorted keys
+//   5. bucket accumulation: one thread per bucket walks its run of sorted
+//      point indices with Jacobian+affine mixed adds (VALU-bound hot kernel)
+//   6. two-level running-sum bucket reduction (256-bucket segments, then
+//      per-window combine), window Horner combine, affine conversion.
+//
+// Work shape: ~16*(n + 2*65536) mixed adds; HBM traffic is only the
+// gathered 128-B points + sorted pairs => VALU-bound, so
+// there is deliberately no MFMA anywhere here.
+// ============================================================================
+#pragma once
+#include <hip/hip_runtime.h>
+#include "gpu_g1.h"
+
+namespace em {
+
+constexpr int MSM_C = 16;                      // window bits
+constexpr int MSM_NWIN = 16;                   // ceil(254/16)
+constexpr uint32_t MSM_NBUCKET_TOTAL = (uint32_t)MSM_NWIN << MSM_C;  // 1M ids
+constexpr int MSM_SEG = 256;                   // buckets per reduction segment
+constexpr int MSM_NSEG = (1 << MSM_C) / MSM_SEG;  // 256 segments per window
+
+// ---- input parsing ----
+
+// 64-byte BE affine -> Montgomery g1a + infinity flag; off-curve -> err.
+// Coordinates reduced mod p, (0,0) = identity (provider.rs:252-268).
+__global__ void k_parse_points(const uint8_t *__restrict__ in,
+                               g1a *__restrict__ pts, uint8_t *__restrict__ inf,
+                               size_t n, uint32_t *__restrict__ err) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    fe4 x = to_mont<Fq>(fe_from_be(in + 64 * i));
+    fe4 y = to_mont<Fq>(fe_from_be(in + 64 * i + 32));
+    if (fe_is_zero(x) && fe_is_zero(y)) {
+        inf[i] = 1;
+        pts[i].x = x;
+        pts[i].y = y;
+        return;
+    }
+    g1a p{x, y};
+    inf[i] = 0;
+    if (!g1a_on_curve(p)) atomicOr(err, 1u);
+    pts[i] = p;
+}
+
+// P_i = (start+i+1)*G directly in HBM; per-thread affine conversion.
+__global__ void k_gen_points(g1a *__restrict__ pts, uint8_t *__restrict__ inf,
+                             size_t n, uint64_t start) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t k = start + i + 1;
+    g1a g = g1_generator();
+    g1j acc = g1_inf();
+    for (int b = 63; b >= 0; b--) {
+        acc = g1_dbl(acc);
+        if ((k >> b) & 1) acc = g1_add_affine(acc, g);
+    }
+    // to affine (k >= 1 and k < r => never infinity)
+    fe4 zi = mont_inv<Fq>(acc.z);
+    fe4 zi2 = mont_sqr<Fq>(zi);
+    pts[i].x = mont_mul<Fq>(acc.x, zi2);
+    pts[i].y = mont_mul<Fq>(acc.y, mont_mul<Fq>(zi2, zi));
+    inf[i] = 0;
+}
+
+// download points as BE affine bytes (for parity tests)
+__global__ void k_points_to_be(const g1a *__restrict__ pts,
+                               const uint8_t *__restrict__ inf,
+                               uint8_t *__restrict__ out, size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    if (inf[i]) {
+        for (int j = 0; j < 8; j++) ((u64 *)(out + 64 * i))[j] = 0;
+        return;
+    }
+    fe_to_be(out + 64 * i, from_mont<Fq>(pts[i].x));
+    fe_to_be(out + 64 * i + 32, from_mont<Fq>(pts[i].y));
+}
+
+// 32-byte BE scalars -> canonical fe4 reduced mod r (from_be_bytes_mod_order)
+__global__ void k_parse_scalars(const uint8_t *__restrict__ in,
+                                fe4 *__restrict__ out, size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    out[i] = from_mont<Fr>(to_mont<Fr>(fe_from_be(in + 32 * i)));
+}
+
+// ---- digit extraction: c=16 => digit w is the w-th u16 of the scalar ----
+__global__ void k_digits(const fe4 *__restrict__ scalars,
+                         const uint8_t *__restrict__ inf,
+                         uint32_t *__restrict__ keys, uint32_t *__restrict__ vals,
+                         size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    fe4 k = scalars[i];
+    bool skip = inf[i];
+#pragma unroll
+    for (int w = 0; w < MSM_NWIN; w++) {
+        uint32_t d = (uint32_t)((k.v[w >> 2] >> ((w & 3) * 16)) & 0xffff);
+        if (skip) d = 0;  // identity points contribute nothing: park in bucket 0
+        keys[(size_t)w * n + i] = ((uint32_t)w << 16) | d;
+        vals[(size_t)w * n + i] = (uint32_t)i;
+    }
+}
+
+// ---- bucket segment offsets: lower_bound of each bucket id ----
+__global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total,
+                          uint32_t *__restrict__ offsets) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b > MSM_NBUCKET_TOTAL) return;
+    if (b == MSM_NBUCKET_TOTAL) {
+        offsets[b] = (uint32_t)total;
+        return;
+    }
+    // lower bound of key b
+    size_t lo = 0, hi = total;
+    while (lo < hi) {
+        size_t mid = (lo + hi) >> 1;
+        if (sorted_keys[mid] < b)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    offsets[b] = (uint32_t)lo;
+}
+
+// ---- bucket accumulation (the hot kernel) ----
+// one thread per bucket id; digit-0 buckets are skipped (never read later).
+__global__ void __launch_bounds__(256)
+k_bucket_acc(const g1a *__restrict__ pts, const uint32_t *__restrict__ vals,
+             const uint32_t *__restrict__ offsets, g1j *__restrict__ buckets) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= MSM_NBUCKET_TOTAL) return;
+    if ((b & 0xffff) == 0) return;  // digit 0
+    uint32_t lo = offsets[b], hi = offsets[b + 1];
+    g1j acc = g1_inf();
+    for (uint32_t t = lo; t < hi; t++) {
+        uint32_t i = vals[t];
+        acc = g1_add_affine(acc, pts[i]);
+    }
+    buckets[b] = acc;
+}
+
+// ---- two-level running-sum reduction ----
+// level 1: per (window, 256-bucket segment): from the top digit down,
+//   run  += B_d           (=> run  = sum of segment buckets)
+//   wsum += run           (=> wsum = sum (d - lo + 1) * B_d)
+__global__ void k_segment_reduce(const g1j *__restrict__ buckets,
+                                 g1j *__restrict__ seg_sum,
+                                 g1j *__restrict__ seg_wsum) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;  // [0, 16*256)
+    if (t >= MSM_NWIN * MSM_NSEG) return;
+    uint32_t w = t / MSM_NSEG, seg = t % MSM_NSEG;
+    uint32_t lo = seg * MSM_SEG;
+    g1j run = g1_inf(), wsum = g1_inf();
+    for (int32_t d = (int32_t)lo + MSM_SEG - 1; d >= (int32_t)lo; d--) {
+        if (d == 0) break;  // digit 0 bucket unused
+        run = g1_add(run, buckets[((uint32_t)w << 16) | (uint32_t)d]);
+        wsum = g1_add(wsum, run);
+    }
+    seg_sum[t] = run;
+    seg_wsum[t] = wsum;
+}
+
+// level 2: per window (16 threads):
+//   W = sum_j wsum_j + sum_j (lo_j - 1) * sum_j
+//     = A - B + 256 * (Wacc - B),  A = sum wsum_j, B = sum sum_j,
+//       Wacc = sum (j+1) * sum_j  (running-sum over segments)
+__global__ void k_window_reduce(const g1j *__restrict__ seg_sum,
+                                const g1j *__restrict__ seg_wsum,
+                                g1j *__restrict__ windows) {
+    uint32_t w = blockIdx.x * blockDim.x + threadIdx.x;
+    if (w >= MSM_NWIN) return;
+    g1j A = g1_inf(), runs = g1_inf(), wacc = g1_inf();
+    for (int j = MSM_NSEG - 1; j >= 0; j--) {
+        uint32_t t = w * MSM_NSEG + j;
+        A = g1_add(A, seg_wsum[t]);
+        runs = g1_add(runs, seg_sum[t]);
+        wacc = g1_add(wacc, runs);
+    }
+    // W = A - B + 256*(Wacc - B);  -B as adding negated point
+    g1j B = runs;  // = sum_j sum_j
+    g1j t1 = wacc;
+    // negate B: y -> -y
+    g1j negB = B;
+    if (!g1_is_inf(negB)) negB.y = mod_sub<Fq>(fe4{{0, 0, 0, 0}}, negB.y);
+    t1 = g1_add(t1, negB);               // Wacc - B
+    for (int d = 0; d < 8; d++) t1 = g1_dbl(t1);  // *256
+    g1j W = g1_add(A, negB);             // A - B
+    W = g1_add(W, t1);
+    windows[w] = W;
+}
+
+// ---- final Horner combine + output ----
+// out_mode 0: 64-byte BE affine (infinity -> zeros)
+// out_mode 1: 96-byte BE Jacobian canonical X||Y||Z (Z=0 -> infinity)
+__global__ void k_final_combine(const g1j *__restrict__ windows,
+                                uint8_t *__restrict__ out, int out_mode) {
+    if (blockIdx.x != 0 || threadIdx.x != 0) return;
+    g1j acc = windows[MSM_NWIN - 1];
+    for (int w = MSM_NWIN - 2; w >= 0; w--) {
+        for (int d = 0; d < MSM_C; d++) acc = g1_dbl(acc);
+        acc = g1_add(acc, windows[w]);
+    }
+    if (out_mode == 0) {
+        g1_to_affine_be(out, acc);
+    } else {
+        if (g1_is_inf(acc)) {
+            for (int j = 0; j < 12; j++) ((u64 *)out)[j] = 0;
+        } else {
+            fe_to_be(out, from_mont<Fq>(acc.x));
+            fe_to_be(out + 32, from_mont<Fq>(acc.y));
+            fe_to_be(out + 64, from_mont<Fq>(acc.z));
+        }
+    }
+}
+
+// ---- single-op kernels (zisk-mirror ABI + Jacobian combine) ----
+
+__global__ void k_g1_add_single(const uint8_t *in /* 128 B: p1||p2 */,
+                                uint8_t *out, uint32_t *err) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    fe4 x1 = to_mont<Fq>(fe_from_be(in));
+    fe4 y1 = to_mont<Fq>(fe_from_be(in + 32));
+    fe4 x2 = to_mont<Fq>(fe_from_be(in + 64));
+    fe4 y2 = to_mont<Fq>(fe_from_be(in + 96));
+    g1j acc = g1_inf();
+    if (!(fe_is_zero(x1) && fe_is_zero(y1))) {
+        g1a p{x1, y1};
+        if (!g1a_on_curve(p)) { atomicOr(err, 1u); return; }
+        acc = g1_add_affine(acc, p);
+    }
+    if (!(fe_is_zero(x2) && fe_is_zero(y2))) {
+        g1a p{x2, y2};
+        if (!g1a_on_curve(p)) { atomicOr(err, 1u); return; }
+        acc = g1_add_affine(acc, p);
+    }
+    g1_to_affine_be(out, acc);
+}
+
+__global__ void k_g1_mul_single(const uint8_t *in /* 96 B: point||scalar */,
+                                uint8_t *out, uint32_t *err) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    fe4 x = to_mont<Fq>(fe_from_be(in));
+    fe4 y = to_mont<Fq>(fe_from_be(in + 32));
+    if (fe_is_zero(x) && fe_is_zero(y)) {
+        for (int j = 0; j < 8; j++) ((u64 *)out)[j] = 0;
+        return;
+    }
+    g1a p{x, y};
+    if (!g1a_on_curve(p)) { atomicOr(err, 1u); return; }
+    fe4 k = from_mont<Fr>(to_mont<Fr>(fe_from_be(in + 64)));
+    if (fe_is_zero(k)) {
+        for (int j = 0; j < 8; j++) ((u64 *)out)[j] = 0;
+        return;
+    }
+    g1_to_affine_be(out, g1_scalar_mul(p, k));
+}
+
+// combine count Jacobian partials (96-B BE canonical each) -> affine
+__global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
+                             uint8_t *__restrict__ out) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    g1j acc = g1_inf();
+    for (size_t i = 0; i < count; i++) {
+        g1j p;
+        p.x = to_mont<Fq>(fe_from_be(in + 96 * i));
+        p.y = to_mont<Fq>(fe_from_be(in + 96 * i + 32));
+        p.z = to_mont<Fq>(fe_from_be(in + 96 * i + 64));
+        acc = g1_add(acc, p);
+    }
+    g1_to_affine_be(out, acc);
+}
+
+}  // namespace em

@@ -1,0 +1,201 @@
+"""ctypes binding to libethrex_mi355.so (the gfx950 HIP core).
+
+Loads the in-tree shared library built by __graft_entry__.build().  Import
+fails loudly if the library is absent: the product path never falls back
+to CPU.
+"""
+import ctypes
+import os
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "libethrex_mi355.so")
+
+if not os.path.exists(_SO):
+    raise ImportError(
+        f"HIP core library missing: {_SO}. Run __graft_entry__.build() "
+        "(hipcc --offload-arch=gfx950). The product path has no CPU fallback.")
+
+_lib = ctypes.CDLL(_SO)
+
+EM_OK = 0
+EM_ERR_POINT = 1
+EM_ERR_INPUT = 2
+EM_ERR_HIP = 3
+
+_lib.ethrex_mi355_version.restype = ctypes.c_char_p
+_lib.ethrex_mi355_last_error.restype = ctypes.c_char_p
+for _f in ("device_count", "set_device", "bn254_g1_add", "bn254_g1_mul",
+           "bn254_g1_msm", "bn254_fr_ntt", "bn254_g1_combine",
+           "msm_plan_create", "msm_plan_destroy", "msm_upload_points",
+           "msm_gen_points", "msm_download_points", "msm_upload_scalars",
+           "msm_run", "msm_run_partial", "msm_last_times",
+           "ntt_plan_create", "ntt_plan_destroy", "ntt_upload", "ntt_run",
+           "ntt_download", "ntt_last_times"):
+    getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
+
+
+class HipCoreError(RuntimeError):
+    def __init__(self, rc, what):
+        self.rc = rc
+        super().__init__(f"{what}: rc={rc} ({last_error()})")
+
+
+def _check(rc, what):
+    if rc != EM_OK:
+        raise HipCoreError(rc, what)
+
+
+def _buf(b):
+    return (ctypes.c_uint8 * len(b)).from_buffer_copy(b)
+
+
+def version() -> str:
+    return _lib.ethrex_mi355_version().decode()
+
+
+def last_error() -> str:
+    return (_lib.ethrex_mi355_last_error() or b"").decode()
+
+
+def device_count() -> int:
+    n = ctypes.c_int(0)
+    _lib.ethrex_mi355_device_count(ctypes.byref(n))
+    return n.value
+
+
+def set_device(d: int):
+    _check(_lib.ethrex_mi355_set_device(ctypes.c_int(d)), "set_device")
+
+
+def g1_add(p1: bytes, p2: bytes):
+    out = (ctypes.c_uint8 * 64)()
+    rc = _lib.ethrex_mi355_bn254_g1_add(_buf(p1), _buf(p2), out)
+    return rc, bytes(out)
+
+
+def g1_mul(point: bytes, scalar: bytes):
+    out = (ctypes.c_uint8 * 64)()
+    rc = _lib.ethrex_mi355_bn254_g1_mul(_buf(point), _buf(scalar), out)
+    return rc, bytes(out)
+
+
+def g1_msm(points: bytes, scalars: bytes, n: int):
+    out = (ctypes.c_uint8 * 64)()
+    rc = _lib.ethrex_mi355_bn254_g1_msm(_buf(points), _buf(scalars),
+                                        ctypes.c_size_t(n), out)
+    return rc, bytes(out)
+
+
+def fr_ntt(elems: bytes, n: int, inverse: bool):
+    buf = _buf(elems)
+    rc = _lib.ethrex_mi355_bn254_fr_ntt(buf, ctypes.c_size_t(n),
+                                        ctypes.c_int(1 if inverse else 0))
+    return rc, bytes(buf)
+
+
+def g1_combine(jacobians: bytes, count: int):
+    out = (ctypes.c_uint8 * 64)()
+    rc = _lib.ethrex_mi355_bn254_g1_combine(_buf(jacobians),
+                                            ctypes.c_size_t(count), out)
+    return rc, bytes(out)
+
+
+def gen_fr(seed: int, n: int) -> bytes:
+    """Deterministic Fr elements (host-side; BASELINE.md input scheme)."""
+    out = (ctypes.c_uint8 * (32 * n))()
+    _lib.ethrex_mi355_gen_fr(ctypes.c_uint64(seed), ctypes.c_size_t(n), out)
+    return bytes(out)
+
+
+class MsmPlan:
+    """Device-resident MSM plan: upload once, run many (bench/pipeline)."""
+
+    def __init__(self, n: int):
+        self.n = n
+        self._p = ctypes.c_void_p()
+        _check(_lib.ethrex_mi355_msm_plan_create(ctypes.c_size_t(n),
+                                                 ctypes.byref(self._p)),
+               "msm_plan_create")
+
+    def upload_points(self, points: bytes):
+        _check(_lib.ethrex_mi355_msm_upload_points(self._p, _buf(points)),
+               "msm_upload_points")
+
+    def gen_points(self, start: int = 0):
+        _check(_lib.ethrex_mi355_msm_gen_points(self._p, ctypes.c_uint64(start)),
+               "msm_gen_points")
+
+    def download_points(self) -> bytes:
+        out = (ctypes.c_uint8 * (64 * self.n))()
+        _check(_lib.ethrex_mi355_msm_download_points(self._p, out),
+               "msm_download_points")
+        return bytes(out)
+
+    def upload_scalars(self, scalars: bytes):
+        _check(_lib.ethrex_mi355_msm_upload_scalars(self._p, _buf(scalars)),
+               "msm_upload_scalars")
+
+    def run(self) -> bytes:
+        out = (ctypes.c_uint8 * 64)()
+        _check(_lib.ethrex_mi355_msm_run(self._p, out), "msm_run")
+        return bytes(out)
+
+    def run_partial(self) -> bytes:
+        out = (ctypes.c_uint8 * 96)()
+        _check(_lib.ethrex_mi355_msm_run_partial(self._p, out), "msm_run_partial")
+        return bytes(out)
+
+    def last_times(self):
+        t = (ctypes.c_double * 5)()
+        _check(_lib.ethrex_mi355_msm_last_times(self._p, t), "msm_last_times")
+        return {"digits_sort_ms": t[0], "bucket_acc_ms": t[1],
+                "reduce_ms": t[2], "combine_ms": t[3], "total_ms": t[4]}
+
+    def destroy(self):
+        if self._p:
+            _lib.ethrex_mi355_msm_plan_destroy(self._p)
+            self._p = ctypes.c_void_p()
+
+    def __del__(self):
+        try:
+            self.destroy()
+        except Exception:
+            pass
+
+
+class NttPlan:
+    def __init__(self, n: int):
+        self.n = n
+        self._p = ctypes.c_void_p()
+        _check(_lib.ethrex_mi355_ntt_plan_create(ctypes.c_size_t(n),
+                                                 ctypes.byref(self._p)),
+               "ntt_plan_create")
+
+    def upload(self, elems: bytes):
+        _check(_lib.ethrex_mi355_ntt_upload(self._p, _buf(elems)), "ntt_upload")
+
+    def run(self, inverse: bool = False):
+        _check(_lib.ethrex_mi355_ntt_run(self._p,
+                                         ctypes.c_int(1 if inverse else 0)),
+               "ntt_run")
+
+    def download(self) -> bytes:
+        out = (ctypes.c_uint8 * (32 * self.n))()
+        _check(_lib.ethrex_mi355_ntt_download(self._p, out), "ntt_download")
+        return bytes(out)
+
+    def last_times(self):
+        t = (ctypes.c_double * 3)()
+        _check(_lib.ethrex_mi355_ntt_last_times(self._p, t), "ntt_last_times")
+        return {"bitrev_ms": t[0], "stages_ms": t[1], "total_ms": t[2]}
+
+    def destroy(self):
+        if self._p:
+            _lib.ethrex_mi355_ntt_plan_destroy(self._p)
+            self._p = ctypes.c_void_p()
+
+    def __del__(self):
+        try:
+            self.destroy()
+        except Exception:
+            pass

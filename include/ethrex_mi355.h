@@ -1,0 +1,108 @@
+/* ============================================================================
+ * ethrex_mi355 — C-ABI boundary of the MI355X-native BN254 MSM/NTT prover
+ * core.
+ *
+ * This is the drop-in seam a Rust host binds behind ethrex's `ProverBackend`
+ * trait (reference: crates/prover/src/backend/mod.rs:87-153; the backend's
+ * `prove` drives these entry points).  The FFI convention mirrors the
+ * in-repo ZisK accelerator ABI (crates/guest-program/src/crypto/zisk.rs:71-137):
+ *   - `int` status returns (0 = OK, nonzero = error),
+ *   - fixed-size byte structs, 8-byte aligned,
+ *   - caller-allocated out-parameters.
+ * The Rust-side binding a maintainer would add is shown in INTEGRATION.md.
+ *
+ * Byte encodings (reference semantics, crates/common/crypto/provider.rs:247-318):
+ *   - G1 affine point: 64 bytes big-endian x||y; (0,0) encodes the identity
+ *     on input and output (EIP-197; crates/vm/levm/src/precompiles.rs:792-795).
+ *     Coordinates are parsed `from_be_bytes_mod_order` (reduced mod p) and
+ *     validated on-curve (off-curve => EM_ERR_POINT).
+ *   - Scalar / Fr element: 32 bytes big-endian; scalars are reduced mod r
+ *     exactly as ark's `from_be_bytes_mod_order` does.  NTT elements must be
+ *     canonical (< r) => EM_ERR_INPUT otherwise.
+ *   - G1 Jacobian partial (multi-GPU exchange payload): 96 bytes big-endian
+ *     X||Y||Z canonical form; Z = 0 encodes the identity.
+ *
+ * Every compute entry point REQUIRES the MI355X GPU path: there is no CPU
+ * fallback anywhere behind this ABI (EM_ERR_HIP if no device).
+ * ==========================================================================*/
+#ifndef ETHREX_MI355_H
+#define ETHREX_MI355_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+enum {
+    EM_OK = 0,
+    EM_ERR_POINT = 1,   /* G1 point not on curve (provider.rs InvalidPoint) */
+    EM_ERR_INPUT = 2,   /* bad size / non-canonical Fr element / null ptr   */
+    EM_ERR_HIP = 3,     /* HIP runtime failure or no gfx950 device          */
+};
+
+/* library identification / device selection */
+const char *ethrex_mi355_version(void);
+int ethrex_mi355_device_count(int *count);
+int ethrex_mi355_set_device(int device);
+/* last HIP error string (for diagnostics after EM_ERR_HIP) */
+const char *ethrex_mi355_last_error(void);
+
+/* ---- single-op entry points (zisk.rs:100-102 zkvm_bn254_g1_add/mul
+ *      mirror; run on the GPU; used for on-device semantics parity) ---- */
+int ethrex_mi355_bn254_g1_add(const uint8_t p1[64], const uint8_t p2[64],
+                              uint8_t out[64]);
+int ethrex_mi355_bn254_g1_mul(const uint8_t point[64], const uint8_t scalar[32],
+                              uint8_t out[64]);
+
+/* ---- one-shot hot-path entry points (host buffers in, host buffer out;
+ *      the drop-in calls a Rust `ProverBackend::prove` makes) ---- */
+int ethrex_mi355_bn254_g1_msm(const uint8_t *points64, const uint8_t *scalars32,
+                              size_t n, uint8_t out[64]);
+int ethrex_mi355_bn254_fr_ntt(uint8_t *elems32, size_t n, int inverse);
+
+/* ---- planned API: device-resident buffers for repeated runs (bench /
+ *      pipeline use; inputs stay in HBM between runs) ---- */
+typedef struct em_msm_plan em_msm_plan;
+typedef struct em_ntt_plan em_ntt_plan;
+
+int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan);
+int ethrex_mi355_msm_plan_destroy(em_msm_plan *plan);
+/* upload host-side affine points (64 B BE each); validates on device */
+int ethrex_mi355_msm_upload_points(em_msm_plan *plan, const uint8_t *points64);
+/* generate P_i = (start+i+1)*G directly in HBM (deterministic inputs) */
+int ethrex_mi355_msm_gen_points(em_msm_plan *plan, uint64_t start);
+/* download the plan's points back as 64 B BE affine (parity checks) */
+int ethrex_mi355_msm_download_points(em_msm_plan *plan, uint8_t *out64);
+int ethrex_mi355_msm_upload_scalars(em_msm_plan *plan, const uint8_t *scalars32);
+/* full MSM -> affine result */
+int ethrex_mi355_msm_run(em_msm_plan *plan, uint8_t out[64]);
+/* shard partial -> Jacobian 96 B (multi-GPU: AllGather these, then combine) */
+int ethrex_mi355_msm_run_partial(em_msm_plan *plan, uint8_t out[96]);
+/* combine Jacobian partials on the GPU -> affine result */
+int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96, size_t count,
+                                  uint8_t out[64]);
+/* per-phase HIP-event timings of the last run, milliseconds:
+ * [0]=digits+sort, [1]=bucket accumulation, [2]=bucket reduction,
+ * [3]=window combine + affine, [4]=total */
+int ethrex_mi355_msm_last_times(em_msm_plan *plan, double times_ms[5]);
+
+int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan);
+int ethrex_mi355_ntt_plan_destroy(em_ntt_plan *plan);
+int ethrex_mi355_ntt_upload(em_ntt_plan *plan, const uint8_t *elems32);
+int ethrex_mi355_ntt_run(em_ntt_plan *plan, int inverse);
+int ethrex_mi355_ntt_download(em_ntt_plan *plan, uint8_t *elems32);
+/* [0]=bit-reverse, [1]=butterfly stages total, [2]=total */
+int ethrex_mi355_ntt_last_times(em_ntt_plan *plan, double times_ms[3]);
+
+/* ---- deterministic input generation (host-side; the product restatement
+ *      of BASELINE.md's xoshiro256++/splitmix64 scheme; parity-tested
+ *      against the oracle's independent restatement) ---- */
+void ethrex_mi355_gen_fr(uint64_t seed, size_t n, uint8_t *out32);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* ETHREX_MI355_H */

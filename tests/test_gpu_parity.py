@@ -1,0 +1,180 @@
+"""GPU parity tests (the parity gate proper, tier rule ③): the HIP/gfx950
+path through the C-ABI vs the CPU oracle, on the committed golden vectors,
+on seeded random inputs at oracle-feasible sizes, and via size-independent
+identities at large sizes.  Bit-exact throughout (integer field work).
+"""
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+R = 0x30644E72E131A029B85045B68181585D2833E84879B9709143E1F593F0000001
+
+
+@pytest.fixture(scope="module")
+def gpu():
+    import ethrex_amd
+    if ethrex_amd.device_count() < 1:
+        pytest.skip("no GPU")
+    ethrex_amd.set_device(0)
+    return ethrex_amd
+
+
+def _b(h):
+    return bytes.fromhex(h)
+
+
+# ---- golden vectors through the GPU single-op ABI ----
+
+def test_gpu_g1_add_golden(gpu, golden):
+    for v in golden["g1_add"]:
+        rc, out = gpu.g1_add(_b(v["a"]), _b(v["b"]))
+        if v.get("error"):
+            assert rc == gpu.EM_ERR_POINT, v["name"]
+        else:
+            assert rc == 0 and out == _b(v["out"]), v["name"]
+
+
+def test_gpu_g1_mul_golden(gpu, golden):
+    for v in golden["g1_mul"]:
+        rc, out = gpu.g1_mul(_b(v["point"]), _b(v["scalar"]))
+        assert rc == 0 and out == _b(v["out"]), v["name"]
+
+
+def test_gpu_msm_golden(gpu, golden):
+    for v in golden["msm"]:
+        rc, out = gpu.g1_msm(_b(v["points"]), _b(v["scalars"]), v["n"])
+        assert rc == 0 and out == _b(v["out"]), f"msm n={v['n']}"
+
+
+def test_gpu_ntt_golden(gpu, golden):
+    for v in golden["ntt"]:
+        rc, fwd = gpu.fr_ntt(_b(v["in"]), v["n"], False)
+        assert rc == 0 and fwd == _b(v["fwd"]), f"ntt n={v['n']}"
+        rc, back = gpu.fr_ntt(fwd, v["n"], True)
+        assert rc == 0 and back == _b(v["inv_of_fwd"])
+
+
+# ---- random-input parity vs oracle ----
+
+def test_gpu_msm_parity_small(gpu, oracle_mod):
+    for n in (1, 2, 3, 100, 4096):
+        pts = oracle_mod.gen_points(0, n)
+        scs = oracle_mod.gen_fr(42, n)
+        rc, got = gpu.g1_msm(pts, scs, n)
+        rc2, want = oracle_mod.g1_msm(pts, scs, n)
+        assert rc == rc2 == 0 and got == want, n
+
+
+def test_gpu_msm_parity_edge_scalars(gpu, oracle_mod):
+    """zero scalars, scalar=r (reduces to 0), max scalar, identity points."""
+    n = 64
+    pts = bytearray(oracle_mod.gen_points(0, n))
+    scs = bytearray(oracle_mod.gen_fr(42, n))
+    scs[0:32] = b"\x00" * 32                      # k_0 = 0
+    scs[32:64] = R.to_bytes(32, "big")            # k_1 = r -> 0
+    scs[64:96] = (R - 1).to_bytes(32, "big")      # k_2 = r-1
+    scs[96:128] = ((1 << 256) - 1).to_bytes(32, "big")  # over-order
+    pts[64 * 5:64 * 6] = b"\x00" * 64             # P_5 = identity
+    pts, scs = bytes(pts), bytes(scs)
+    rc, got = gpu.g1_msm(pts, scs, n)
+    rc2, want = oracle_mod.g1_msm(pts, scs, n)
+    assert rc == rc2 == 0 and got == want
+    rc3, want_naive = oracle_mod.g1_msm_naive(pts, scs, n)
+    assert rc3 == 0 and got == want_naive
+
+
+def test_gpu_msm_rejects_offcurve(gpu):
+    bad = (1).to_bytes(32, "big") + (1).to_bytes(32, "big")
+    rc, _ = gpu.g1_msm(bad, (5).to_bytes(32, "big"), 1)
+    assert rc == gpu.EM_ERR_POINT
+
+
+def test_gpu_msm_duplicate_points(gpu, oracle_mod):
+    """All points identical -> every bucket run hits the doubling branch."""
+    n = 1024
+    g = (1).to_bytes(32, "big") + (2).to_bytes(32, "big")
+    pts = g * n
+    scs = oracle_mod.gen_fr(44, n)
+    rc, got = gpu.g1_msm(pts, scs, n)
+    k = sum(int.from_bytes(scs[32 * i:32 * i + 32], "big") for i in range(n)) % R
+    rc2, want = oracle_mod.g1_mul(g, k.to_bytes(32, "big"))
+    assert rc == rc2 == 0 and got == want
+
+
+def test_gpu_msm_parity_2_16(gpu, oracle_mod):
+    n = 1 << 16
+    plan = gpu.MsmPlan(n)
+    plan.gen_points(0)
+    pts = plan.download_points()
+    assert pts == oracle_mod.gen_points(0, n), "device gen_points parity"
+    scs = gpu.gen_fr(42, n)
+    plan.upload_scalars(scs)
+    got = plan.run()
+    plan.destroy()
+    rc, want = oracle_mod.g1_msm(pts, scs, n)
+    assert rc == 0 and got == want
+
+
+def test_gpu_msm_shard_combine_parity(gpu, oracle_mod):
+    """4-shard partial-sum path (the RCCL exchange payload) vs full MSM and
+    vs the oracle's shard combine."""
+    n, shards = 1 << 14, 4
+    sh = n // shards
+    parts = b""
+    all_pts, all_scs = b"", b""
+    for s in range(shards):
+        plan = gpu.MsmPlan(sh)
+        plan.gen_points(s * sh)
+        scs = gpu.gen_fr(42 + s, sh)  # per-rank seed
+        plan.upload_scalars(scs)
+        parts += plan.run_partial()
+        all_pts += plan.download_points()
+        all_scs += scs
+        plan.destroy()
+    rc, got = gpu.g1_combine(parts, shards)
+    assert rc == 0
+    rc, want = oracle_mod.g1_msm(all_pts, all_scs, n)
+    assert rc == 0 and got == want
+    rc, want2 = oracle_mod.g1_combine_jacobian(parts, shards)
+    assert rc == 0 and got == want2
+
+
+def test_gpu_ntt_parity(gpu, oracle_mod):
+    for n in (1, 2, 256, 1 << 12, 1 << 16):
+        elems = gpu.gen_fr(43, n)
+        rc, fwd = gpu.fr_ntt(elems, n, False)
+        rc2, want = oracle_mod.fr_ntt(elems, n, False)
+        assert rc == rc2 == 0 and fwd == want, n
+        rc, back = gpu.fr_ntt(fwd, n, True)
+        assert rc == 0 and back == elems, n
+
+
+def test_gpu_ntt_rejects_noncanonical(gpu):
+    rc, _ = gpu.fr_ntt(b"\xff" * 32 * 2, 2, False)
+    assert rc == gpu.EM_ERR_INPUT
+
+
+# ---- large-size identities (oracle-infeasible sizes; tier rule ③) ----
+
+def test_gpu_msm_2_20_sum_identity(gpu, oracle_mod):
+    """All points = G at 2^20: MSM == (sum k_i mod r) * G."""
+    n = 1 << 20
+    g = (1).to_bytes(32, "big") + (2).to_bytes(32, "big")
+    plan = gpu.MsmPlan(n)
+    plan.upload_points(g * n)
+    scs = gpu.gen_fr(45, n)
+    plan.upload_scalars(scs)
+    got = plan.run()
+    plan.destroy()
+    k = sum(int.from_bytes(scs[32 * i:32 * i + 32], "big") for i in range(n)) % R
+    rc, want = oracle_mod.g1_mul(g, k.to_bytes(32, "big"))
+    assert rc == 0 and got == want
+
+
+def test_gpu_ntt_roundtrip_2_20(gpu):
+    n = 1 << 20
+    elems = gpu.gen_fr(43, n)
+    rc, fwd = gpu.fr_ntt(elems, n, False)
+    assert rc == 0 and fwd != elems
+    rc, back = gpu.fr_ntt(fwd, n, True)
+    assert rc == 0 and back == elems
