@@ -162,8 +162,11 @@ __global__ void k_segment_reduce(const g1j *__restrict__ buckets,
     uint32_t lo = seg * MSM_SEG;
     g1j run = g1_inf(), wsum = g1_inf();
     for (int32_t d = (int32_t)lo + MSM_SEG - 1; d >= (int32_t)lo; d--) {
-        if (d == 0) break;  // digit 0 bucket unused
-        run = g1_add(run, buckets[((uint32_t)w << 16) | (uint32_t)d]);
+        // digit-0 bucket is unused: skip the add but keep the wsum step so
+        // segment 0 carries the same (d - lo + 1) weights as every other
+        // segment (verified integer model: see DESIGN.md "Bucket reduction")
+        if (d != 0)
+            run = g1_add(run, buckets[((uint32_t)w << 16) | (uint32_t)d]);
         wsum = g1_add(wsum, run);
     }
     seg_sum[t] = run;
