@@ -90,6 +90,7 @@ struct em_msm_plan {
     g1j *d_buckets = nullptr;         // NBUCKET_TOTAL
     g1j *d_seg_sum = nullptr;         // NWIN*NSEG
     g1j *d_seg_wsum = nullptr;
+    g1j *d_partials = nullptr;        // NWIN*NBLK_PER_WIN
     g1j *d_windows = nullptr;         // NWIN
     uint8_t *d_out = nullptr;         // 96 B
     uint32_t *d_err = nullptr;
@@ -122,6 +123,7 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1j));
     mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1j));
     mal((void **)&p->d_seg_wsum, MSM_NWIN * MSM_NSEG * sizeof(g1j));
+    mal((void **)&p->d_partials, MSM_NWIN * MSM_NBLK_PER_WIN * sizeof(g1j));
     mal((void **)&p->d_windows, MSM_NWIN * sizeof(g1j));
     mal((void **)&p->d_out, 96);
     mal((void **)&p->d_err, 4);
@@ -156,6 +158,7 @@ extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
     hipFree(p->d_buckets);
     hipFree(p->d_seg_sum);
     hipFree(p->d_seg_wsum);
+    hipFree(p->d_partials);
     hipFree(p->d_windows);
     hipFree(p->d_out);
     hipFree(p->d_err);
@@ -236,8 +239,11 @@ static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
     hipLaunchKernelGGL(k_segment_reduce,
                        dim3(blocks_for(MSM_NWIN * MSM_NSEG, 256)), dim3(256), 0, 0,
                        p->d_buckets, p->d_seg_sum, p->d_seg_wsum);
-    hipLaunchKernelGGL(k_window_reduce, dim3(1), dim3(64), 0, 0, p->d_seg_sum,
-                       p->d_seg_wsum, p->d_windows);
+    hipLaunchKernelGGL(k_weighted_reduce,
+                       dim3(MSM_NWIN * MSM_NBLK_PER_WIN), dim3(MSM_RED_BLOCK), 0,
+                       0, p->d_seg_sum, p->d_seg_wsum, p->d_partials);
+    hipLaunchKernelGGL(k_window_sum, dim3(1), dim3(64), 0, 0, p->d_partials,
+                       p->d_windows);
     HIP_TRY(hipEventRecord(p->ev[3], 0));
     hipLaunchKernelGGL(k_final_combine, dim3(1), dim3(64), 0, 0, p->d_windows,
                        p->d_out, out_mode);

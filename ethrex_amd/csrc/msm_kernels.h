@@ -25,8 +25,10 @@ namespace em {
 constexpr int MSM_C = 16;                      // window bits
 constexpr int MSM_NWIN = 16;                   // ceil(254/16)
 constexpr uint32_t MSM_NBUCKET_TOTAL = (uint32_t)MSM_NWIN << MSM_C;  // 1M ids
-constexpr int MSM_SEG = 256;                   // buckets per reduction segment
-constexpr int MSM_NSEG = (1 << MSM_C) / MSM_SEG;  // 256 segments per window
+constexpr int MSM_SEG = 32;                    // buckets per reduction segment
+constexpr int MSM_NSEG = (1 << MSM_C) / MSM_SEG;  // 2048 segments per window
+constexpr int MSM_RED_BLOCK = 256;             // threads per level-2 block
+constexpr int MSM_NBLK_PER_WIN = MSM_NSEG / MSM_RED_BLOCK;  // 8
 
 // ---- input parsing ----
 
@@ -173,33 +175,57 @@ __global__ void k_segment_reduce(const g1j *__restrict__ buckets,
     seg_wsum[t] = wsum;
 }
 
-// level 2: per window (16 threads):
-//   W = sum_j wsum_j + sum_j (lo_j - 1) * sum_j
-//     = A - B + 256 * (Wacc - B),  A = sum wsum_j, B = sum sum_j,
-//       Wacc = sum (j+1) * sum_j  (running-sum over segments)
-__global__ void k_window_reduce(const g1j *__restrict__ seg_sum,
-                                const g1j *__restrict__ seg_wsum,
-                                g1j *__restrict__ windows) {
+// level 2: fully parallel weighted combine + LDS tree reduction.
+//   W_w = sum_j [ wsum_j + (j*SEG - 1) * sum_j ]   (j=0 term: -sum_0)
+// One thread per segment computes its weighted value (17-bit double-and-add
+// for the (j*SEG-1) scalar), then a 256-wide LDS tree sums the block; 8
+// blocks per window leave 16*8 partials for the tiny level-3 kernel.
+// (Integer-model verification: DESIGN.md "Bucket reduction".)
+__global__ void __launch_bounds__(MSM_RED_BLOCK)
+k_weighted_reduce(const g1j *__restrict__ seg_sum,
+                  const g1j *__restrict__ seg_wsum,
+                  g1j *__restrict__ partials /* NWIN*NBLK_PER_WIN */) {
+    __shared__ g1j lds[MSM_RED_BLOCK];
+    uint32_t t = blockIdx.x * MSM_RED_BLOCK + threadIdx.x;
+    uint32_t j = t % MSM_NSEG;
+    g1j ws = seg_wsum[t];
+    g1j ss = seg_sum[t];
+    g1j val;
+    if (j == 0) {
+        // weight -1: subtract sum_0
+        if (!g1_is_inf(ss)) ss.y = mod_sub<Fq>(fe4{{0, 0, 0, 0}}, ss.y);
+        val = g1_add(ws, ss);
+    } else {
+        uint32_t weight = j * MSM_SEG - 1;  // <= 65535
+        g1j acc = g1_inf();
+        for (int b = 16; b >= 0; b--) {
+            acc = g1_dbl(acc);
+            if ((weight >> b) & 1) acc = g1_add(acc, ss);
+        }
+        val = g1_add(ws, acc);
+    }
+    lds[threadIdx.x] = val;
+    __syncthreads();
+    for (int s = MSM_RED_BLOCK / 2; s > 0; s >>= 1) {
+        if (threadIdx.x < (uint32_t)s) {
+            g1j o = lds[threadIdx.x + s];
+            g1j m = g1_add(lds[threadIdx.x], o);
+            lds[threadIdx.x] = m;
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) partials[blockIdx.x] = lds[0];
+}
+
+// level 3: 16 threads, 8 partials each -> per-window sums
+__global__ void k_window_sum(const g1j *__restrict__ partials,
+                             g1j *__restrict__ windows) {
     uint32_t w = blockIdx.x * blockDim.x + threadIdx.x;
     if (w >= MSM_NWIN) return;
-    g1j A = g1_inf(), runs = g1_inf(), wacc = g1_inf();
-    for (int j = MSM_NSEG - 1; j >= 0; j--) {
-        uint32_t t = w * MSM_NSEG + j;
-        A = g1_add(A, seg_wsum[t]);
-        runs = g1_add(runs, seg_sum[t]);
-        wacc = g1_add(wacc, runs);
-    }
-    // W = A - B + 256*(Wacc - B);  -B as adding negated point
-    g1j B = runs;  // = sum_j sum_j
-    g1j t1 = wacc;
-    // negate B: y -> -y
-    g1j negB = B;
-    if (!g1_is_inf(negB)) negB.y = mod_sub<Fq>(fe4{{0, 0, 0, 0}}, negB.y);
-    t1 = g1_add(t1, negB);               // Wacc - B
-    for (int d = 0; d < 8; d++) t1 = g1_dbl(t1);  // *256
-    g1j W = g1_add(A, negB);             // A - B
-    W = g1_add(W, t1);
-    windows[w] = W;
+    g1j acc = g1_inf();
+    for (int b = 0; b < MSM_NBLK_PER_WIN; b++)
+        acc = g1_add(acc, partials[w * MSM_NBLK_PER_WIN + b]);
+    windows[w] = acc;
 }
 
 // ---- final Horner combine + output ----
