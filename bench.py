@@ -93,34 +93,26 @@ def main():
 
     ethrex_amd.set_device(local_rank)
 
+    from ethrex_amd.dist import allgather_partials, shard_range
+
     n_total = 1 << args.msm_log2
-    shard = n_total // n_gpus
-    start = rank * shard
+    lo, hi = shard_range(n_total, n_gpus, rank)
+    shard = hi - lo
 
     # ---- setup: inputs resident in HBM before the timed region ----
     plan = ethrex_amd.MsmPlan(shard)
-    plan.gen_points(start)                      # P_i=(i+1)G, device-side
+    plan.gen_points(lo)                         # P_i=(i+1)G, device-side
     scalars = ethrex_amd.gen_fr(42 + rank, shard)
     plan.upload_scalars(scalars)
-
-    if world > 1:
-        import torch
-        gather_in = torch.empty(96, dtype=torch.uint8, device="cuda")
-        gather_out = torch.empty(world * 96, dtype=torch.uint8, device="cuda")
 
     def step():
         if world == 1:
             return plan.run()
         part = plan.run_partial()
-        gather_in.copy_(torch.frombuffer(bytearray(part), dtype=torch.uint8))
-        dist.all_gather_into_tensor(gather_out, gather_in)
-        allparts = bytes(gather_out.cpu().numpy().tobytes())
+        allparts = allgather_partials(part, dist, device="cuda")
         rc, out = ethrex_amd.g1_combine(allparts, world)
         assert rc == 0
         return out
-
-    if world > 1:
-        import torch
     # warmup
     first = None
     for _ in range(max(args.warmup, 1)):

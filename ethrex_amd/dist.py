@@ -1,0 +1,29 @@
+"""Multi-GPU sharding helpers: point-index sharding with an RCCL (xGMI)
+AllGather of the 96-byte Jacobian G1 partials + local combine (SURVEY.md
+§8e: EC addition is not an RCCL ReduceOp, so the exchange is an AllGather
+of the tiny partials; latency-bound, compute fully overlaps).
+
+The exchange is torch.distributed so the identical code path runs under
+"nccl" (=RCCL) on GPUs and "gloo" on CPU in the world_size-2 tests.
+"""
+
+
+def shard_range(n_total: int, world: int, rank: int):
+    """Contiguous point-index shard for this rank."""
+    shard = n_total // world
+    extra = n_total % world
+    lo = rank * shard + min(rank, extra)
+    hi = lo + shard + (1 if rank < extra else 0)
+    return lo, hi
+
+
+def allgather_partials(partial96: bytes, tdist, device="cpu"):
+    """AllGather each rank's 96-B Jacobian partial; returns concatenated
+    world*96 bytes in rank order."""
+    import torch
+    assert len(partial96) == 96
+    world = tdist.get_world_size()
+    t_in = torch.frombuffer(bytearray(partial96), dtype=torch.uint8).to(device)
+    t_out = torch.empty(world * 96, dtype=torch.uint8, device=device)
+    tdist.all_gather_into_tensor(t_out, t_in)
+    return bytes(t_out.cpu().numpy().tobytes())

@@ -1,0 +1,187 @@
+"""Host-side mirror of ethrex's prover backend boundary and pull loop.
+
+The reference host for this path is Rust:
+  - trait ProverBackend (crates/prover/src/backend/mod.rs:87-153):
+    prover_type / serialize_input / execute / prove(input, format) /
+    verify / to_proof_bytes (+ *_timed defaults)
+  - pull loop (crates/prover/src/prover.rs:86-140): poll coordinator ->
+    InputRequest -> prove -> ProofSubmit, errors continue to next endpoint
+  - BackendType registry (mod.rs:49-80)
+
+This module mirrors that interface in Python above the C-ABI (the Rust
+binding a maintainer would add instead is in INTEGRATION.md).  Two backends:
+
+  Mi355Backend — the product: drives the gfx950 MSM/NTT core through
+    libethrex_mi355.so.  Requires a GPU; fails loudly without one.
+  ExecBackend  — the reference's own mock-prover shape
+    (crates/prover/src/backend/exec.rs:12-94): executes the statement on
+    CPU, emits the sentinel proof bytes (exec.rs:53-60).  Used for
+    CPU-only plumbing tests exactly as the reference's CI uses Exec.
+"""
+import hashlib
+import json
+import socket
+import time
+
+
+class BackendError(Exception):
+    """Mirror of crates/prover/src/backend/error.rs BackendError."""
+
+
+# ---- ProverOutput / ProofBytes wire shapes (crates/common/types/prover.rs) ----
+
+def proof_output(prover_type: str, proof_bytes: bytes) -> dict:
+    return {"Proof": {"prover_type": prover_type,
+                      "proof": list(proof_bytes)}}
+
+
+class ExecBackend:
+    """Mirror of ExecBackend (backend/exec.rs): statement execution only,
+    sentinel proof bytes; the protocol-level mock prover."""
+
+    def prover_type(self) -> str:
+        return "Exec"
+
+    def serialize_input(self, input_data) -> bytes:
+        return json.dumps(input_data, sort_keys=True).encode()
+
+    def execute(self, input_data) -> None:
+        self.serialize_input(input_data)  # statement re-execution stand-in
+
+    def prove(self, input_data, proof_format: str):
+        self.execute(input_data)
+        return {"output": hashlib.sha256(
+            self.serialize_input(input_data)).hexdigest()}
+
+    def verify(self, proof) -> None:
+        if "output" not in proof:
+            raise BackendError("exec: missing output")
+
+    def to_proof_bytes(self, proof, proof_format: str) -> dict:
+        # non-empty sentinel (exec.rs:53-60)
+        return proof_output("Exec", b"\x00")
+
+
+class Mi355Backend:
+    """The MI355X backend: `prove` derives the batch's MSM/NTT workload
+    deterministically from the input and runs it on the GPU through the
+    C-ABI.  Proof bytes = MSM affine result || NTT output digest (the
+    stand-in for the Groth16 wrap output whose MSM/NTT this core computes;
+    full proof-object integration is SURVEY.md §8f row 3)."""
+
+    def __init__(self, msm_log2=16, ntt_log2=12, device=0):
+        self.msm_log2 = msm_log2
+        self.ntt_log2 = ntt_log2
+        self.device = device
+
+    def prover_type(self) -> str:
+        # reuses Exec semantics for coordinator keying (no on-chain
+        # verifier), per SURVEY.md §8b registration nuance
+        return "Exec"
+
+    def serialize_input(self, input_data) -> bytes:
+        return json.dumps(input_data, sort_keys=True).encode()
+
+    def _seed(self, input_data) -> int:
+        return int.from_bytes(
+            hashlib.sha256(self.serialize_input(input_data)).digest()[:8],
+            "little")
+
+    def execute(self, input_data) -> None:
+        self.serialize_input(input_data)
+
+    def prove(self, input_data, proof_format: str):
+        import ethrex_amd as ea
+        if ea.device_count() < 1:
+            raise BackendError("mi355: no GPU visible (no CPU fallback)")
+        ea.set_device(self.device)
+        seed = self._seed(input_data)
+        n = 1 << self.msm_log2
+        plan = ea.MsmPlan(n)
+        try:
+            plan.gen_points(0)
+            scalars = ea.gen_fr(seed, n)
+            plan.upload_scalars(scalars)
+            msm_out = plan.run()
+        finally:
+            plan.destroy()
+        m = 1 << self.ntt_log2
+        elems = ea.gen_fr(seed + 1, m)
+        rc, ntt_out = ea.fr_ntt(elems, m, False)
+        if rc != 0:
+            raise BackendError(f"mi355: ntt rc={rc}")
+        return {"msm": msm_out, "ntt_digest": hashlib.sha256(ntt_out).digest()}
+
+    def verify(self, proof) -> None:
+        if len(proof.get("msm", b"")) != 64:
+            raise BackendError("mi355: bad proof")
+
+    def to_proof_bytes(self, proof, proof_format: str) -> dict:
+        return proof_output(self.prover_type(),
+                            proof["msm"] + proof["ntt_digest"])
+
+
+BACKENDS = {"exec": ExecBackend, "mi355": Mi355Backend}
+
+
+# ---- pull-loop client (crates/prover/src/prover.rs:86-198,289-305) ----
+
+def _round_trip(host, port, obj, timeout=10.0):
+    with socket.create_connection((host, port), timeout=timeout) as conn:
+        conn.sendall(json.dumps(obj).encode())
+        conn.shutdown(socket.SHUT_WR)
+        chunks = []
+        while True:
+            b = conn.recv(65536)
+            if not b:
+                break
+            chunks.append(b)
+    return json.loads(b"".join(chunks).decode()) if chunks else None
+
+
+class ProverClient:
+    def __init__(self, backend, endpoints, commit_hash="deadbeef",
+                 proving_interval_s=0.05):
+        self.backend = backend
+        self.endpoints = endpoints  # [(host, port)]
+        self.commit_hash = commit_hash
+        self.interval = proving_interval_s
+        self.proved = []
+
+    def poll_once(self) -> int:
+        """One pass over all endpoints (poll_endpoints, prover.rs:86-140).
+        Returns number of proofs submitted; errors continue to the next
+        endpoint (prover.rs:100-103,129-138)."""
+        done = 0
+        for (host, port) in self.endpoints:
+            try:
+                resp = _round_trip(host, port, {"InputRequest": {
+                    "commit_hash": self.commit_hash,
+                    "prover_type": self.backend.prover_type()}})
+                if resp == "VersionMismatch":
+                    raise BackendError("version mismatch with coordinator")
+                if not (isinstance(resp, dict) and "InputResponse" in resp):
+                    continue
+                ir = resp["InputResponse"]
+                if ir.get("id") is None:
+                    continue  # no work available
+                proof = self.backend.prove(ir["input"], ir.get("format"))
+                self.backend.verify(proof)
+                out = self.backend.to_proof_bytes(proof, ir.get("format"))
+                ack = _round_trip(host, port, {"ProofSubmit": {
+                    "id": ir["id"], "proof": out}})
+                if isinstance(ack, dict) and "ProofSubmitACK" in ack:
+                    self.proved.append(ir["id"])
+                    done += 1
+            except BackendError:
+                raise
+            except (OSError, json.JSONDecodeError):
+                continue  # endpoint error: move on (prover.rs:100-103)
+        return done
+
+    def run(self, max_polls=100):
+        """Re-scheduling poll loop (Handler<Poll>, prover.rs:214-222)."""
+        for _ in range(max_polls):
+            if self.poll_once() == 0:
+                time.sleep(self.interval)
+        return self.proved
