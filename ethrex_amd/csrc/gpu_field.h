@@ -18,6 +18,7 @@
 
 namespace em {
 
+using u32 = uint32_t;
 using u64 = uint64_t;
 using u128 = unsigned __int128;
 
@@ -104,54 +105,67 @@ __device__ __host__ __forceinline__ fe4 mod_dbl(const fe4 &a) {
     return mod_add<F>(a, a);
 }
 
-// ---- CIOS Montgomery multiplication ----
-// t has 6 limbs (4 + 2 guard); after each a_i row we fold in m_i * MOD.
-// Inputs < MOD  =>  output < MOD.
+// ---- CIOS Montgomery multiplication, 32-bit-limb form ----
+// On CDNA4 the 64-bit-limb form compiles to long VCC carry chains
+// (v_addc + mandatory s_nop wait states — measured 863 instructions); the
+// 32-bit-limb form below maps every step onto carry-free u64 arithmetic
+// (v_mad_u64_u32 + v_lshl_add_u64, zero s_nop / zero v_addc — 569
+// instructions), because each `c += (u64)a32*b32 + t32` fits u64 exactly.
+// Inputs < MOD  =>  output < MOD (standard CIOS bound; the to_mont case
+// with a < 2^256 and b = R2 < MOD also stays within the guard limb).
 
 template <typename F>
-__device__ __host__ __forceinline__ fe4 mont_mul(const fe4 &a, const fe4 &b) {
-    u64 t0 = 0, t1 = 0, t2 = 0, t3 = 0, t4 = 0, t5 = 0;
+__device__ __host__ __forceinline__ fe4 mont_mul(const fe4 &A, const fe4 &B) {
+    u32 a[8], b[8], n[8];
 #pragma unroll
     for (int i = 0; i < 4; i++) {
-        // t += a[i] * b
-        u128 c = (u128)a.v[i] * b.v[0] + t0;
-        t0 = (u64)c;
-        c = (c >> 64) + (u128)a.v[i] * b.v[1] + t1;
-        t1 = (u64)c;
-        c = (c >> 64) + (u128)a.v[i] * b.v[2] + t2;
-        t2 = (u64)c;
-        c = (c >> 64) + (u128)a.v[i] * b.v[3] + t3;
-        t3 = (u64)c;
-        c = (c >> 64) + t4;
-        t4 = (u64)c;
-        t5 = (u64)(c >> 64);
-
-        // m = t0 * n0inv;  t += m * MOD;  t >>= 64
-        u64 m = t0 * F::N0INV;
-        c = (u128)m * F::MOD[0] + t0;
-        c >>= 64;
-        c += (u128)m * F::MOD[1] + t1;
-        t0 = (u64)c;
-        c = (c >> 64) + (u128)m * F::MOD[2] + t2;
-        t1 = (u64)c;
-        c = (c >> 64) + (u128)m * F::MOD[3] + t3;
-        t2 = (u64)c;
-        c = (c >> 64) + t4;
-        t3 = (u64)c;
-        t4 = t5 + (u64)(c >> 64);
-        t5 = 0;
+        a[2 * i] = (u32)A.v[i];
+        a[2 * i + 1] = (u32)(A.v[i] >> 32);
+        b[2 * i] = (u32)B.v[i];
+        b[2 * i + 1] = (u32)(B.v[i] >> 32);
+        n[2 * i] = (u32)F::MOD[i];
+        n[2 * i + 1] = (u32)(F::MOD[i] >> 32);
     }
-    fe4 out{{t0, t1, t2, t3}};
-    // final conditional subtraction (t4 is 0 or 1)
-    bool ge = t4 != 0;
+    const u32 n0inv32 = (u32)F::N0INV;  // -MOD^-1 mod 2^32
+    u32 t[10] = {0, 0, 0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+    for (int i = 0; i < 8; i++) {
+        u64 c = 0;
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+            c += (u64)a[i] * b[j] + t[j];
+            t[j] = (u32)c;
+            c >>= 32;
+        }
+        c += t[8];
+        t[8] = (u32)c;
+        t[9] = (u32)(c >> 32);
+        u32 m = t[0] * n0inv32;
+        c = (u64)m * n[0] + t[0];
+        c >>= 32;
+#pragma unroll
+        for (int j = 1; j < 8; j++) {
+            c += (u64)m * n[j] + t[j];
+            t[j - 1] = (u32)c;
+            c >>= 32;
+        }
+        c += t[8];
+        t[7] = (u32)c;
+        t[8] = t[9] + (u32)(c >> 32);
+        t[9] = 0;
+    }
+    fe4 out{{(u64)t[0] | ((u64)t[1] << 32), (u64)t[2] | ((u64)t[3] << 32),
+             (u64)t[4] | ((u64)t[5] << 32), (u64)t[6] | ((u64)t[7] << 32)}};
+    // final conditional subtraction (t[8] is 0 or 1)
+    bool ge = t[8] != 0;
     if (!ge) ge = fe_geq(out, fe4{{F::MOD[0], F::MOD[1], F::MOD[2], F::MOD[3]}});
     if (ge) {
         u128 bor = 0;
 #pragma unroll
         for (int i = 0; i < 4; i++) {
-            u128 t = (u128)out.v[i] - F::MOD[i] - bor;
-            out.v[i] = (u64)t;
-            bor = (t >> 64) & 1;
+            u128 t2 = (u128)out.v[i] - F::MOD[i] - bor;
+            out.v[i] = (u64)t2;
+            bor = (t2 >> 64) & 1;
         }
     }
     return out;
