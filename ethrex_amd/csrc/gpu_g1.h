@@ -82,7 +82,7 @@ __device__ __forceinline__ g1j g1_add(const g1j &p, const g1j &q) {
 
 // mixed add: q affine (implicit z=1), q must not be infinity
 __device__ __forceinline__ g1j g1_add_affine(const g1j &p, const g1a &q) {
-    if (g1_is_inf(p)) {
+    if (__builtin_expect(g1_is_inf(p), 0)) {
         g1j o;
         o.x = q.x;
         o.y = q.y;
@@ -94,7 +94,7 @@ __device__ __forceinline__ g1j g1_add_affine(const g1j &p, const g1a &q) {
     fe4 s2 = mont_mul<Fq>(q.y, mont_mul<Fq>(p.z, z1z1));
     fe4 h = mod_sub<Fq>(u2, p.x);
     fe4 r = mod_sub<Fq>(s2, p.y);
-    if (fe_is_zero(h)) {
+    if (__builtin_expect(fe_is_zero(h), 0)) {
         if (fe_is_zero(r)) return g1_dbl(p);
         return g1_inf();
     }

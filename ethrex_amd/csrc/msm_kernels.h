@@ -144,9 +144,18 @@ k_bucket_acc(const g1a *__restrict__ pts, const uint32_t *__restrict__ vals,
     if ((b & 0xffff) == 0) return;  // digit 0
     uint32_t lo = offsets[b], hi = offsets[b + 1];
     g1j acc = g1_inf();
+    if (lo >= hi) {
+        buckets[b] = acc;
+        return;
+    }
+    // software pipeline: issue the NEXT point's gather before the ~7k-inst
+    // mixed add so the dependent idx->point load chain (~2x L2/HBM latency)
+    // overlaps the VALU work of the current add.
+    g1a p = pts[vals[lo]];
     for (uint32_t t = lo; t < hi; t++) {
-        uint32_t i = vals[t];
-        acc = g1_add_affine(acc, pts[i]);
+        g1a cur = p;
+        if (t + 1 < hi) p = pts[vals[t + 1]];
+        acc = g1_add_affine(acc, cur);
     }
     buckets[b] = acc;
 }
