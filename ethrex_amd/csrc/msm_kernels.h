@@ -154,7 +154,8 @@ k_bucket_acc(const g1a *__restrict__ pts, const uint32_t *__restrict__ vals,
     g1a p = pts[vals[lo]];
     for (uint32_t t = lo; t < hi; t++) {
         g1a cur = p;
-        if (t + 1 < hi) p = pts[vals[t + 1]];
+        uint32_t nxt = t + 1 < hi ? t + 1 : t;  // clamped: loads always issue
+        p = pts[vals[nxt]];
         acc = g1_add_affine(acc, cur);
     }
     buckets[b] = acc;
