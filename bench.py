@@ -35,7 +35,7 @@ def point_adds(n_total):
     return MSM_WINDOWS * (n_total + (1 << (MSM_C + 1)))
 
 
-def cpu_baseline_leg(log2n=20):
+def cpu_baseline_leg(log2n=22):
     """Oracle (CPU restatement of the reference's ark-bn254 path) timed on
     this box's host cores — tier rule ④'s `cpu_baseline`, kind="port".
     Bounded sample: 2^log2n points (~10-30 s of CPU work)."""

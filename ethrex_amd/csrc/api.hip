@@ -519,7 +519,7 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
                            0, cur, oth, N1, N2);
         HIP_TRY(hipEventRecord(p->ev[1], 0));
         // P1: row NTT_N1 over each of the N2 rows + w^(k1*c) twiddle
-        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512), 0, 0, oth,
+        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(1024), 0, 0, oth,
                            p->logN1, inverse ? p->d_twrow1_inv : p->d_twrow1,
                            inverse ? p->d_twfull_inv : p->d_twfull,
                            (const fe4 *)nullptr);
@@ -527,7 +527,7 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256), 0,
                            0, oth, cur, N2, N1);
         // P2: row NTT_N2 (+ 1/n scale on iNTT)
-        hipLaunchKernelGGL(k_ntt_row, dim3(N1), dim3(512), 0, 0, cur,
+        hipLaunchKernelGGL(k_ntt_row, dim3(N1), dim3(1024), 0, 0, cur,
                            p->logN2, inverse ? p->d_twrow2_inv : p->d_twrow2,
                            (const fe4 *)nullptr,
                            inverse ? p->d_ninv : (const fe4 *)nullptr);

@@ -109,7 +109,7 @@ k_transpose_fe4(const fe4 *__restrict__ src, fe4 *__restrict__ dst,
 // tw_full: if non-null, epilogue multiplies element k by tw_full[k*blockIdx.x]
 //          (the four-step inter-NTT twiddle; index < n always).
 // scale: if non-null, epilogue multiplies by *scale (iNTT 1/n, Montgomery).
-__global__ void __launch_bounds__(512)
+__global__ void __launch_bounds__(1024)
 k_ntt_row(fe4 *__restrict__ data, int logM, const fe4 *__restrict__ tw_row,
           const fe4 *__restrict__ tw_full, const fe4 *__restrict__ scale) {
     // static 128 KiB LDS (max row length 4096 fe4); gfx950 has 160 KiB/CU —
