@@ -72,7 +72,7 @@ __global__ void k_ciosstep(u64* out, u64 seed) {
 }
 
 static double run(void(*k)(u64*,u64), const char* name, int ops_per_inner) {
-    u64* d; (void)hipMalloc(&d, 256*256*8);
+    u64* d; (void)hipMalloc(&d, (size_t)2048*256*8);
     hipEvent_t e0,e1; (void)hipEventCreate(&e0); (void)hipEventCreate(&e1);
     // warm
     hipLaunchKernelGGL(k, dim3(2048), dim3(256), 0, 0, d, 12345ull);
