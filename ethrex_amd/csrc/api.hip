@@ -20,7 +20,7 @@
 
 #include "../../include/ethrex_mi355.h"
 #include "gpu_field.h"
-#include "gpu_g1.h"
+#include "gpu_g1_9.h"
 #include "msm_kernels.h"
 #include "ntt_kernels.h"
 
@@ -76,7 +76,7 @@ static inline uint32_t blocks_for(size_t n, int bs) {
 
 struct em_msm_plan {
     size_t n;
-    g1a *d_pts = nullptr;
+    g1a9 *d_pts = nullptr;
     uint8_t *d_inf = nullptr;
     fe4 *d_scalars = nullptr;
     uint8_t *d_scratch = nullptr;     // 64n bytes: point/scalar byte staging
@@ -87,11 +87,11 @@ struct em_msm_plan {
     void *d_sort_tmp = nullptr;
     size_t sort_tmp_bytes = 0;
     uint32_t *d_offsets = nullptr;    // NBUCKET_TOTAL + 1
-    g1j *d_buckets = nullptr;         // NBUCKET_TOTAL
-    g1j *d_seg_sum = nullptr;         // NWIN*NSEG
-    g1j *d_seg_wsum = nullptr;
-    g1j *d_partials = nullptr;        // NWIN*NBLK_PER_WIN
-    g1j *d_windows = nullptr;         // NWIN
+    g1j9 *d_buckets = nullptr;         // NBUCKET_TOTAL
+    g1j9 *d_seg_sum = nullptr;         // NWIN*NSEG
+    g1j9 *d_seg_wsum = nullptr;
+    g1j9 *d_partials = nullptr;        // NWIN*NBLK_PER_WIN
+    g1j9 *d_windows = nullptr;         // NWIN
     uint8_t *d_out = nullptr;         // 96 B
     uint32_t *d_err = nullptr;
     bool have_scalars = false;
@@ -111,7 +111,7 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     auto mal = [&](void **ptr, size_t bytes) {
         if (e == hipSuccess) e = hipMalloc(ptr, bytes);
     };
-    mal((void **)&p->d_pts, n * sizeof(g1a));
+    mal((void **)&p->d_pts, n * sizeof(g1a9));
     mal((void **)&p->d_inf, n);
     mal((void **)&p->d_scalars, n * sizeof(fe4));
     mal((void **)&p->d_scratch, n * 64);
@@ -120,11 +120,11 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     mal((void **)&p->d_keys_out, total * 4);
     mal((void **)&p->d_vals_out, total * 4);
     mal((void **)&p->d_offsets, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
-    mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1j));
-    mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1j));
-    mal((void **)&p->d_seg_wsum, MSM_NWIN * MSM_NSEG * sizeof(g1j));
-    mal((void **)&p->d_partials, MSM_NWIN * MSM_NBLK_PER_WIN * sizeof(g1j));
-    mal((void **)&p->d_windows, MSM_NWIN * sizeof(g1j));
+    mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1j9));
+    mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1j9));
+    mal((void **)&p->d_seg_wsum, MSM_NWIN * MSM_NSEG * sizeof(g1j9));
+    mal((void **)&p->d_partials, MSM_NWIN * MSM_NBLK_PER_WIN * sizeof(g1j9));
+    mal((void **)&p->d_windows, MSM_NWIN * sizeof(g1j9));
     mal((void **)&p->d_out, 96);
     mal((void **)&p->d_err, 4);
     if (e == hipSuccess) {

@@ -26,6 +26,9 @@ struct fe4 {
     u64 v[4];
 };
 
+using Fq = bn254::Fq;
+using Fr = bn254::Fr;
+
 __device__ __host__ __forceinline__ bool fe_is_zero(const fe4 &a) {
     return (a.v[0] | a.v[1] | a.v[2] | a.v[3]) == 0;
 }

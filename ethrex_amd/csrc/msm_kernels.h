@@ -3,22 +3,25 @@
 //
 // Window size c = 16 => 16 windows over the 254-bit scalar, 65535 live
 // buckets per window.  Pipeline per run:
-//   1. parse/reduce scalars (ark from_be_bytes_mod_order semantics)
+//   1. parse/reduce scalars (ark from_be_bytes_mod_order semantics; Fr stays
+//      on the 4x64 path — digit extraction wants packed u64 limbs)
 //   2. digit extraction -> (key = window<<16 | digit, value = point index)
 //   3. device radix sort of the 16n pairs on 20 key bits (rocPRIM)
 //   4. per-bucket segment offsets by binary search in the sorted keys
 //   5. bucket accumulation: one thread per bucket walks its run of sorted
-//      point indices with Jacobian+affine mixed adds (VALU-bound hot kernel)
-//   6. two-level running-sum bucket reduction (256-bucket segments, then
-//      per-window combine), window Horner combine, affine conversion.
+//      point indices with Jacobian+affine mixed adds — the VALU-bound hot
+//      kernel, on the 9x29-bit field core (gpu_field9.h)
+//   6. segment running sums -> weighted LDS-tree reduce (folds the 2^(16w)
+//      window factor) -> window sums -> 16-way combine + affine conversion.
 //
 // Work shape: ~16*(n + 2*65536) mixed adds; HBM traffic is only the
-// gathered 128-B points + sorted pairs => VALU-bound, so
+// gathered 72-B points + sorted pairs => VALU-bound, so
 // there is deliberately no MFMA anywhere here.
 // ============================================================================
 #pragma once
 #include <hip/hip_runtime.h>
-#include "gpu_g1.h"
+#include "gpu_field.h"   // fe4: Fr scalar handling
+#include "gpu_g1_9.h"    // fe9: Fq / G1 compute core
 
 namespace em {
 
@@ -32,49 +35,49 @@ constexpr int MSM_NBLK_PER_WIN = MSM_NSEG / MSM_RED_BLOCK;  // 8
 
 // ---- input parsing ----
 
-// 64-byte BE affine -> Montgomery g1a + infinity flag; off-curve -> err.
-// Coordinates reduced mod p, (0,0) = identity (provider.rs:252-268).
+// 64-byte BE affine -> Montgomery(2^261) g1a9 + infinity flag; off-curve ->
+// err.  Coordinates reduced mod p, (0,0) = identity (provider.rs:252-268).
 __global__ void k_parse_points(const uint8_t *__restrict__ in,
-                               g1a *__restrict__ pts, uint8_t *__restrict__ inf,
+                               g1a9 *__restrict__ pts, uint8_t *__restrict__ inf,
                                size_t n, uint32_t *__restrict__ err) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    fe4 x = to_mont<Fq>(fe_from_be(in + 64 * i));
-    fe4 y = to_mont<Fq>(fe_from_be(in + 64 * i + 32));
-    if (fe_is_zero(x) && fe_is_zero(y)) {
+    fe9 x = to_mont9(fe9_from_be(in + 64 * i));
+    fe9 y = to_mont9(fe9_from_be(in + 64 * i + 32));
+    if (fe9_is_zero_modp(x) && fe9_is_zero_modp(y)) {
         inf[i] = 1;
-        pts[i].x = x;
-        pts[i].y = y;
+        pts[i].x = fe9_zero();
+        pts[i].y = fe9_zero();
         return;
     }
-    g1a p{x, y};
+    g1a9 p{x, y};
     inf[i] = 0;
-    if (!g1a_on_curve(p)) atomicOr(err, 1u);
+    if (!g1a9_on_curve(p)) atomicOr(err, 1u);
     pts[i] = p;
 }
 
 // P_i = (start+i+1)*G directly in HBM; per-thread affine conversion.
-__global__ void k_gen_points(g1a *__restrict__ pts, uint8_t *__restrict__ inf,
+__global__ void k_gen_points(g1a9 *__restrict__ pts, uint8_t *__restrict__ inf,
                              size_t n, uint64_t start) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    uint64_t k = start + i + 1;
-    g1a g = g1_generator();
-    g1j acc = g1_inf();
+    u64 k[4] = {start + i + 1, 0, 0, 0};
+    g1a9 g = g1_generator9();
+    g1j9 acc = g1_inf9();
     for (int b = 63; b >= 0; b--) {
-        acc = g1_dbl(acc);
-        if ((k >> b) & 1) acc = g1_add_affine(acc, g);
+        acc = g1_dbl9(acc);
+        if ((k[0] >> b) & 1) acc = g1_add_affine9(acc, g);
     }
     // to affine (k >= 1 and k < r => never infinity)
-    fe4 zi = mont_inv<Fq>(acc.z);
-    fe4 zi2 = mont_sqr<Fq>(zi);
-    pts[i].x = mont_mul<Fq>(acc.x, zi2);
-    pts[i].y = mont_mul<Fq>(acc.y, mont_mul<Fq>(zi2, zi));
+    fe9 zi = mont_inv9(acc.z);
+    fe9 zi2 = mont_sqr9(zi);
+    pts[i].x = fe9_csub2p(mont_mul9(acc.x, zi2));
+    pts[i].y = fe9_csub2p(mont_mul9(acc.y, mont_mul9(zi2, zi)));
     inf[i] = 0;
 }
 
 // download points as BE affine bytes (for parity tests)
-__global__ void k_points_to_be(const g1a *__restrict__ pts,
+__global__ void k_points_to_be(const g1a9 *__restrict__ pts,
                                const uint8_t *__restrict__ inf,
                                uint8_t *__restrict__ out, size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -83,8 +86,8 @@ __global__ void k_points_to_be(const g1a *__restrict__ pts,
         for (int j = 0; j < 8; j++) ((u64 *)(out + 64 * i))[j] = 0;
         return;
     }
-    fe_to_be(out + 64 * i, from_mont<Fq>(pts[i].x));
-    fe_to_be(out + 64 * i + 32, from_mont<Fq>(pts[i].y));
+    fe9_to_be(out + 64 * i, from_mont9(pts[i].x));
+    fe9_to_be(out + 64 * i + 32, from_mont9(pts[i].y));
 }
 
 // 32-byte BE scalars -> canonical fe4 reduced mod r (from_be_bytes_mod_order)
@@ -122,7 +125,6 @@ __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total
         offsets[b] = (uint32_t)total;
         return;
     }
-    // lower bound of key b
     size_t lo = 0, hi = total;
     while (lo < hi) {
         size_t mid = (lo + hi) >> 1;
@@ -137,94 +139,86 @@ __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total
 // ---- bucket accumulation (the hot kernel) ----
 // one thread per bucket id; digit-0 buckets are skipped (never read later).
 __global__ void __launch_bounds__(256)
-k_bucket_acc(const g1a *__restrict__ pts, const uint32_t *__restrict__ vals,
-             const uint32_t *__restrict__ offsets, g1j *__restrict__ buckets) {
+k_bucket_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
+             const uint32_t *__restrict__ offsets, g1j9 *__restrict__ buckets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= MSM_NBUCKET_TOTAL) return;
     if ((b & 0xffff) == 0) return;  // digit 0
     uint32_t lo = offsets[b], hi = offsets[b + 1];
-    g1j acc = g1_inf();
+    g1j9 acc = g1_inf9();
     if (lo >= hi) {
         buckets[b] = acc;
         return;
     }
-    // software pipeline: issue the NEXT point's gather before the ~7k-inst
-    // mixed add so the dependent idx->point load chain (~2x L2/HBM latency)
-    // overlaps the VALU work of the current add.
-    g1a p = pts[vals[lo]];
+    // software pipeline: issue the NEXT point's gather before the long mixed
+    // add so the dependent idx->point load chain overlaps the VALU work.
+    g1a9 p = pts[vals[lo]];
     for (uint32_t t = lo; t < hi; t++) {
-        g1a cur = p;
-        uint32_t nxt = t + 1 < hi ? t + 1 : t;  // clamped: loads always issue
+        g1a9 cur = p;
+        uint32_t nxt = t + 1 < hi ? t + 1 : t;
         p = pts[vals[nxt]];
-        acc = g1_add_affine(acc, cur);
+        acc = g1_add_affine9(acc, cur);
     }
     buckets[b] = acc;
 }
 
 // ---- two-level running-sum reduction ----
-// level 1: per (window, 256-bucket segment): from the top digit down,
+// level 1: per (window, 32-bucket segment): from the top digit down,
 //   run  += B_d           (=> run  = sum of segment buckets)
 //   wsum += run           (=> wsum = sum (d - lo + 1) * B_d)
-__global__ void k_segment_reduce(const g1j *__restrict__ buckets,
-                                 g1j *__restrict__ seg_sum,
-                                 g1j *__restrict__ seg_wsum) {
-    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;  // [0, 16*256)
+__global__ void k_segment_reduce(const g1j9 *__restrict__ buckets,
+                                 g1j9 *__restrict__ seg_sum,
+                                 g1j9 *__restrict__ seg_wsum) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;  // [0, 16*2048)
     if (t >= MSM_NWIN * MSM_NSEG) return;
     uint32_t w = t / MSM_NSEG, seg = t % MSM_NSEG;
     uint32_t lo = seg * MSM_SEG;
-    g1j run = g1_inf(), wsum = g1_inf();
+    g1j9 run = g1_inf9(), wsum = g1_inf9();
     for (int32_t d = (int32_t)lo + MSM_SEG - 1; d >= (int32_t)lo; d--) {
         // digit-0 bucket is unused: skip the add but keep the wsum step so
-        // segment 0 carries the same (d - lo + 1) weights as every other
-        // segment (verified integer model: see DESIGN.md "Bucket reduction")
+        // segment 0 carries the same (d - lo + 1) weights
         if (d != 0)
-            run = g1_add(run, buckets[((uint32_t)w << 16) | (uint32_t)d]);
-        wsum = g1_add(wsum, run);
+            run = g1_add9(run, buckets[((uint32_t)w << 16) | (uint32_t)d]);
+        wsum = g1_add9(wsum, run);
     }
     seg_sum[t] = run;
     seg_wsum[t] = wsum;
 }
 
 // level 2: fully parallel weighted combine + LDS tree reduction.
-//   W_w = sum_j [ wsum_j + (j*SEG - 1) * sum_j ]   (j=0 term: -sum_0)
-// One thread per segment computes its weighted value (17-bit double-and-add
-// for the (j*SEG-1) scalar), then a 256-wide LDS tree sums the block; 8
-// blocks per window leave 16*8 partials for the tiny level-3 kernel.
-// (Integer-model verification: DESIGN.md "Bucket reduction".)
+//   W_w = sum_j [ wsum_j + (j*SEG - 1) * sum_j ]   (j=0 term: -sum_0),
+// then scaled by 2^(16w) (the doubling chains run SIMD-wide here).
 __global__ void __launch_bounds__(MSM_RED_BLOCK)
-k_weighted_reduce(const g1j *__restrict__ seg_sum,
-                  const g1j *__restrict__ seg_wsum,
-                  g1j *__restrict__ partials /* NWIN*NBLK_PER_WIN */) {
-    __shared__ g1j lds[MSM_RED_BLOCK];
+k_weighted_reduce(const g1j9 *__restrict__ seg_sum,
+                  const g1j9 *__restrict__ seg_wsum,
+                  g1j9 *__restrict__ partials /* NWIN*NBLK_PER_WIN */) {
+    __shared__ g1j9 lds[MSM_RED_BLOCK];
     uint32_t t = blockIdx.x * MSM_RED_BLOCK + threadIdx.x;
     uint32_t j = t % MSM_NSEG;
-    g1j ws = seg_wsum[t];
-    g1j ss = seg_sum[t];
-    g1j val;
+    g1j9 ws = seg_wsum[t];
+    g1j9 ss = seg_sum[t];
+    g1j9 val;
     if (j == 0) {
         // weight -1: subtract sum_0
-        if (!g1_is_inf(ss)) ss.y = mod_sub<Fq>(fe4{{0, 0, 0, 0}}, ss.y);
-        val = g1_add(ws, ss);
+        if (!g1_is_inf9(ss)) ss.y = neg9(ss.y);
+        val = g1_add9(ws, ss);
     } else {
         uint32_t weight = j * MSM_SEG - 1;  // <= 65535
-        g1j acc = g1_inf();
+        g1j9 acc = g1_inf9();
         for (int b = 16; b >= 0; b--) {
-            acc = g1_dbl(acc);
-            if ((weight >> b) & 1) acc = g1_add(acc, ss);
+            acc = g1_dbl9(acc);
+            if ((weight >> b) & 1) acc = g1_add9(acc, ss);
         }
-        val = g1_add(ws, acc);
+        val = g1_add9(ws, acc);
     }
-    // fold the window factor 2^(16w) in HERE: the doubling chains run
-    // SIMD-wide across all 32K threads (~0.4 ms wall) instead of in the
-    // 16-thread final combine (measured 3.5 ms single-wave).
     uint32_t w = t / MSM_NSEG;
-    for (uint32_t d = 0; d < (uint32_t)MSM_C * w; d++) val = g1_dbl(val);
+    for (uint32_t d = 0; d < (uint32_t)MSM_C * w; d++) val = g1_dbl9(val);
     lds[threadIdx.x] = val;
     __syncthreads();
     for (int s = MSM_RED_BLOCK / 2; s > 0; s >>= 1) {
         if (threadIdx.x < (uint32_t)s) {
-            g1j o = lds[threadIdx.x + s];
-            g1j m = g1_add(lds[threadIdx.x], o);
+            g1j9 o = lds[threadIdx.x + s];
+            g1j9 m = g1_add9(lds[threadIdx.x], o);
             lds[threadIdx.x] = m;
         }
         __syncthreads();
@@ -232,36 +226,35 @@ k_weighted_reduce(const g1j *__restrict__ seg_sum,
     if (threadIdx.x == 0) partials[blockIdx.x] = lds[0];
 }
 
-// level 3: 16 threads, 8 partials each -> per-window sums (already scaled
-// by 2^(16w) in k_weighted_reduce)
-__global__ void k_window_sum(const g1j *__restrict__ partials,
-                             g1j *__restrict__ windows) {
+// level 3: 16 threads, 8 partials each -> per-window sums (pre-scaled)
+__global__ void k_window_sum(const g1j9 *__restrict__ partials,
+                             g1j9 *__restrict__ windows) {
     uint32_t w = blockIdx.x * blockDim.x + threadIdx.x;
     if (w >= MSM_NWIN) return;
-    g1j acc = g1_inf();
+    g1j9 acc = g1_inf9();
     for (int b = 0; b < MSM_NBLK_PER_WIN; b++)
-        acc = g1_add(acc, partials[w * MSM_NBLK_PER_WIN + b]);
+        acc = g1_add9(acc, partials[w * MSM_NBLK_PER_WIN + b]);
     windows[w] = acc;
 }
 
-// ---- final Horner combine + output ----
+// ---- final combine + output ----
 // out_mode 0: 64-byte BE affine (infinity -> zeros)
 // out_mode 1: 96-byte BE Jacobian canonical X||Y||Z (Z=0 -> infinity)
-__global__ void k_final_combine(const g1j *__restrict__ windows,
+__global__ void k_final_combine(const g1j9 *__restrict__ windows,
                                 uint8_t *__restrict__ out, int out_mode) {
     if (blockIdx.x != 0 || threadIdx.x != 0) return;
-    // windows[] arrive pre-scaled by 2^(16w) (k_window_sum)
-    g1j acc = windows[MSM_NWIN - 1];
-    for (int w = MSM_NWIN - 2; w >= 0; w--) acc = g1_add(acc, windows[w]);
+    // windows[] arrive pre-scaled by 2^(16w) (k_weighted_reduce)
+    g1j9 acc = windows[MSM_NWIN - 1];
+    for (int w = MSM_NWIN - 2; w >= 0; w--) acc = g1_add9(acc, windows[w]);
     if (out_mode == 0) {
-        g1_to_affine_be(out, acc);
+        g1_to_affine_be9(out, acc);
     } else {
-        if (g1_is_inf(acc)) {
+        if (g1_is_inf9(acc)) {
             for (int j = 0; j < 12; j++) ((u64 *)out)[j] = 0;
         } else {
-            fe_to_be(out, from_mont<Fq>(acc.x));
-            fe_to_be(out + 32, from_mont<Fq>(acc.y));
-            fe_to_be(out + 64, from_mont<Fq>(acc.z));
+            fe9_to_be(out, from_mont9(acc.x));
+            fe9_to_be(out + 32, from_mont9(acc.y));
+            fe9_to_be(out + 64, from_mont9(acc.z));
         }
     }
 }
@@ -271,56 +264,60 @@ __global__ void k_final_combine(const g1j *__restrict__ windows,
 __global__ void k_g1_add_single(const uint8_t *in /* 128 B: p1||p2 */,
                                 uint8_t *out, uint32_t *err) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
-    fe4 x1 = to_mont<Fq>(fe_from_be(in));
-    fe4 y1 = to_mont<Fq>(fe_from_be(in + 32));
-    fe4 x2 = to_mont<Fq>(fe_from_be(in + 64));
-    fe4 y2 = to_mont<Fq>(fe_from_be(in + 96));
-    g1j acc = g1_inf();
-    if (!(fe_is_zero(x1) && fe_is_zero(y1))) {
-        g1a p{x1, y1};
-        if (!g1a_on_curve(p)) { atomicOr(err, 1u); return; }
-        acc = g1_add_affine(acc, p);
+    g1j9 acc = g1_inf9();
+#pragma unroll
+    for (int k = 0; k < 2; k++) {
+        fe9 x = to_mont9(fe9_from_be(in + 64 * k));
+        fe9 y = to_mont9(fe9_from_be(in + 64 * k + 32));
+        if (fe9_is_zero_modp(x) && fe9_is_zero_modp(y)) continue;
+        g1a9 p{x, y};
+        if (!g1a9_on_curve(p)) {
+            atomicOr(err, 1u);
+            return;
+        }
+        acc = g1_add_affine9(acc, p);
     }
-    if (!(fe_is_zero(x2) && fe_is_zero(y2))) {
-        g1a p{x2, y2};
-        if (!g1a_on_curve(p)) { atomicOr(err, 1u); return; }
-        acc = g1_add_affine(acc, p);
-    }
-    g1_to_affine_be(out, acc);
+    g1_to_affine_be9(out, acc);
 }
 
 __global__ void k_g1_mul_single(const uint8_t *in /* 96 B: point||scalar */,
                                 uint8_t *out, uint32_t *err) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
-    fe4 x = to_mont<Fq>(fe_from_be(in));
-    fe4 y = to_mont<Fq>(fe_from_be(in + 32));
-    if (fe_is_zero(x) && fe_is_zero(y)) {
+    fe9 x = to_mont9(fe9_from_be(in));
+    fe9 y = to_mont9(fe9_from_be(in + 32));
+    if (fe9_is_zero_modp(x) && fe9_is_zero_modp(y)) {
         for (int j = 0; j < 8; j++) ((u64 *)out)[j] = 0;
         return;
     }
-    g1a p{x, y};
-    if (!g1a_on_curve(p)) { atomicOr(err, 1u); return; }
+    g1a9 p{x, y};
+    if (!g1a9_on_curve(p)) {
+        atomicOr(err, 1u);
+        return;
+    }
     fe4 k = from_mont<Fr>(to_mont<Fr>(fe_from_be(in + 64)));
     if (fe_is_zero(k)) {
         for (int j = 0; j < 8; j++) ((u64 *)out)[j] = 0;
         return;
     }
-    g1_to_affine_be(out, g1_scalar_mul(p, k));
+    g1_to_affine_be9(out, g1_scalar_mul9(p, k.v));
 }
 
 // combine count Jacobian partials (96-B BE canonical each) -> affine
 __global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
                              uint8_t *__restrict__ out) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
-    g1j acc = g1_inf();
+    g1j9 acc = g1_inf9();
     for (size_t i = 0; i < count; i++) {
-        g1j p;
-        p.x = to_mont<Fq>(fe_from_be(in + 96 * i));
-        p.y = to_mont<Fq>(fe_from_be(in + 96 * i + 32));
-        p.z = to_mont<Fq>(fe_from_be(in + 96 * i + 64));
-        acc = g1_add(acc, p);
+        g1j9 p;
+        p.x = to_mont9(fe9_from_be(in + 96 * i));
+        p.y = to_mont9(fe9_from_be(in + 96 * i + 32));
+        p.z = to_mont9(fe9_from_be(in + 96 * i + 64));
+        // canonical zero Z parses to a 0 (mod p) Montgomery value: exact-zero
+        // it so g1_is_inf9's raw check holds
+        if (fe9_is_zero_modp(p.z)) p.z = fe9_zero();
+        acc = g1_add9(acc, p);
     }
-    g1_to_affine_be(out, acc);
+    g1_to_affine_be9(out, acc);
 }
 
 }  // namespace em

@@ -12,11 +12,11 @@ using u32=uint32_t; using u64=uint64_t;
 // independent mads (ILP): 4 chains
 __global__ void k_mad64_ilp(u64* out, u64 seed) {
     u64 a=seed+threadIdx.x, b=seed^threadIdx.x, c=seed|1, d=seed+7;
-    u32 x=(u32)(seed>>3)|1, y=(u32)(seed>>5)|3;
+    u32 y=(u32)(seed>>5)|3;
     for (int i=0;i<ITER;i++){
 #pragma unroll
         for(int u=0;u<UNROLL/4;u++){
-            a = (u64)x*y + a; b = (u64)y*x + b; c = (u64)(x^3)*y + c; d = (u64)x*(y^5) + d;
+            a = (u64)(u32)a*y + a; b = (u64)(u32)b*y + b; c = (u64)(u32)c*y + c; d = (u64)(u32)d*y + d;
         }
     }
     out[blockIdx.x*blockDim.x+threadIdx.x] = a+b+c+d;
@@ -24,10 +24,10 @@ __global__ void k_mad64_ilp(u64* out, u64 seed) {
 // fully dependent mad chain
 __global__ void k_mad64_dep(u64* out, u64 seed) {
     u64 a=seed+threadIdx.x;
-    u32 x=(u32)(seed>>3)|1, y=(u32)(seed>>5)|3;
+    u32 y=(u32)(seed>>5)|3;
     for (int i=0;i<ITER;i++){
 #pragma unroll
-        for(int u=0;u<UNROLL;u++) a = (u64)x*y + a;
+        for(int u=0;u<UNROLL;u++) a = (u64)(u32)a*y + a;
     }
     out[blockIdx.x*blockDim.x+threadIdx.x] = a;
 }
@@ -44,19 +44,19 @@ __global__ void k_mul32_ilp(u64* out, u64 seed) {
 }
 // 64-bit adds, independent
 __global__ void k_add64_ilp(u64* out, u64 seed) {
-    u64 a=seed+threadIdx.x, b=seed^threadIdx.x, c=seed|1, d=seed+7, e=seed<<1;
+    u64 a=seed+threadIdx.x, b=seed^threadIdx.x, c=seed|1, d=seed+7;
     for (int i=0;i<ITER;i++){
 #pragma unroll
-        for(int u=0;u<UNROLL/4;u++){ a+=e; b+=e; c+=e; d+=e; }
+        for(int u=0;u<UNROLL/4;u++){ a+=b; b+=c; c+=d; d+=a; }
     }
     out[blockIdx.x*blockDim.x+threadIdx.x] = a+b+c+d;
 }
 // 32-bit adds
 __global__ void k_add32_ilp(u64* out, u64 seed) {
-    u32 a=(u32)seed+threadIdx.x, b=(u32)seed^threadIdx.x, c=(u32)seed|1, d=(u32)seed+7, e=(u32)seed|3;
+    u32 a=(u32)seed+threadIdx.x, b=(u32)seed^threadIdx.x, c=(u32)seed|1, d=(u32)seed+7;
     for (int i=0;i<ITER;i++){
 #pragma unroll
-        for(int u=0;u<UNROLL/4;u++){ a+=e; b+=e; c+=e; d+=e; }
+        for(int u=0;u<UNROLL/4;u++){ a+=b; b+=c; c+=d; d+=a; }
     }
     out[blockIdx.x*blockDim.x+threadIdx.x] = (u64)a+b+c+d;
 }
