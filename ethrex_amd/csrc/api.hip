@@ -80,11 +80,12 @@ struct em_msm_plan {
     uint8_t *d_inf = nullptr;
     fe4 *d_scalars = nullptr;
     uint8_t *d_scratch = nullptr;     // 64n bytes: point/scalar byte staging
-    uint32_t *d_vals = nullptr;       // up to 16n grouped point indices
-    uint32_t *d_hist = nullptr;       // NBUCKET_TOTAL + 1
-    uint32_t *d_cursor = nullptr;     // NBUCKET_TOTAL (scatter cursors)
-    void *d_scan_tmp = nullptr;
-    size_t scan_tmp_bytes = 0;
+    uint32_t *d_keys = nullptr;       // 16n
+    uint32_t *d_vals = nullptr;
+    uint32_t *d_keys_out = nullptr;
+    uint32_t *d_vals_out = nullptr;
+    void *d_sort_tmp = nullptr;
+    size_t sort_tmp_bytes = 0;
     uint32_t *d_offsets = nullptr;    // NBUCKET_TOTAL + 1
     g1j9 *d_buckets = nullptr;         // NBUCKET_TOTAL
     g1j9 *d_seg_sum = nullptr;         // NWIN*NSEG
@@ -114,9 +115,10 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     mal((void **)&p->d_inf, n);
     mal((void **)&p->d_scalars, n * sizeof(fe4));
     mal((void **)&p->d_scratch, n * 64);
+    mal((void **)&p->d_keys, total * 4);
     mal((void **)&p->d_vals, total * 4);
-    mal((void **)&p->d_hist, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
-    mal((void **)&p->d_cursor, (size_t)MSM_NBUCKET_TOTAL * 4);
+    mal((void **)&p->d_keys_out, total * 4);
+    mal((void **)&p->d_vals_out, total * 4);
     mal((void **)&p->d_offsets, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
     mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1j9));
     mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1j9));
@@ -126,11 +128,11 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     mal((void **)&p->d_out, 96);
     mal((void **)&p->d_err, 4);
     if (e == hipSuccess) {
-        // rocPRIM exclusive-scan temp-storage size query
-        e = rocprim::exclusive_scan(nullptr, p->scan_tmp_bytes, p->d_hist,
-                                    p->d_offsets,
-                                    (uint32_t)0, (size_t)MSM_NBUCKET_TOTAL + 1);
-        if (e == hipSuccess) e = hipMalloc(&p->d_scan_tmp, p->scan_tmp_bytes);
+        // rocPRIM temp-storage size query
+        e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
+                                      p->d_keys_out, p->d_vals, p->d_vals_out,
+                                      total, 0, 20);
+        if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
     }
     for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
     if (e != hipSuccess) {
@@ -147,10 +149,11 @@ extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
     hipFree(p->d_inf);
     hipFree(p->d_scalars);
     hipFree(p->d_scratch);
+    hipFree(p->d_keys);
     hipFree(p->d_vals);
-    hipFree(p->d_hist);
-    hipFree(p->d_cursor);
-    hipFree(p->d_scan_tmp);
+    hipFree(p->d_keys_out);
+    hipFree(p->d_vals_out);
+    hipFree(p->d_sort_tmp);
     hipFree(p->d_offsets);
     hipFree(p->d_buckets);
     hipFree(p->d_seg_sum);
@@ -211,25 +214,25 @@ static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
         g_last_err = "msm_run: points/scalars not uploaded";
         return EM_ERR_INPUT;
     }
+    size_t total = p->n * MSM_NWIN;
     HIP_TRY(hipEventRecord(p->ev[0], 0));
-    // counting-sort grouping: histogram -> exclusive scan -> scatter
-    HIP_TRY(hipMemsetAsync(p->d_hist, 0, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4, 0));
-    hipLaunchKernelGGL(k_histogram, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       p->d_scalars, p->d_inf, p->d_hist, p->n);
-    size_t tmp = p->scan_tmp_bytes;
-    hipError_t e = rocprim::exclusive_scan(p->d_scan_tmp, tmp, p->d_hist,
-                                           p->d_offsets, (uint32_t)0,
-                                           (size_t)MSM_NBUCKET_TOTAL + 1);
-    if (e != hipSuccess) return hip_fail(e, "exclusive_scan");
-    HIP_TRY(hipMemcpyAsync(p->d_cursor, p->d_offsets,
-                           (size_t)MSM_NBUCKET_TOTAL * 4,
-                           hipMemcpyDeviceToDevice, 0));
-    hipLaunchKernelGGL(k_scatter, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       p->d_scalars, p->d_inf, p->d_cursor, p->d_vals, p->n);
+    // digits
+    hipLaunchKernelGGL(k_digits, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_scalars, p->d_inf, p->d_keys, p->d_vals, p->n);
+    // sort on 20 key bits
+    size_t tmp = p->sort_tmp_bytes;
+    hipError_t e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
+                                             p->d_keys_out, p->d_vals,
+                                             p->d_vals_out, total, 0, 20);
+    if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs");
+    // offsets
+    hipLaunchKernelGGL(k_offsets,
+                       dim3(blocks_for((size_t)MSM_NBUCKET_TOTAL + 1, 256)),
+                       dim3(256), 0, 0, p->d_keys_out, total, p->d_offsets);
     HIP_TRY(hipEventRecord(p->ev[1], 0));
     // bucket accumulation (hot)
     hipLaunchKernelGGL(k_bucket_acc, dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)),
-                       dim3(256), 0, 0, p->d_pts, p->d_vals, p->d_offsets,
+                       dim3(256), 0, 0, p->d_pts, p->d_vals_out, p->d_offsets,
                        p->d_buckets);
     HIP_TRY(hipEventRecord(p->ev[2], 0));
     // reductions

@@ -98,42 +98,42 @@ __global__ void k_parse_scalars(const uint8_t *__restrict__ in,
     out[i] = from_mont<Fr>(to_mont<Fr>(fe_from_be(in + 32 * i)));
 }
 
-// ---- counting-sort bucket grouping (replaces a full radix sort) ----
-// c=16 => digit w is the w-th u16 of the scalar.  Zero digits (and identity
-// points) contribute nothing and get no slot.  Bucket sums are independent
-// of intra-bucket order (EC addition is commutative/associative and every
-// exposed result is canonical-affine), so atomic slot assignment is fine.
-
-__global__ void k_histogram(const fe4 *__restrict__ scalars,
-                            const uint8_t *__restrict__ inf,
-                            uint32_t *__restrict__ hist, size_t n) {
+// ---- digit extraction: c=16 => digit w is the w-th u16 of the scalar ----
+__global__ void k_digits(const fe4 *__restrict__ scalars,
+                         const uint8_t *__restrict__ inf,
+                         uint32_t *__restrict__ keys, uint32_t *__restrict__ vals,
+                         size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    if (inf[i]) return;
     fe4 k = scalars[i];
+    bool skip = inf[i];
 #pragma unroll
     for (int w = 0; w < MSM_NWIN; w++) {
         uint32_t d = (uint32_t)((k.v[w >> 2] >> ((w & 3) * 16)) & 0xffff);
-        if (d) atomicAdd(&hist[((uint32_t)w << 16) | d], 1u);
+        if (skip) d = 0;  // identity points contribute nothing: park in bucket 0
+        keys[(size_t)w * n + i] = ((uint32_t)w << 16) | d;
+        vals[(size_t)w * n + i] = (uint32_t)i;
     }
 }
 
-__global__ void k_scatter(const fe4 *__restrict__ scalars,
-                          const uint8_t *__restrict__ inf,
-                          uint32_t *__restrict__ cursor,
-                          uint32_t *__restrict__ vals, size_t n) {
-    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n) return;
-    if (inf[i]) return;
-    fe4 k = scalars[i];
-#pragma unroll
-    for (int w = 0; w < MSM_NWIN; w++) {
-        uint32_t d = (uint32_t)((k.v[w >> 2] >> ((w & 3) * 16)) & 0xffff);
-        if (d) {
-            uint32_t slot = atomicAdd(&cursor[((uint32_t)w << 16) | d], 1u);
-            vals[slot] = (uint32_t)i;
-        }
+// ---- bucket segment offsets: lower_bound of each bucket id ----
+__global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total,
+                          uint32_t *__restrict__ offsets) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b > MSM_NBUCKET_TOTAL) return;
+    if (b == MSM_NBUCKET_TOTAL) {
+        offsets[b] = (uint32_t)total;
+        return;
     }
+    size_t lo = 0, hi = total;
+    while (lo < hi) {
+        size_t mid = (lo + hi) >> 1;
+        if (sorted_keys[mid] < b)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    offsets[b] = (uint32_t)lo;
 }
 
 // ---- bucket accumulation (the hot kernel) ----
