@@ -131,7 +131,7 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
         // rocPRIM temp-storage size query
         e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
                                       p->d_keys_out, p->d_vals, p->d_vals_out,
-                                      total, 0, 20);
+                                      total, 0, MSM_SORT_BITS);
         if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
     }
     for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
@@ -223,7 +223,7 @@ static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
     size_t tmp = p->sort_tmp_bytes;
     hipError_t e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
                                              p->d_keys_out, p->d_vals,
-                                             p->d_vals_out, total, 0, 20);
+                                             p->d_vals_out, total, 0, MSM_SORT_BITS);
     if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs");
     // offsets
     hipLaunchKernelGGL(k_offsets,

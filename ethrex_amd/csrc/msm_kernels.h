@@ -25,13 +25,24 @@
 
 namespace em {
 
-constexpr int MSM_C = 16;                      // window bits
-constexpr int MSM_NWIN = 16;                   // ceil(254/16)
-constexpr uint32_t MSM_NBUCKET_TOTAL = (uint32_t)MSM_NWIN << MSM_C;  // 1M ids
+constexpr int MSM_C = 20;                      // window bits
+constexpr int MSM_NWIN = 13;                   // ceil(254/20)
+constexpr uint32_t MSM_DMASK = (1u << MSM_C) - 1;
+constexpr uint32_t MSM_NBUCKET_TOTAL = (uint32_t)MSM_NWIN << MSM_C;  // 13.6M
+constexpr int MSM_SORT_BITS = 24;              // 20 digit + 4 window bits
 constexpr int MSM_SEG = 32;                    // buckets per reduction segment
-constexpr int MSM_NSEG = (1 << MSM_C) / MSM_SEG;  // 2048 segments per window
+constexpr int MSM_NSEG = (1 << MSM_C) / MSM_SEG;  // 32768 segments per window
 constexpr int MSM_RED_BLOCK = 256;             // threads per level-2 block
-constexpr int MSM_NBLK_PER_WIN = MSM_NSEG / MSM_RED_BLOCK;  // 8
+constexpr int MSM_NBLK_PER_WIN = MSM_NSEG / MSM_RED_BLOCK;  // 128
+
+// digit w = bits [C*w, C*w+C) of the canonical scalar (spans u64 limbs)
+__device__ __forceinline__ uint32_t msm_digit(const fe4 &k, int w) {
+    int bit = MSM_C * w;
+    int limb = bit >> 6, off = bit & 63;
+    uint64_t d = k.v[limb] >> off;
+    if (off > 64 - MSM_C && limb < 3) d |= k.v[limb + 1] << (64 - off);
+    return (uint32_t)d & MSM_DMASK;
+}
 
 // ---- input parsing ----
 
@@ -98,7 +109,7 @@ __global__ void k_parse_scalars(const uint8_t *__restrict__ in,
     out[i] = from_mont<Fr>(to_mont<Fr>(fe_from_be(in + 32 * i)));
 }
 
-// ---- digit extraction: c=16 => digit w is the w-th u16 of the scalar ----
+// ---- digit extraction ----
 __global__ void k_digits(const fe4 *__restrict__ scalars,
                          const uint8_t *__restrict__ inf,
                          uint32_t *__restrict__ keys, uint32_t *__restrict__ vals,
@@ -109,9 +120,9 @@ __global__ void k_digits(const fe4 *__restrict__ scalars,
     bool skip = inf[i];
 #pragma unroll
     for (int w = 0; w < MSM_NWIN; w++) {
-        uint32_t d = (uint32_t)((k.v[w >> 2] >> ((w & 3) * 16)) & 0xffff);
+        uint32_t d = msm_digit(k, w);
         if (skip) d = 0;  // identity points contribute nothing: park in bucket 0
-        keys[(size_t)w * n + i] = ((uint32_t)w << 16) | d;
+        keys[(size_t)w * n + i] = ((uint32_t)w << MSM_C) | d;
         vals[(size_t)w * n + i] = (uint32_t)i;
     }
 }
@@ -143,7 +154,7 @@ k_bucket_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
              const uint32_t *__restrict__ offsets, g1j9 *__restrict__ buckets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= MSM_NBUCKET_TOTAL) return;
-    if ((b & 0xffff) == 0) return;  // digit 0
+    if ((b & MSM_DMASK) == 0) return;  // digit 0
     uint32_t lo = offsets[b], hi = offsets[b + 1];
     g1j9 acc = g1_inf9();
     if (lo >= hi) {
@@ -178,7 +189,7 @@ __global__ void k_segment_reduce(const g1j9 *__restrict__ buckets,
         // digit-0 bucket is unused: skip the add but keep the wsum step so
         // segment 0 carries the same (d - lo + 1) weights
         if (d != 0)
-            run = g1_add9(run, buckets[((uint32_t)w << 16) | (uint32_t)d]);
+            run = g1_add9(run, buckets[((uint32_t)w << MSM_C) | (uint32_t)d]);
         wsum = g1_add9(wsum, run);
     }
     seg_sum[t] = run;
@@ -203,16 +214,14 @@ k_weighted_reduce(const g1j9 *__restrict__ seg_sum,
         if (!g1_is_inf9(ss)) ss.y = neg9(ss.y);
         val = g1_add9(ws, ss);
     } else {
-        uint32_t weight = j * MSM_SEG - 1;  // <= 65535
+        uint32_t weight = j * MSM_SEG - 1;  // < 2^C
         g1j9 acc = g1_inf9();
-        for (int b = 16; b >= 0; b--) {
+        for (int b = MSM_C; b >= 0; b--) {
             acc = g1_dbl9(acc);
             if ((weight >> b) & 1) acc = g1_add9(acc, ss);
         }
         val = g1_add9(ws, acc);
     }
-    uint32_t w = t / MSM_NSEG;
-    for (uint32_t d = 0; d < (uint32_t)MSM_C * w; d++) val = g1_dbl9(val);
     lds[threadIdx.x] = val;
     __syncthreads();
     for (int s = MSM_RED_BLOCK / 2; s > 0; s >>= 1) {
@@ -223,7 +232,14 @@ k_weighted_reduce(const g1j9 *__restrict__ seg_sum,
         }
         __syncthreads();
     }
-    if (threadIdx.x == 0) partials[blockIdx.x] = lds[0];
+    if (threadIdx.x == 0) {
+        // fold the window factor 2^(C*w): one doubling chain per BLOCK
+        // (NWIN*NBLK_PER_WIN chains run concurrently; wall = longest chain)
+        uint32_t w = t / MSM_NSEG;
+        g1j9 acc = lds[0];
+        for (uint32_t d = 0; d < (uint32_t)MSM_C * w; d++) acc = g1_dbl9(acc);
+        partials[blockIdx.x] = acc;
+    }
 }
 
 // level 3: 16 threads, 8 partials each -> per-window sums (pre-scaled)
