@@ -26,8 +26,8 @@ import time
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
-MSM_WINDOWS = 13          # c = 20 (ethrex_amd/csrc/msm_kernels.h)
-MSM_C = 20
+MSM_WINDOWS = 16          # c = 16 (ethrex_amd/csrc/msm_kernels.h)
+MSM_C = 16
 
 
 def point_adds(n_total):
@@ -154,7 +154,7 @@ def main():
     # writes (DESIGN.md "Measurement").  The kernel is VALU-bound (big-int
     # Montgomery mul), so the HBM fraction is expectedly far below 1; VALU
     # evidence lives in profiles/ (rocprofv3 PMC).
-    alg_bytes = MSM_WINDOWS * shard * 76 + (MSM_WINDOWS << 20) * 108
+    alg_bytes = MSM_WINDOWS * shard * 76 + (MSM_WINDOWS << MSM_C) * 108
     avg_bucket_ms = sum(bucket_ms) / len(bucket_ms)
     hbm_peak = 8.0e12
     roofline = {

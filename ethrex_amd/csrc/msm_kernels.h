@@ -25,15 +25,15 @@
 
 namespace em {
 
-constexpr int MSM_C = 20;                      // window bits
-constexpr int MSM_NWIN = 13;                   // ceil(254/20)
+constexpr int MSM_C = 16;                      // window bits
+constexpr int MSM_NWIN = 16;                   // ceil(254/16)
 constexpr uint32_t MSM_DMASK = (1u << MSM_C) - 1;
-constexpr uint32_t MSM_NBUCKET_TOTAL = (uint32_t)MSM_NWIN << MSM_C;  // 13.6M
-constexpr int MSM_SORT_BITS = 24;              // 20 digit + 4 window bits
+constexpr uint32_t MSM_NBUCKET_TOTAL = (uint32_t)MSM_NWIN << MSM_C;  // 1M ids
+constexpr int MSM_SORT_BITS = 20;              // 16 digit + 4 window bits
 constexpr int MSM_SEG = 32;                    // buckets per reduction segment
-constexpr int MSM_NSEG = (1 << MSM_C) / MSM_SEG;  // 32768 segments per window
+constexpr int MSM_NSEG = (1 << MSM_C) / MSM_SEG;  // 2048 segments per window
 constexpr int MSM_RED_BLOCK = 256;             // threads per level-2 block
-constexpr int MSM_NBLK_PER_WIN = MSM_NSEG / MSM_RED_BLOCK;  // 128
+constexpr int MSM_NBLK_PER_WIN = MSM_NSEG / MSM_RED_BLOCK;  // 8
 
 // digit w = bits [C*w, C*w+C) of the canonical scalar (spans u64 limbs)
 __device__ __forceinline__ uint32_t msm_digit(const fe4 &k, int w) {
