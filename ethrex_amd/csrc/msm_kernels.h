@@ -80,10 +80,7 @@ __global__ void k_gen_points(g1a9 *__restrict__ pts, uint8_t *__restrict__ inf,
         if ((k[0] >> b) & 1) acc = g1_add_affine9(acc, g);
     }
     // to affine (k >= 1 and k < r => never infinity)
-    fe9 zi = mont_inv9(acc.z);
-    fe9 zi2 = mont_sqr9(zi);
-    pts[i].x = fe9_csub2p(mont_mul9(acc.x, zi2));
-    pts[i].y = fe9_csub2p(mont_mul9(acc.y, mont_mul9(zi2, zi)));
+    pts[i] = g1_to_affine9(acc);
     inf[i] = 0;
 }
 
@@ -265,12 +262,15 @@ __global__ void k_final_combine(const g1j9 *__restrict__ windows,
     if (out_mode == 0) {
         g1_to_affine_be9(out, acc);
     } else {
+        // 96-B Jacobian exchange payload (ABI): convert XYZZ -> Jacobian
         if (g1_is_inf9(acc)) {
             for (int j = 0; j < 12; j++) ((u64 *)out)[j] = 0;
         } else {
-            fe9_to_be(out, from_mont9(acc.x));
-            fe9_to_be(out + 32, from_mont9(acc.y));
-            fe9_to_be(out + 64, from_mont9(acc.z));
+            fe9 X, Y, Z;
+            g1_xyzz_to_jacobian9(X, Y, Z, acc);
+            fe9_to_be(out, from_mont9(X));
+            fe9_to_be(out + 32, from_mont9(Y));
+            fe9_to_be(out + 64, from_mont9(Z));
         }
     }
 }
@@ -324,14 +324,11 @@ __global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     g1j9 acc = g1_inf9();
     for (size_t i = 0; i < count; i++) {
-        g1j9 p;
-        p.x = to_mont9(fe9_from_be(in + 96 * i));
-        p.y = to_mont9(fe9_from_be(in + 96 * i + 32));
-        p.z = to_mont9(fe9_from_be(in + 96 * i + 64));
-        // canonical zero Z parses to a 0 (mod p) Montgomery value: exact-zero
-        // it so g1_is_inf9's raw check holds
-        if (fe9_is_zero_modp(p.z)) p.z = fe9_zero();
-        acc = g1_add9(acc, p);
+        fe9 X = to_mont9(fe9_from_be(in + 96 * i));
+        fe9 Y = to_mont9(fe9_from_be(in + 96 * i + 32));
+        fe9 Z = to_mont9(fe9_from_be(in + 96 * i + 64));
+        if (fe9_is_zero_modp(Z)) continue;  // Z=0 encodes infinity
+        acc = g1_add9(acc, g1_jacobian_to_xyzz9(X, Y, Z));
     }
     g1_to_affine_be9(out, acc);
 }
