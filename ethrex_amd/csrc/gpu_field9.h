@@ -31,6 +31,26 @@ struct fe9 {
     u32 v[9];
 };
 
+// field traits: same 9x29 scheme for Fq (G1) and Fr (NTT)
+struct Fq9T {
+    static constexpr const u32 (&P)[9] = bn254::FQ9_P;
+    static constexpr const u32 (&TWOP)[9] = bn254::FQ9_2P;
+    static constexpr const u32 (&C2P)[9] = bn254::FQ9_C2P;
+    static constexpr const u32 (&C4P)[9] = bn254::FQ9_C4P;
+    static constexpr const u32 (&R2)[9] = bn254::FQ9_R2;
+    static constexpr const u32 (&ONE)[9] = bn254::FQ9_ONE;
+    static constexpr u32 N0INV = bn254::FQ9_N0INV;
+};
+struct Fr9T {
+    static constexpr const u32 (&P)[9] = bn254::FR9_P;
+    static constexpr const u32 (&TWOP)[9] = bn254::FR9_2P;
+    static constexpr const u32 (&C2P)[9] = bn254::FR9_C2P;
+    static constexpr const u32 (&C4P)[9] = bn254::FR9_C4P;
+    static constexpr const u32 (&R2)[9] = bn254::FR9_R2;
+    static constexpr const u32 (&ONE)[9] = bn254::FR9_ONE;
+    static constexpr u32 N0INV = bn254::FR9_N0INV;
+};
+
 __device__ __host__ __forceinline__ fe9 fe9_zero() {
     return fe9{{0, 0, 0, 0, 0, 0, 0, 0, 0}};
 }
@@ -57,12 +77,13 @@ __device__ __host__ __forceinline__ bool fe9_is_zero_raw(const fe9 &a) {
 }
 
 // x ≡ 0 mod p for x norm2p (< 2p): x == 0 or x == p
+template <typename T = Fq9T>
 __device__ __forceinline__ bool fe9_is_zero_modp(const fe9 &a) {
     u32 z = 0, e = 0;
 #pragma unroll
     for (int i = 0; i < 9; i++) {
         z |= a.v[i];
-        e |= a.v[i] ^ bn254::FQ9_P[i];
+        e |= a.v[i] ^ T::P[i];
     }
     return z == 0 || e == 0;
 }
@@ -81,12 +102,13 @@ __device__ __host__ __forceinline__ fe9 fe9_norm(const fe9 &a) {
 }
 
 // conditional subtract 2p: input norm limbs, value < 4p  =>  value < 2p
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 fe9_csub2p(const fe9 &a) {
     fe9 s;
     u32 bor = 0;
 #pragma unroll
     for (int i = 0; i < 9; i++) {
-        u32 t = a.v[i] - bn254::FQ9_2P[i] - bor;
+        u32 t = a.v[i] - T::TWOP[i] - bor;
         bor = (t >> 31) & 1;           // limbs < 2^29 so sign bit = borrow
         s.v[i] = t & bn254::FQ9_MASK;
     }
@@ -97,12 +119,13 @@ __device__ __host__ __forceinline__ fe9 fe9_csub2p(const fe9 &a) {
 }
 
 // conditional subtract p: input norm limbs, value < 2p  =>  canonical < p
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 fe9_csubp(const fe9 &a) {
     fe9 s;
     u32 bor = 0;
 #pragma unroll
     for (int i = 0; i < 9; i++) {
-        u32 t = a.v[i] - bn254::FQ9_P[i] - bor;
+        u32 t = a.v[i] - T::P[i] - bor;
         bor = (t >> 31) & 1;
         s.v[i] = t & bn254::FQ9_MASK;
     }
@@ -121,35 +144,40 @@ __device__ __host__ __forceinline__ fe9 add9(const fe9 &a, const fe9 &b) {
 }
 
 // normalizing add: out norm2p (inputs: limb sum < 2^32, value sum < 4p)
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 add9_n(const fe9 &a, const fe9 &b) {
-    return fe9_csub2p(fe9_norm(add9(a, b)));
+    return fe9_csub2p<T>(fe9_norm(add9(a, b)));
 }
 
 // a - b + 2p; b a mul output (< 1.5p, norm limbs); a norm limbs, value < 2p
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 subm9(const fe9 &a, const fe9 &b) {
     fe9 t;
 #pragma unroll
-    for (int i = 0; i < 9; i++) t.v[i] = a.v[i] + bn254::FQ9_C2P[i] - b.v[i];
-    return fe9_csub2p(fe9_norm(t));    // < 4p -> < 2p
+    for (int i = 0; i < 9; i++) t.v[i] = a.v[i] + T::C2P[i] - b.v[i];
+    return fe9_csub2p<T>(fe9_norm(t));    // < 4p -> < 2p
 }
 
 // a - b + 4p; b norm2p (< 2p); a norm limbs, value < 2p  => out < 6p -> 2 csubs
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 subn9(const fe9 &a, const fe9 &b) {
     fe9 t;
 #pragma unroll
-    for (int i = 0; i < 9; i++) t.v[i] = a.v[i] + bn254::FQ9_C4P[i] - b.v[i];
-    return fe9_csub2p(fe9_csub2p(fe9_norm(t)));
+    for (int i = 0; i < 9; i++) t.v[i] = a.v[i] + T::C4P[i] - b.v[i];
+    return fe9_csub2p<T>(fe9_csub2p<T>(fe9_norm(t)));
 }
 
 // -y mod p for y norm2p: 4p - y -> norm2p
+template <typename T = Fq9T>
 __device__ __forceinline__ fe9 neg9(const fe9 &y) {
     fe9 t;
 #pragma unroll
-    for (int i = 0; i < 9; i++) t.v[i] = bn254::FQ9_C4P[i] - y.v[i];
-    return fe9_csub2p(fe9_csub2p(fe9_norm(t)));
+    for (int i = 0; i < 9; i++) t.v[i] = T::C4P[i] - y.v[i];
+    return fe9_csub2p<T>(fe9_csub2p<T>(fe9_norm(t)));
 }
 
 // ---- Montgomery multiplication: column SOS, radix 2^29 ----
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 mont_mul9(const fe9 &A, const fe9 &B) {
     u64 t[17];
 #pragma unroll
@@ -161,9 +189,9 @@ __device__ __host__ __forceinline__ fe9 mont_mul9(const fe9 &A, const fe9 &B) {
     }
 #pragma unroll
     for (int k = 0; k < 9; k++) {
-        u32 m = ((u32)t[k] * bn254::FQ9_N0INV) & bn254::FQ9_MASK;
+        u32 m = ((u32)t[k] * T::N0INV) & bn254::FQ9_MASK;
 #pragma unroll
-        for (int j = 0; j < 9; j++) t[k + j] += (u64)m * bn254::FQ9_P[j];
+        for (int j = 0; j < 9; j++) t[k + j] += (u64)m * T::P[j];
         t[k + 1] += t[k] >> 29;        // t[k] ≡ 0 mod 2^29 now
     }
     fe9 r;
@@ -178,8 +206,9 @@ __device__ __host__ __forceinline__ fe9 mont_mul9(const fe9 &A, const fe9 &B) {
     return r;
 }
 
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 mont_sqr9(const fe9 &a) {
-    return mont_mul9(a, a);
+    return mont_mul9<T>(a, a);
 }
 
 // ---- conversions fe4 (4x64 canonical) <-> fe9 ----
@@ -212,22 +241,25 @@ __device__ __host__ __forceinline__ void fe9_to_u64x4(u64 w[4], const fe9 &a) {
 }
 
 // to Montgomery(2^261): x any value < 2^256 (raw 29-limbs) -> norm2p
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 to_mont9(const fe9 &x) {
-    return mont_mul9(x, fe9_load(bn254::FQ9_R2));
+    return mont_mul9<T>(x, fe9_load(T::R2));
 }
 
 // from Montgomery: norm2p -> canonical (< p, norm limbs)
+template <typename T = Fq9T>
 __device__ __host__ __forceinline__ fe9 from_mont9(const fe9 &x) {
     fe9 one{{1, 0, 0, 0, 0, 0, 0, 0, 0}};
-    return fe9_csubp(mont_mul9(x, one));
+    return fe9_csubp<T>(mont_mul9<T>(x, one));
 }
 
 // x^e (Montgomery in/out), e canonical 4x64
+template <typename T = Fq9T>
 __device__ __forceinline__ fe9 mont_pow9(const fe9 &x, const u64 e[4]) {
-    fe9 acc = fe9_load(bn254::FQ9_ONE);
+    fe9 acc = fe9_load(T::ONE);
     for (int i = 255; i >= 0; i--) {
-        acc = mont_sqr9(acc);
-        if ((e[i >> 6] >> (i & 63)) & 1) acc = mont_mul9(acc, x);
+        acc = mont_sqr9<T>(acc);
+        if ((e[i >> 6] >> (i & 63)) & 1) acc = mont_mul9<T>(acc, x);
     }
     return acc;
 }
@@ -239,9 +271,19 @@ __device__ __forceinline__ fe9 mont_inv9(const fe9 &x) {
     return mont_pow9(x, e);
 }
 
+// raw lexicographic compare of norm-limb values: a >= b ?
+__device__ __forceinline__ bool fe9_geq_raw(const fe9 &a, const u32 (&b)[9]) {
+#pragma unroll
+    for (int i = 8; i >= 0; i--) {
+        if (a.v[i] != b[i]) return a.v[i] > b[i];
+    }
+    return true;
+}
+
 // equality mod p of two norm2p values (canonicalize then compare)
+template <typename T = Fq9T>
 __device__ __forceinline__ bool fe9_eq_modp(const fe9 &a, const fe9 &b) {
-    return fe9_eq_raw(fe9_csubp(a), fe9_csubp(b));
+    return fe9_eq_raw(fe9_csubp<T>(a), fe9_csubp<T>(b));
 }
 
 }  // namespace em

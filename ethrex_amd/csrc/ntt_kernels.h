@@ -1,71 +1,82 @@
 // ============================================================================
-// Radix-2 NTT over BN254 Fr — gfx950 kernels.
+// Radix-2 NTT over BN254 Fr — gfx950 kernels, on the fe9 (9x29-bit) field
+// core (carry-free v_mad_u64_u32 columns; see gpu_field9.h).
 //
 // Transform (same definition as the oracle, oracle/bn254_oracle.c):
 //   forward:  A_j = sum_i a_i w^(ij) mod r,  w = W28^(2^(28-log2 n))
 //   inverse:  a_i = n^-1 sum_j A_j w^(-ij)
 // In/out: 32-byte big-endian canonical Fr elements, natural order.
+// Device-resident data is fe9 Montgomery(2^261), 36 B per element.
 //
-// HBM-bound path: data stays resident in fe4 (Montgomery) form; one stage
-// kernel pass streams 2 x 32 B per butterfly pair.  Twiddle tables
-// (n/2 entries, forward and inverse) are built once per plan on device.
+// Two paths (selected in api.hip):
+//   13 <= logn <= 24: four-step fused (3 tiled transposes + 2 LDS row-NTT
+//   passes); otherwise: bit-reverse + logn radix-2 stage launches.
+// Both produce identical integer results (same DFT), parity-pinned against
+// the oracle.
 // ============================================================================
 #pragma once
 #include <hip/hip_runtime.h>
-#include "gpu_field.h"
+#include "gpu_field9.h"
 
 namespace em {
 
 // ---- conversion / validation ----
 
-// BE bytes -> Montgomery fe4; flags err if elem >= r (canonical required)
+// BE bytes -> fe9 Montgomery; flags err if elem >= r (canonical required)
 __global__ void k_fr_from_be(const uint8_t *__restrict__ in,
-                             fe4 *__restrict__ out, size_t n,
+                             fe9 *__restrict__ out, size_t n,
                              uint32_t *__restrict__ err) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    fe4 v = fe_from_be(in + 32 * i);
-    if (fe_geq(v, fe4{{Fr::MOD[0], Fr::MOD[1], Fr::MOD[2], Fr::MOD[3]}}))
-        atomicOr(err, 1u);
-    out[i] = to_mont<Fr>(v);
+    const u64 *w = (const u64 *)(in + 32 * i);
+    u64 v[4] = {__builtin_bswap64(w[3]), __builtin_bswap64(w[2]),
+                __builtin_bswap64(w[1]), __builtin_bswap64(w[0])};
+    fe9 raw = fe9_from_u64x4(v);
+    if (fe9_geq_raw(raw, bn254::FR9_P)) atomicOr(err, 1u);
+    out[i] = to_mont9<Fr9T>(raw);
 }
 
-__global__ void k_fr_to_be(const fe4 *__restrict__ in, uint8_t *__restrict__ out,
+__global__ void k_fr_to_be(const fe9 *__restrict__ in, uint8_t *__restrict__ out,
                            size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    fe_to_be(out + 32 * i, from_mont<Fr>(in[i]));
+    fe9 c = from_mont9<Fr9T>(in[i]);
+    u64 w[4];
+    fe9_to_u64x4(w, c);
+    u64 *o = (u64 *)(out + 32 * i);
+    o[0] = __builtin_bswap64(w[3]);
+    o[1] = __builtin_bswap64(w[2]);
+    o[2] = __builtin_bswap64(w[1]);
+    o[3] = __builtin_bswap64(w[0]);
 }
 
-// ---- twiddle generation: tw[j] = w^j (Montgomery), j in [0, n/2) ----
-// w2k[k] = w^(2^k) precomputed on host (gpu_field host path).
-__global__ void k_gen_twiddles(fe4 *__restrict__ tw, size_t half,
-                               const fe4 *__restrict__ w2k, int logn) {
+// ---- twiddle generation: tw[j] = w^j (fe9 Montgomery), j in [0, count) ----
+// w2k[k] = w^(2^k) precomputed on host.
+__global__ void k_gen_twiddles(fe9 *__restrict__ tw, size_t count,
+                               const fe9 *__restrict__ w2k, int bits) {
     size_t j = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (j >= half) return;
-    fe4 acc = fe_one_mont<Fr>();
+    if (j >= count) return;
+    fe9 acc = fe9_load(bn254::FR9_ONE);
     size_t e = j;
-    for (int k = 0; k < logn && e; k++, e >>= 1)
-        if (e & 1) acc = mont_mul<Fr>(acc, w2k[k]);
+    for (int k = 0; k < bits && e; k++, e >>= 1)
+        if (e & 1) acc = mont_mul9<Fr9T>(acc, w2k[k]);
     tw[j] = acc;
 }
 
 // ---- bit-reverse permutation (in-place swap) ----
-__global__ void k_bit_reverse(fe4 *__restrict__ a, size_t n, int logn) {
+__global__ void k_bit_reverse(fe9 *__restrict__ a, size_t n, int logn) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     size_t j = __brevll(i) >> (64 - logn);
     if (j > i) {
-        fe4 t = a[i];
+        fe9 t = a[i];
         a[i] = a[j];
         a[j] = t;
     }
 }
 
-// ---- one radix-2 DIT stage ----
-// stage s (1-based): m = 2^s, half = m/2; thread t handles butterfly
-// (g*m + j, g*m + j + half), twiddle tw[j << (logn - s)].
-__global__ void k_ntt_stage(fe4 *__restrict__ a, const fe4 *__restrict__ tw,
+// ---- one radix-2 DIT stage (fallback path) ----
+__global__ void k_ntt_stage(fe9 *__restrict__ a, const fe9 *__restrict__ tw,
                             size_t n, int logn, int s) {
     size_t t = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= (n >> 1)) return;
@@ -73,28 +84,22 @@ __global__ void k_ntt_stage(fe4 *__restrict__ a, const fe4 *__restrict__ tw,
     size_t j = t & (half - 1);
     size_t g = t >> (s - 1);
     size_t idx = (g << s) + j;
-    fe4 u = a[idx];
-    fe4 v = mont_mul<Fr>(a[idx + half], tw[j << (logn - s)]);
-    a[idx] = mod_add<Fr>(u, v);
-    a[idx + half] = mod_sub<Fr>(u, v);
+    fe9 u = a[idx];
+    fe9 v = mont_mul9<Fr9T>(a[idx + half], tw[j << (logn - s)]);
+    a[idx] = add9_n<Fr9T>(u, v);
+    a[idx + half] = subm9<Fr9T>(u, v);  // b = mul output
 }
 
 // ---- four-step fused path (13 <= logn <= 24) ----
-// n = N1*N2; A[r][c] = a[r*N2+c].  Five passes instead of logn:
-//   T0: A1[c][r] = A[r][c]            (tiled transpose)
-//   P1: per row c: NTT_N1 over r in LDS (12 stages max, 128 KB tile),
-//       epilogue multiplies element k1 by w^(k1*c)   (tw_full table)
-//   T1: C[k1][c] = C1[c][k1]
-//   P2: per row k1: NTT_N2 over c in LDS (+ 1/n scale for iNTT)
-//   T2: out[k2*N1+k1] = D[k1][k2]
-// Identical integer results to the radix-2 path (same DFT), parity-pinned
-// against the oracle.
+// n = N1*N2; A[r][c] = a[r*N2+c]:
+//   T0: A1[c][r] = A[r][c]; P1: row NTT_N1 + w^(k1*c) twiddle;
+//   T1; P2: row NTT_N2 (+ 1/n for iNTT); T2 -> natural order.
 
-// tiled fe4 transpose, 32x32 tiles (+1 pad column kills LDS bank conflicts)
+// tiled fe9 transpose, 32x32 tiles (+1 pad column for LDS banking)
 __global__ void __launch_bounds__(256)
-k_transpose_fe4(const fe4 *__restrict__ src, fe4 *__restrict__ dst,
+k_transpose_fe9(const fe9 *__restrict__ src, fe9 *__restrict__ dst,
                 uint32_t R, uint32_t C) {
-    __shared__ fe4 tile[32][33];
+    __shared__ fe9 tile[32][33];
     uint32_t c0 = blockIdx.x * 32, r0 = blockIdx.y * 32;
     uint32_t tx = threadIdx.x & 31, ty = threadIdx.x >> 5;  // 8 rows/pass
     for (uint32_t dy = ty; dy < 32; dy += 8)
@@ -104,19 +109,13 @@ k_transpose_fe4(const fe4 *__restrict__ src, fe4 *__restrict__ dst,
         dst[(size_t)(c0 + dy) * R + r0 + tx] = tile[tx][dy];
 }
 
-// one row NTT of length M = 2^logM fully in LDS.
-// tw_row: M/2 twiddles of the size-M transform.
-// tw_full: if non-null, epilogue multiplies element k by tw_full[k*blockIdx.x]
-//          (the four-step inter-NTT twiddle; index < n always).
-// scale: if non-null, epilogue multiplies by *scale (iNTT 1/n, Montgomery).
+// one row NTT of length M = 2^logM fully in LDS (fe9: 4096*36 B = 144 KiB).
 __global__ void __launch_bounds__(1024)
-k_ntt_row(fe4 *__restrict__ data, int logM, const fe4 *__restrict__ tw_row,
-          const fe4 *__restrict__ tw_full, const fe4 *__restrict__ scale) {
-    // static 128 KiB LDS (max row length 4096 fe4); gfx950 has 160 KiB/CU —
-    // one block/CU, 8 waves, streaming kernel
-    __shared__ fe4 smem[4096];
+k_ntt_row(fe9 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
+          const fe9 *__restrict__ tw_full, const fe9 *__restrict__ scale) {
+    __shared__ fe9 smem[4096];
     const uint32_t M = 1u << logM;
-    fe4 *row = data + (size_t)blockIdx.x * M;
+    fe9 *row = data + (size_t)blockIdx.x * M;
     for (uint32_t i = threadIdx.x; i < M; i += blockDim.x) {
         uint32_t j = __brev(i) >> (32 - logM);
         smem[j] = row[i];
@@ -127,29 +126,27 @@ k_ntt_row(fe4 *__restrict__ data, int logM, const fe4 *__restrict__ tw_row,
         for (uint32_t t = threadIdx.x; t < (M >> 1); t += blockDim.x) {
             uint32_t j = t & (half - 1);
             uint32_t idx = ((t >> (s - 1)) << s) + j;
-            fe4 u = smem[idx];
-            fe4 v = mont_mul<Fr>(smem[idx + half], tw_row[j << (logM - s)]);
-            smem[idx] = mod_add<Fr>(u, v);
-            smem[idx + half] = mod_sub<Fr>(u, v);
+            fe9 u = smem[idx];
+            fe9 v = mont_mul9<Fr9T>(smem[idx + half], tw_row[j << (logM - s)]);
+            smem[idx] = add9_n<Fr9T>(u, v);
+            smem[idx + half] = subm9<Fr9T>(u, v);
         }
         __syncthreads();
     }
     uint64_t c = blockIdx.x;
     for (uint32_t k = threadIdx.x; k < M; k += blockDim.x) {
-        fe4 x = smem[k];
-        if (tw_full) x = mont_mul<Fr>(x, tw_full[(size_t)k * c]);
-        if (scale) x = mont_mul<Fr>(x, *scale);
+        fe9 x = smem[k];
+        if (tw_full) x = mont_mul9<Fr9T>(x, tw_full[(size_t)k * c]);
+        if (scale) x = mont_mul9<Fr9T>(x, *scale);
         row[k] = x;
     }
 }
 
-// ---- scale by n^-1 (iNTT) ----
-__global__ void k_ntt_scale(fe4 *__restrict__ a, size_t n, int logn) {
+// ---- scale by n^-1 (iNTT, fallback path) ----
+__global__ void k_ntt_scale(fe9 *__restrict__ a, size_t n, int logn) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    fe4 ninv{{bn254::FR_INV_POW2_MONT[logn][0], bn254::FR_INV_POW2_MONT[logn][1],
-              bn254::FR_INV_POW2_MONT[logn][2], bn254::FR_INV_POW2_MONT[logn][3]}};
-    a[i] = mont_mul<Fr>(a[i], ninv);
+    a[i] = mont_mul9<Fr9T>(a[i], fe9_load(bn254::FR9_INV_POW2[logn]));
 }
 
 }  // namespace em
