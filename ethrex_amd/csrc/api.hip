@@ -87,6 +87,7 @@ struct em_msm_plan {
     void *d_sort_tmp = nullptr;
     size_t sort_tmp_bytes = 0;
     uint32_t *d_offsets = nullptr;    // NBUCKET_TOTAL + 1
+    msm_aux *d_aux = nullptr;         // 2 per chunk (uniform partitioning)
     g1j9 *d_buckets = nullptr;         // NBUCKET_TOTAL
     g1j9 *d_seg_sum = nullptr;         // NWIN*NSEG
     g1j9 *d_seg_wsum = nullptr;
@@ -120,6 +121,8 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     mal((void **)&p->d_keys_out, total * 4);
     mal((void **)&p->d_vals_out, total * 4);
     mal((void **)&p->d_offsets, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
+    mal((void **)&p->d_aux,
+        2 * ((total + MSM_CHUNK - 1) / MSM_CHUNK) * sizeof(msm_aux));
     mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1j9));
     mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1j9));
     mal((void **)&p->d_seg_wsum, MSM_NWIN * MSM_NSEG * sizeof(g1j9));
@@ -155,6 +158,7 @@ extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
     hipFree(p->d_vals_out);
     hipFree(p->d_sort_tmp);
     hipFree(p->d_offsets);
+    hipFree(p->d_aux);
     hipFree(p->d_buckets);
     hipFree(p->d_seg_sum);
     hipFree(p->d_seg_wsum);
@@ -230,10 +234,14 @@ static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
                        dim3(blocks_for((size_t)MSM_NBUCKET_TOTAL + 1, 256)),
                        dim3(256), 0, 0, p->d_keys_out, total, p->d_offsets);
     HIP_TRY(hipEventRecord(p->ev[1], 0));
-    // bucket accumulation (hot)
-    hipLaunchKernelGGL(k_bucket_acc, dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)),
-                       dim3(256), 0, 0, p->d_pts, p->d_vals_out, p->d_offsets,
-                       p->d_buckets);
+    // bucket accumulation (hot): uniform chunks + straddle merge
+    size_t nchunks = (total + MSM_CHUNK - 1) / MSM_CHUNK;
+    hipLaunchKernelGGL(k_chunk_acc, dim3(blocks_for(nchunks, 256)), dim3(256),
+                       0, 0, p->d_pts, p->d_vals_out, p->d_offsets,
+                       p->d_buckets, p->d_aux, total);
+    hipLaunchKernelGGL(k_bucket_merge,
+                       dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)), dim3(256), 0,
+                       0, p->d_offsets, p->d_aux, p->d_buckets);
     HIP_TRY(hipEventRecord(p->ev[2], 0));
     // reductions
     hipLaunchKernelGGL(k_segment_reduce,
