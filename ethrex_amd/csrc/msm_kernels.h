@@ -179,9 +179,10 @@ k_chunk_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
     }
     uint32_t b = lo;
     bool carry_in = offsets[b] < (uint32_t)s;
-    msm_aux a0, a1;
-    a0.bid = MSM_AUX_INVALID;
-    a1.bid = MSM_AUX_INVALID;
+    // aux slots written straight to global on emission (keeping them in
+    // registers costs ~80 VGPRs and halves occupancy)
+    aux[2 * t].bid = MSM_AUX_INVALID;
+    aux[2 * t + 1].bid = MSM_AUX_INVALID;
     g1j9 acc = g1_inf9();
     uint32_t idx = (uint32_t)s;
     while (idx < e) {
@@ -197,8 +198,8 @@ k_chunk_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
             // bucket b closes inside this chunk
             if ((b & MSM_DMASK) != 0) {
                 if (carry_in) {
-                    a0.bid = b;
-                    a0.pt = acc;
+                    aux[2 * t].pt = acc;
+                    aux[2 * t].bid = b;
                 } else {
                     buckets[b] = acc;  // fully contained: final
                 }
@@ -210,19 +211,13 @@ k_chunk_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
         } else {
             // bucket continues past the chunk: emit partial
             if ((b & MSM_DMASK) != 0) {
-                if (carry_in) {
-                    a0.bid = b;
-                    a0.pt = acc;
-                } else {
-                    a1.bid = b;
-                    a1.pt = acc;
-                }
+                size_t slot = carry_in ? 2 * t : 2 * t + 1;
+                aux[slot].pt = acc;
+                aux[slot].bid = b;
             }
             break;
         }
     }
-    aux[2 * t] = a0;
-    aux[2 * t + 1] = a1;
 }
 
 // assemble straddling buckets from chunk partials; also writes empty buckets
