@@ -185,14 +185,23 @@ k_chunk_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
     aux[2 * t + 1].bid = MSM_AUX_INVALID;
     g1j9 acc = g1_inf9();
     uint32_t idx = (uint32_t)s;
+    // software pipeline: keep the NEXT entry's point in flight across the
+    // ~7k-cycle mixed add (the dependent vals->pts chain is ~2 memory
+    // round trips); prefetch crosses bucket boundaries freely.
+    g1a9 p = pts[vals[idx]];
     while (idx < e) {
         uint32_t bend = offsets[b + 1];
         uint32_t stop = bend < e ? bend : e;
         if ((b & MSM_DMASK) != 0) {
-            for (; idx < stop; idx++)
-                acc = g1_add_affine9(acc, pts[vals[idx]]);
+            for (; idx < stop; idx++) {
+                g1a9 cur = p;
+                uint32_t nidx = idx + 1 < e ? idx + 1 : idx;
+                p = pts[vals[nidx]];
+                acc = g1_add_affine9(acc, cur);
+            }
         } else {
             idx = stop;  // digit-0 run: consume, contribute nothing
+            if (idx < e) p = pts[vals[idx]];
         }
         if (stop == bend) {
             // bucket b closes inside this chunk
