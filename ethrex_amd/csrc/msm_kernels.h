@@ -144,112 +144,28 @@ __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total
     offsets[b] = (uint32_t)lo;
 }
 
-// ---- bucket accumulation (the hot kernel): uniform work partitioning ----
-// Every thread processes EXACTLY MSM_CHUNK consecutive sorted entries, so
-// waves have no variable-run-length divergence (a thread-per-bucket kernel
-// pays max-of-64 Poisson run lengths, ~1.16x at c=16).  Buckets fully
-// contained in a chunk are written directly; the chunk's first/last
-// straddling buckets emit <=2 tagged partials merged by k_bucket_merge.
-
-constexpr uint32_t MSM_CHUNK = 64;
-constexpr uint32_t MSM_AUX_INVALID = 0xffffffffu;
-
-struct msm_aux {
-    g1j9 pt;
-    uint32_t bid;
-    uint32_t pad[3];
-};
-
+// ---- bucket accumulation (the hot kernel) ----
+// one thread per bucket id; digit-0 buckets are skipped (never read later).
 __global__ void __launch_bounds__(256)
-k_chunk_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
-            const uint32_t *__restrict__ offsets, g1j9 *__restrict__ buckets,
-            msm_aux *__restrict__ aux, size_t total) {
-    size_t t = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-    size_t s = t * MSM_CHUNK;
-    if (s >= total) return;
-    uint32_t e = (uint32_t)(s + MSM_CHUNK < total ? s + MSM_CHUNK : total);
-    // b = bucket containing entry s: offsets[b] <= s < offsets[b+1]
-    uint32_t lo = 0, hi = MSM_NBUCKET_TOTAL;
-    while (lo < hi) {
-        uint32_t mid = (lo + hi + 1) >> 1;
-        if (offsets[mid] <= (uint32_t)s)
-            lo = mid;
-        else
-            hi = mid - 1;
-    }
-    uint32_t b = lo;
-    bool carry_in = offsets[b] < (uint32_t)s;
-    // aux slots written straight to global on emission (keeping them in
-    // registers costs ~80 VGPRs and halves occupancy)
-    aux[2 * t].bid = MSM_AUX_INVALID;
-    aux[2 * t + 1].bid = MSM_AUX_INVALID;
-    g1j9 acc = g1_inf9();
-    uint32_t idx = (uint32_t)s;
-    // software pipeline: keep the NEXT entry's point in flight across the
-    // ~7k-cycle mixed add (the dependent vals->pts chain is ~2 memory
-    // round trips); prefetch crosses bucket boundaries freely.
-    g1a9 p = pts[vals[idx]];
-    while (idx < e) {
-        uint32_t bend = offsets[b + 1];
-        uint32_t stop = bend < e ? bend : e;
-        if ((b & MSM_DMASK) != 0) {
-            for (; idx < stop; idx++) {
-                g1a9 cur = p;
-                uint32_t nidx = idx + 1 < e ? idx + 1 : idx;
-                p = pts[vals[nidx]];
-                acc = g1_add_affine9(acc, cur);
-            }
-        } else {
-            idx = stop;  // digit-0 run: consume, contribute nothing
-            if (idx < e) p = pts[vals[idx]];
-        }
-        if (stop == bend) {
-            // bucket b closes inside this chunk
-            if ((b & MSM_DMASK) != 0) {
-                if (carry_in) {
-                    aux[2 * t].pt = acc;
-                    aux[2 * t].bid = b;
-                } else {
-                    buckets[b] = acc;  // fully contained: final
-                }
-            }
-            carry_in = false;
-            acc = g1_inf9();
-            // advance to the bucket containing idx (skip empties)
-            while (b + 1 < MSM_NBUCKET_TOTAL && offsets[b + 1] <= idx) b++;
-        } else {
-            // bucket continues past the chunk: emit partial
-            if ((b & MSM_DMASK) != 0) {
-                size_t slot = carry_in ? 2 * t : 2 * t + 1;
-                aux[slot].pt = acc;
-                aux[slot].bid = b;
-            }
-            break;
-        }
-    }
-}
-
-// assemble straddling buckets from chunk partials; also writes empty buckets
-__global__ void __launch_bounds__(256)
-k_bucket_merge(const uint32_t *__restrict__ offsets,
-               const msm_aux *__restrict__ aux, g1j9 *__restrict__ buckets) {
+k_bucket_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
+             const uint32_t *__restrict__ offsets, g1j9 *__restrict__ buckets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= MSM_NBUCKET_TOTAL) return;
-    if ((b & MSM_DMASK) == 0) return;  // digit-0 ids unused
-    uint32_t s = offsets[b], e = offsets[b + 1];
-    if (s == e) {
-        buckets[b] = g1_inf9();
+    if ((b & MSM_DMASK) == 0) return;  // digit 0
+    uint32_t lo = offsets[b], hi = offsets[b + 1];
+    g1j9 acc = g1_inf9();
+    if (lo >= hi) {
+        buckets[b] = acc;
         return;
     }
-    uint32_t c0 = s / MSM_CHUNK, c1 = (e - 1) / MSM_CHUNK;
-    if (c0 == c1) return;  // fully contained: direct-written by k_chunk_acc
-    g1j9 acc = g1_inf9();
-    for (uint32_t c = c0; c <= c1; c++) {
-#pragma unroll
-        for (int slot = 0; slot < 2; slot++) {
-            msm_aux a = aux[2 * (size_t)c + slot];
-            if (a.bid == b) acc = g1_add9(acc, a.pt);
-        }
+    // software pipeline: issue the NEXT point's gather before the long mixed
+    // add so the dependent idx->point load chain overlaps the VALU work.
+    g1a9 p = pts[vals[lo]];
+    for (uint32_t t = lo; t < hi; t++) {
+        g1a9 cur = p;
+        uint32_t nxt = t + 1 < hi ? t + 1 : t;
+        p = pts[vals[nxt]];
+        acc = g1_add_affine9(acc, cur);
     }
     buckets[b] = acc;
 }
