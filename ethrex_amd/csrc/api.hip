@@ -368,36 +368,33 @@ struct em_ntt_plan {
     bool fused = false;       // four-step path (13 <= logn <= 24)
     int logN1 = 0, logN2 = 0;
     int cur = 0;              // which buffer holds the data: 0=d_data 1=d_work
-    fe4 *d_data = nullptr;
-    fe4 *d_work = nullptr;    // fused: transpose ping-pong buffer
+    fe9 *d_data = nullptr;
+    fe9 *d_work = nullptr;    // fused: transpose ping-pong buffer
     uint8_t *d_bytes = nullptr;
-    fe4 *d_tw = nullptr;      // fallback: forward twiddles, n/2
-    fe4 *d_tw_inv = nullptr;
-    fe4 *d_twfull = nullptr;      // fused: w^j, j<n
-    fe4 *d_twfull_inv = nullptr;
-    fe4 *d_twrow1 = nullptr;      // fused: N1/2 row twiddles (+inv)
-    fe4 *d_twrow1_inv = nullptr;
-    fe4 *d_twrow2 = nullptr;      // fused: N2/2 (+inv)
-    fe4 *d_twrow2_inv = nullptr;
-    fe4 *d_ninv = nullptr;        // 1/n (Montgomery)
+    fe9 *d_tw = nullptr;      // fallback: forward twiddles, n/2
+    fe9 *d_tw_inv = nullptr;
+    fe9 *d_twfull = nullptr;      // fused: w^j, j<n
+    fe9 *d_twfull_inv = nullptr;
+    fe9 *d_twrow1 = nullptr;      // fused: N1/2 row twiddles (+inv)
+    fe9 *d_twrow1_inv = nullptr;
+    fe9 *d_twrow2 = nullptr;      // fused: N2/2 (+inv)
+    fe9 *d_twrow2_inv = nullptr;
+    fe9 *d_ninv = nullptr;        // 1/n (fe9 Montgomery)
     uint32_t *d_err = nullptr;
     hipEvent_t ev[4];
     double last_ms[3] = {0, 0, 0};
 };
 
 // build w2k powers on host and launch k_gen_twiddles: tw[j] = w^j, j < count
-static int gen_tw_table(fe4 *d_out, size_t count, int bits, bool inverse,
+static int gen_tw_table(fe9 *d_out, size_t count, int bits, bool inverse,
                         int log_size /* transform size 2^log_size */) {
-    fe4 w = inverse
-                ? fe4{{bn254::FR_W28_INV_MONT[0], bn254::FR_W28_INV_MONT[1],
-                       bn254::FR_W28_INV_MONT[2], bn254::FR_W28_INV_MONT[3]}}
-                : fe4{{bn254::FR_W28_MONT[0], bn254::FR_W28_MONT[1],
-                       bn254::FR_W28_MONT[2], bn254::FR_W28_MONT[3]}};
-    for (int k = bn254::FR_TWO_ADICITY; k > log_size; k--) w = mont_sqr<Fr>(w);
-    fe4 w2k[32];
-    w2k[0] = log_size == 0 ? fe_one_mont<Fr>() : w;
-    for (int k = 1; k < bits && k < 32; k++) w2k[k] = mont_sqr<Fr>(w2k[k - 1]);
-    fe4 *d_w2k;
+    fe9 w = fe9_load(inverse ? bn254::FR9_W28_INV : bn254::FR9_W28);
+    for (int k = bn254::FR_TWO_ADICITY; k > log_size; k--)
+        w = mont_sqr9<Fr9T>(w);
+    fe9 w2k[32];
+    w2k[0] = log_size == 0 ? fe9_load(bn254::FR9_ONE) : w;
+    for (int k = 1; k < bits && k < 32; k++) w2k[k] = mont_sqr9<Fr9T>(w2k[k - 1]);
+    fe9 *d_w2k;
     HIP_TRY(hipMalloc(&d_w2k, sizeof(w2k)));
     HIP_TRY(hipMemcpy(d_w2k, w2k, sizeof(w2k), hipMemcpyHostToDevice));
     hipLaunchKernelGGL(k_gen_twiddles, dim3(blocks_for(count, 256)), dim3(256),
@@ -424,25 +421,25 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
     auto mal = [&](void **ptr, size_t bytes) {
         if (e == hipSuccess) e = hipMalloc(ptr, bytes);
     };
-    mal((void **)&p->d_data, n * sizeof(fe4));
+    mal((void **)&p->d_data, n * sizeof(fe9));
     mal((void **)&p->d_bytes, n * 32);
     mal((void **)&p->d_err, 4);
     if (p->fused) {
         p->logN1 = (logn + 1) / 2;
         p->logN2 = logn / 2;
-        mal((void **)&p->d_work, n * sizeof(fe4));
-        mal((void **)&p->d_twfull, n * sizeof(fe4));
-        mal((void **)&p->d_twfull_inv, n * sizeof(fe4));
-        mal((void **)&p->d_twrow1, ((size_t)1 << (p->logN1 - 1)) * sizeof(fe4));
-        mal((void **)&p->d_twrow1_inv, ((size_t)1 << (p->logN1 - 1)) * sizeof(fe4));
+        mal((void **)&p->d_work, n * sizeof(fe9));
+        mal((void **)&p->d_twfull, n * sizeof(fe9));
+        mal((void **)&p->d_twfull_inv, n * sizeof(fe9));
+        mal((void **)&p->d_twrow1, ((size_t)1 << (p->logN1 - 1)) * sizeof(fe9));
+        mal((void **)&p->d_twrow1_inv, ((size_t)1 << (p->logN1 - 1)) * sizeof(fe9));
         mal((void **)&p->d_twrow2,
-            ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe4));
+            ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe9));
         mal((void **)&p->d_twrow2_inv,
-            ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe4));
-        mal((void **)&p->d_ninv, sizeof(fe4));
+            ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe9));
+        mal((void **)&p->d_ninv, sizeof(fe9));
     } else {
-        mal((void **)&p->d_tw, half * sizeof(fe4));
-        mal((void **)&p->d_tw_inv, half * sizeof(fe4));
+        mal((void **)&p->d_tw, half * sizeof(fe9));
+        mal((void **)&p->d_tw_inv, half * sizeof(fe9));
     }
     for (int i = 0; i < 4 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
     if (e != hipSuccess) {
@@ -461,9 +458,8 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
                                 p->logN2, false, p->logN2))) return rc2;
         if ((rc2 = gen_tw_table(p->d_twrow2_inv, (size_t)1 << (p->logN2 - 1),
                                 p->logN2, true, p->logN2))) return rc2;
-        fe4 ninv{{bn254::FR_INV_POW2_MONT[logn][0], bn254::FR_INV_POW2_MONT[logn][1],
-                  bn254::FR_INV_POW2_MONT[logn][2], bn254::FR_INV_POW2_MONT[logn][3]}};
-        HIP_TRY(hipMemcpy(p->d_ninv, &ninv, sizeof(fe4), hipMemcpyHostToDevice));
+        fe9 ninv = fe9_load(bn254::FR9_INV_POW2[logn]);
+        HIP_TRY(hipMemcpy(p->d_ninv, &ninv, sizeof(fe9), hipMemcpyHostToDevice));
     } else {
         int rc2;
         if ((rc2 = gen_tw_table(p->d_tw, half, logn ? logn : 1, false, logn)))
@@ -509,30 +505,30 @@ extern "C" int ethrex_mi355_ntt_upload(em_ntt_plan *p, const uint8_t *elems32) {
 extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
     if (!p) return EM_ERR_INPUT;
     size_t n = p->n;
-    fe4 *cur = p->cur ? p->d_work : p->d_data;
-    fe4 *oth = p->cur ? p->d_data : p->d_work;
+    fe9 *cur = p->cur ? p->d_work : p->d_data;
+    fe9 *oth = p->cur ? p->d_data : p->d_work;
     HIP_TRY(hipEventRecord(p->ev[0], 0));
     if (p->fused) {
         uint32_t N1 = 1u << p->logN1, N2 = 1u << p->logN2;
         // T0: A[r][c] -> A1[c][r]
-        hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256), 0,
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256), 0,
                            0, cur, oth, N1, N2);
         HIP_TRY(hipEventRecord(p->ev[1], 0));
         // P1: row NTT_N1 over each of the N2 rows + w^(k1*c) twiddle
         hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(1024), 0, 0, oth,
                            p->logN1, inverse ? p->d_twrow1_inv : p->d_twrow1,
                            inverse ? p->d_twfull_inv : p->d_twfull,
-                           (const fe4 *)nullptr);
+                           (const fe9 *)nullptr);
         // T1
-        hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256), 0,
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(N1 / 32, N2 / 32), dim3(256), 0,
                            0, oth, cur, N2, N1);
         // P2: row NTT_N2 (+ 1/n scale on iNTT)
         hipLaunchKernelGGL(k_ntt_row, dim3(N1), dim3(1024), 0, 0, cur,
                            p->logN2, inverse ? p->d_twrow2_inv : p->d_twrow2,
-                           (const fe4 *)nullptr,
-                           inverse ? p->d_ninv : (const fe4 *)nullptr);
+                           (const fe9 *)nullptr,
+                           inverse ? p->d_ninv : (const fe9 *)nullptr);
         // T2: natural order
-        hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256), 0,
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256), 0,
                            0, cur, oth, N1, N2);
         p->cur ^= 1;
     } else {
@@ -541,7 +537,7 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
                                0, 0, cur, n, p->logn);
         }
         HIP_TRY(hipEventRecord(p->ev[1], 0));
-        const fe4 *tw = inverse ? p->d_tw_inv : p->d_tw;
+        const fe9 *tw = inverse ? p->d_tw_inv : p->d_tw;
         for (int s = 1; s <= p->logn; s++) {
             hipLaunchKernelGGL(k_ntt_stage, dim3(blocks_for(n / 2, 256)),
                                dim3(256), 0, 0, cur, tw, n, p->logn, s);
@@ -565,7 +561,7 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
 
 extern "C" int ethrex_mi355_ntt_download(em_ntt_plan *p, uint8_t *elems32) {
     if (!p || !elems32) return EM_ERR_INPUT;
-    fe4 *cur = p->cur ? p->d_work : p->d_data;
+    fe9 *cur = p->cur ? p->d_work : p->d_data;
     hipLaunchKernelGGL(k_fr_to_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
                        cur, p->d_bytes, p->n);
     HIP_TRY(hipMemcpy(elems32, p->d_bytes, p->n * 32, hipMemcpyDeviceToHost));
