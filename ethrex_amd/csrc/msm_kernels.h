@@ -174,23 +174,28 @@ k_bucket_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
 // level 1: per (window, 32-bucket segment): from the top digit down,
 //   run  += B_d           (=> run  = sum of segment buckets)
 //   wsum += run           (=> wsum = sum (d - lo + 1) * B_d)
-__global__ void k_segment_reduce(const g1j9 *__restrict__ buckets,
-                                 g1j9 *__restrict__ seg_sum,
-                                 g1j9 *__restrict__ seg_wsum) {
+__global__ void __launch_bounds__(256)
+k_segment_reduce(const g1j9 *__restrict__ buckets,
+                 g1j9 *__restrict__ seg_sum,
+                 g1j9 *__restrict__ seg_wsum) {
+    // the second accumulator lives in LDS: two register XYZZ accumulators
+    // plus mul temporaries spill 232 B/lane to scratch otherwise
+    __shared__ g1j9 lds_wsum[256];
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;  // [0, 16*2048)
     if (t >= MSM_NWIN * MSM_NSEG) return;
     uint32_t w = t / MSM_NSEG, seg = t % MSM_NSEG;
     uint32_t lo = seg * MSM_SEG;
-    g1j9 run = g1_inf9(), wsum = g1_inf9();
+    g1j9 run = g1_inf9();
+    lds_wsum[threadIdx.x] = g1_inf9();
     for (int32_t d = (int32_t)lo + MSM_SEG - 1; d >= (int32_t)lo; d--) {
         // digit-0 bucket is unused: skip the add but keep the wsum step so
         // segment 0 carries the same (d - lo + 1) weights
         if (d != 0)
             run = g1_add9(run, buckets[((uint32_t)w << MSM_C) | (uint32_t)d]);
-        wsum = g1_add9(wsum, run);
+        lds_wsum[threadIdx.x] = g1_add9(lds_wsum[threadIdx.x], run);
     }
     seg_sum[t] = run;
-    seg_wsum[t] = wsum;
+    seg_wsum[t] = lds_wsum[threadIdx.x];
 }
 
 // level 2: fully parallel weighted combine + LDS tree reduction.

@@ -178,3 +178,18 @@ def test_gpu_ntt_roundtrip_2_20(gpu):
     assert rc == 0 and fwd != elems
     rc, back = gpu.fr_ntt(fwd, n, True)
     assert rc == 0 and back == elems
+
+
+def test_gpu_msm_parity_2_20_direct(gpu, oracle_mod):
+    """Direct bit-exact parity vs the oracle at 2^20 (oracle ~1-4 s on the
+    GPU box's host cores)."""
+    n = 1 << 20
+    plan = gpu.MsmPlan(n)
+    plan.gen_points(0)
+    pts = plan.download_points()
+    scs = gpu.gen_fr(42, n)
+    plan.upload_scalars(scs)
+    got = plan.run()
+    plan.destroy()
+    rc, want = oracle_mod.g1_msm(pts, scs, n)
+    assert rc == 0 and got == want
