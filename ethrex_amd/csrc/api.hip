@@ -231,7 +231,7 @@ static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
                        dim3(256), 0, 0, p->d_keys_out, total, p->d_offsets);
     HIP_TRY(hipEventRecord(p->ev[1], 0));
     // bucket accumulation (hot)
-    hipLaunchKernelGGL(k_bucket_acc, dim3(blocks_for(MSM_NBUCKET_TOTAL / 2, 256)),
+    hipLaunchKernelGGL(k_bucket_acc, dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)),
                        dim3(256), 0, 0, p->d_pts, p->d_vals_out, p->d_offsets,
                        p->d_buckets);
     HIP_TRY(hipEventRecord(p->ev[2], 0));

@@ -145,43 +145,29 @@ __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total
 }
 
 // ---- bucket accumulation (the hot kernel) ----
-// TWO independent bucket chains per thread: the Montgomery reduce rounds
-// are a serial dependency chain, so interleaving two mixed adds in one
-// instruction stream lets the scheduler hide each chain's latency in the
-// other's mads (PMC showed 57% issue-stall on the single-chain version).
-// Digit-0 buckets are skipped (never read later).
+// one thread per bucket id; digit-0 buckets are skipped (never read later).
 __global__ void __launch_bounds__(256)
 k_bucket_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
              const uint32_t *__restrict__ offsets, g1j9 *__restrict__ buckets) {
-    uint32_t h = blockIdx.x * blockDim.x + threadIdx.x;
-    if (h >= MSM_NBUCKET_TOTAL / 2) return;
-    uint32_t b1 = h, b2 = h + MSM_NBUCKET_TOTAL / 2;
-    bool use1 = (b1 & MSM_DMASK) != 0, use2 = (b2 & MSM_DMASK) != 0;
-    uint32_t t1 = 0, hi1 = 0, t2 = 0, hi2 = 0;
-    if (use1) {
-        t1 = offsets[b1];
-        hi1 = offsets[b1 + 1];
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= MSM_NBUCKET_TOTAL) return;
+    if ((b & MSM_DMASK) == 0) return;  // digit 0
+    uint32_t lo = offsets[b], hi = offsets[b + 1];
+    g1j9 acc = g1_inf9();
+    if (lo >= hi) {
+        buckets[b] = acc;
+        return;
     }
-    if (use2) {
-        t2 = offsets[b2];
-        hi2 = offsets[b2 + 1];
+    // software pipeline: issue the NEXT point's gather before the long mixed
+    // add so the dependent idx->point load chain overlaps the VALU work.
+    g1a9 p = pts[vals[lo]];
+    for (uint32_t t = lo; t < hi; t++) {
+        g1a9 cur = p;
+        uint32_t nxt = t + 1 < hi ? t + 1 : t;
+        p = pts[vals[nxt]];
+        acc = g1_add_affine9(acc, cur);
     }
-    g1j9 acc1 = g1_inf9(), acc2 = g1_inf9();
-    g1a9 p1, p2;
-    if (t1 < hi1) p1 = pts[vals[t1]];
-    if (t2 < hi2) p2 = pts[vals[t2]];
-    while (t1 < hi1 || t2 < hi2) {
-        bool d1 = t1 < hi1, d2 = t2 < hi2;
-        g1a9 c1 = p1, c2 = p2;
-        if (d1) p1 = pts[vals[t1 + 1 < hi1 ? t1 + 1 : t1]];
-        if (d2) p2 = pts[vals[t2 + 1 < hi2 ? t2 + 1 : t2]];
-        if (d1) acc1 = g1_add_affine9(acc1, c1);
-        if (d2) acc2 = g1_add_affine9(acc2, c2);
-        t1 += d1;
-        t2 += d2;
-    }
-    if (use1) buckets[b1] = acc1;
-    if (use2) buckets[b2] = acc2;
+    buckets[b] = acc;
 }
 
 // ---- two-level running-sum reduction ----
