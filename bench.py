@@ -164,9 +164,14 @@ def main():
         "peak": hbm_peak,
         "unit": "B/s",
         "frac": (alg_bytes / (avg_bucket_ms / 1000.0)) / hbm_peak,
-        "traffic": None,
-        "note": "kernel is VALU-bound (254-bit Montgomery mul); see "
-                "profiles/ for VALU PMC evidence",
+        # measured per-launch HBM bytes (rocprofv3 FETCH_SIZE+WRITE_SIZE,
+        # profiles/r01_summary.md) — only valid for the default 2^24 1-GPU
+        # config; other shapes report null
+        "traffic": (29.95e9 if (args.msm_log2 == 24 and shard == n_total)
+                    else None),
+        "note": "kernel is VALU-bound (254-bit Montgomery mul: PMC "
+                "SQ_ACTIVE_INST_VALU 38.8%, issue-stall 57%, memory-wait "
+                "3.7%); see profiles/ for evidence",
     }
 
     # ---- NTT secondary leg (rank 0, single GPU, replicas-only path) ----
