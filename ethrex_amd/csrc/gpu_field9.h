@@ -211,54 +211,6 @@ __device__ __host__ __forceinline__ fe9 mont_sqr9(const fe9 &a) {
     return mont_mul9<T>(a, a);
 }
 
-// two INDEPENDENT Montgomery muls with interleaved reduction rounds: the
-// serial per-round dependency chains (m_k -> mads -> carry) of the two
-// muls zip together so the scheduler hides each chain's latency in the
-// other's mads (the single-mul kernel measures 57% issue-stall).
-template <typename T = Fq9T>
-__device__ __forceinline__ void mont_mul9_pair(const fe9 &A1, const fe9 &B1,
-                                               const fe9 &A2, const fe9 &B2,
-                                               fe9 &O1, fe9 &O2) {
-    u64 t1[17], t2[17];
-#pragma unroll
-    for (int k = 0; k < 17; k++) {
-        t1[k] = 0;
-        t2[k] = 0;
-    }
-#pragma unroll
-    for (int i = 0; i < 9; i++) {
-#pragma unroll
-        for (int j = 0; j < 9; j++) {
-            t1[i + j] += (u64)A1.v[i] * B1.v[j];
-            t2[i + j] += (u64)A2.v[i] * B2.v[j];
-        }
-    }
-#pragma unroll
-    for (int k = 0; k < 9; k++) {
-        u32 m1 = ((u32)t1[k] * T::N0INV) & bn254::FQ9_MASK;
-        u32 m2 = ((u32)t2[k] * T::N0INV) & bn254::FQ9_MASK;
-#pragma unroll
-        for (int j = 0; j < 9; j++) {
-            t1[k + j] += (u64)m1 * T::P[j];
-            t2[k + j] += (u64)m2 * T::P[j];
-        }
-        t1[k + 1] += t1[k] >> 29;
-        t2[k + 1] += t2[k] >> 29;
-    }
-    u64 c1 = 0, c2 = 0;
-#pragma unroll
-    for (int k = 9; k < 17; k++) {
-        c1 += t1[k];
-        c2 += t2[k];
-        O1.v[k - 9] = (u32)c1 & bn254::FQ9_MASK;
-        O2.v[k - 9] = (u32)c2 & bn254::FQ9_MASK;
-        c1 >>= 29;
-        c2 >>= 29;
-    }
-    O1.v[8] = (u32)c1;
-    O2.v[8] = (u32)c2;
-}
-
 // ---- conversions fe4 (4x64 canonical) <-> fe9 ----
 
 // canonical (or any < 2^256) u64[4] -> 29-bit limbs (raw, norm limbs)

@@ -86,26 +86,22 @@ __device__ __forceinline__ g1j9 g1_add_affine9(const g1j9 &p, const g1a9 &q) {
         o.zzz = fe9_load(bn254::FQ9_ONE);
         return o;
     }
-    // the 10 muls pair into 5 independent dual-muls (mont_mul9_pair):
-    // reduce-round latency of each hides in the other's mads
-    fe9 u2, s2;
-    mont_mul9_pair(q.x, p.zz, q.y, p.zzz, u2, s2);
+    fe9 u2 = mont_mul9(q.x, p.zz);
+    fe9 s2 = mont_mul9(q.y, p.zzz);
     fe9 P = subn9(u2, p.x);                    // b=X1 norm2p
     fe9 R = subn9(s2, p.y);
     if (__builtin_expect(fe9_is_zero_modp(P), 0)) {
         if (fe9_is_zero_modp(R)) return g1_dbl9(p);
         return g1_inf9();
     }
-    fe9 PP, RR;
-    mont_mul9_pair(P, P, R, R, PP, RR);
-    fe9 PPP, Q;
-    mont_mul9_pair(P, PP, p.x, PP, PPP, Q);
+    fe9 PP = mont_sqr9(P);
+    fe9 PPP = mont_mul9(P, PP);
+    fe9 Q = mont_mul9(p.x, PP);
     g1j9 o;
-    o.x = subm9(subm9(subm9(RR, PPP), Q), Q);
-    fe9 t1, t2;
-    mont_mul9_pair(R, subn9(Q, o.x), p.y, PPP, t1, t2);
-    o.y = subm9(t1, t2);
-    mont_mul9_pair(p.zz, PP, p.zzz, PPP, o.zz, o.zzz);
+    o.x = subm9(subm9(subm9(mont_sqr9(R), PPP), Q), Q);
+    o.y = subm9(mont_mul9(R, subn9(Q, o.x)), mont_mul9(p.y, PPP));
+    o.zz = mont_mul9(p.zz, PP);
+    o.zzz = mont_mul9(p.zzz, PPP);
     return o;
 }
 
