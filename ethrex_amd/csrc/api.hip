@@ -80,11 +80,10 @@ struct em_msm_plan {
     uint8_t *d_inf = nullptr;
     fe4 *d_scalars = nullptr;
     uint8_t *d_scratch = nullptr;     // 64n bytes: point/scalar byte staging
-    uint16_t *d_keys = nullptr;       // 16n digit keys (u16)
+    uint32_t *d_keys = nullptr;       // 16n
     uint32_t *d_vals = nullptr;
-    uint16_t *d_keys_out = nullptr;
+    uint32_t *d_keys_out = nullptr;
     uint32_t *d_vals_out = nullptr;
-    uint32_t *d_segofs = nullptr;     // 17 window-segment offsets
     void *d_sort_tmp = nullptr;
     size_t sort_tmp_bytes = 0;
     uint32_t *d_offsets = nullptr;    // NBUCKET_TOTAL + 1
@@ -116,11 +115,10 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     mal((void **)&p->d_inf, n);
     mal((void **)&p->d_scalars, n * sizeof(fe4));
     mal((void **)&p->d_scratch, n * 64);
-    mal((void **)&p->d_keys, total * 2);
+    mal((void **)&p->d_keys, total * 4);
     mal((void **)&p->d_vals, total * 4);
-    mal((void **)&p->d_keys_out, total * 2);
+    mal((void **)&p->d_keys_out, total * 4);
     mal((void **)&p->d_vals_out, total * 4);
-    mal((void **)&p->d_segofs, (MSM_NWIN + 1) * 4);
     mal((void **)&p->d_offsets, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
     mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1j9));
     mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1j9));
@@ -130,18 +128,10 @@ extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     mal((void **)&p->d_out, 96);
     mal((void **)&p->d_err, 4);
     if (e == hipSuccess) {
-        // window-segment boundaries [0, n, 2n, ...]
-        uint32_t h_segofs[MSM_NWIN + 1];
-        for (int i = 0; i <= MSM_NWIN; i++) h_segofs[i] = (uint32_t)(i * n);
-        e = hipMemcpy(p->d_segofs, h_segofs, sizeof(h_segofs),
-                      hipMemcpyHostToDevice);
-        // rocPRIM segmented-sort temp-storage size query (u16 digit keys,
-        // one segment per window: 2 radix passes instead of 3)
-        if (e == hipSuccess)
-            e = rocprim::segmented_radix_sort_pairs(
-                nullptr, p->sort_tmp_bytes, p->d_keys, p->d_keys_out,
-                p->d_vals, p->d_vals_out, total, MSM_NWIN, p->d_segofs,
-                p->d_segofs + 1, 0, 16);
+        // rocPRIM temp-storage size query
+        e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
+                                      p->d_keys_out, p->d_vals, p->d_vals_out,
+                                      total, 0, MSM_SORT_BITS);
         if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
     }
     for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
@@ -163,7 +153,6 @@ extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
     hipFree(p->d_vals);
     hipFree(p->d_keys_out);
     hipFree(p->d_vals_out);
-    hipFree(p->d_segofs);
     hipFree(p->d_sort_tmp);
     hipFree(p->d_offsets);
     hipFree(p->d_buckets);
@@ -232,14 +221,14 @@ static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
                        p->d_scalars, p->d_inf, p->d_keys, p->d_vals, p->n);
     // sort on 20 key bits
     size_t tmp = p->sort_tmp_bytes;
-    hipError_t e = rocprim::segmented_radix_sort_pairs(
-        p->d_sort_tmp, tmp, p->d_keys, p->d_keys_out, p->d_vals,
-        p->d_vals_out, total, MSM_NWIN, p->d_segofs, p->d_segofs + 1, 0, 16);
-    if (e != hipSuccess) return hip_fail(e, "segmented_radix_sort_pairs");
+    hipError_t e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
+                                             p->d_keys_out, p->d_vals,
+                                             p->d_vals_out, total, 0, MSM_SORT_BITS);
+    if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs");
     // offsets
     hipLaunchKernelGGL(k_offsets,
                        dim3(blocks_for((size_t)MSM_NBUCKET_TOTAL + 1, 256)),
-                       dim3(256), 0, 0, p->d_keys_out, p->n, p->d_offsets);
+                       dim3(256), 0, 0, p->d_keys_out, total, p->d_offsets);
     HIP_TRY(hipEventRecord(p->ev[1], 0));
     // bucket accumulation (hot)
     hipLaunchKernelGGL(k_bucket_acc, dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)),

@@ -106,10 +106,10 @@ __global__ void k_parse_scalars(const uint8_t *__restrict__ in,
     out[i] = from_mont<Fr>(to_mont<Fr>(fe_from_be(in + 32 * i)));
 }
 
-// ---- digit extraction (u16 keys; window = sort segment) ----
+// ---- digit extraction ----
 __global__ void k_digits(const fe4 *__restrict__ scalars,
                          const uint8_t *__restrict__ inf,
-                         uint16_t *__restrict__ keys, uint32_t *__restrict__ vals,
+                         uint32_t *__restrict__ keys, uint32_t *__restrict__ vals,
                          size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
@@ -119,33 +119,29 @@ __global__ void k_digits(const fe4 *__restrict__ scalars,
     for (int w = 0; w < MSM_NWIN; w++) {
         uint32_t d = msm_digit(k, w);
         if (skip) d = 0;  // identity points contribute nothing: park in bucket 0
-        keys[(size_t)w * n + i] = (uint16_t)d;
+        keys[(size_t)w * n + i] = ((uint32_t)w << MSM_C) | d;
         vals[(size_t)w * n + i] = (uint32_t)i;
     }
 }
 
-// ---- bucket segment offsets: lower_bound of the digit WITHIN its window's
-// sorted segment [w*n, (w+1)*n) ----
-__global__ void k_offsets(const uint16_t *__restrict__ sorted_keys, size_t n,
+// ---- bucket segment offsets: lower_bound of each bucket id ----
+__global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total,
                           uint32_t *__restrict__ offsets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b > MSM_NBUCKET_TOTAL) return;
     if (b == MSM_NBUCKET_TOTAL) {
-        offsets[b] = (uint32_t)(n * MSM_NWIN);
+        offsets[b] = (uint32_t)total;
         return;
     }
-    uint32_t w = b >> MSM_C;
-    uint16_t d = (uint16_t)(b & MSM_DMASK);
-    const uint16_t *seg = sorted_keys + (size_t)w * n;
-    size_t lo = 0, hi = n;
+    size_t lo = 0, hi = total;
     while (lo < hi) {
         size_t mid = (lo + hi) >> 1;
-        if (seg[mid] < d)
+        if (sorted_keys[mid] < b)
             lo = mid + 1;
         else
             hi = mid;
     }
-    offsets[b] = (uint32_t)((size_t)w * n + lo);
+    offsets[b] = (uint32_t)lo;
 }
 
 // ---- bucket accumulation (the hot kernel) ----
