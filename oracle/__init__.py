@@ -113,3 +113,66 @@ def fr_mulmod(a: bytes, b: bytes) -> bytes:
 
 def num_threads() -> int:
     return _lib.oracle_num_threads()
+
+
+# ---- BLS12-381 G1 (semantics: bls_blst.rs) ----
+
+_lib.oracle_bls_g1_add.restype = ctypes.c_int
+_lib.oracle_bls_g1_mul.restype = ctypes.c_int
+_lib.oracle_bls_g1_msm.restype = ctypes.c_int
+_lib.oracle_bls_g1_msm_naive.restype = ctypes.c_int
+_lib.oracle_bls_g1_msm_jacobian.restype = ctypes.c_int
+_lib.oracle_bls_g1_combine_jacobian.restype = ctypes.c_int
+_lib.oracle_bls_gen_points.restype = ctypes.c_int
+
+
+def bls_g1_add(p1: bytes, p2: bytes):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.oracle_bls_g1_add(_buf(p1), _buf(p2), out)
+    return rc, bytes(out)
+
+
+def bls_g1_mul(point: bytes, scalar: bytes):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.oracle_bls_g1_mul(_buf(point), _buf(scalar), out)
+    return rc, bytes(out)
+
+
+def bls_g1_msm(points: bytes, scalars: bytes, n: int):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.oracle_bls_g1_msm(_buf(points), _buf(scalars), ctypes.c_size_t(n), out)
+    return rc, bytes(out)
+
+
+def bls_g1_msm_naive(points: bytes, scalars: bytes, n: int):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.oracle_bls_g1_msm_naive(_buf(points), _buf(scalars),
+                                      ctypes.c_size_t(n), out)
+    return rc, bytes(out)
+
+
+def bls_g1_msm_jacobian(points: bytes, scalars: bytes, n: int):
+    out = (ctypes.c_uint8 * 144)()
+    rc = _lib.oracle_bls_g1_msm_jacobian(_buf(points), _buf(scalars),
+                                         ctypes.c_size_t(n), out)
+    return rc, bytes(out)
+
+
+def bls_g1_combine_jacobian(jacobians: bytes, g: int):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.oracle_bls_g1_combine_jacobian(_buf(jacobians),
+                                             ctypes.c_size_t(g), out)
+    return rc, bytes(out)
+
+
+def bls_gen_points(start: int, n: int) -> bytes:
+    out = (ctypes.c_uint8 * (96 * n))()
+    rc = _lib.oracle_bls_gen_points(ctypes.c_uint64(start), ctypes.c_size_t(n), out)
+    assert rc == 0
+    return bytes(out)
+
+
+def bls_gen_fr(seed: int, n: int) -> bytes:
+    out = (ctypes.c_uint8 * (32 * n))()
+    _lib.oracle_bls_gen_fr(ctypes.c_uint64(seed), ctypes.c_size_t(n), out)
+    return bytes(out)
