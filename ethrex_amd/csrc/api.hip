@@ -15,6 +15,7 @@
 #include <cstdio>
 #include <cstring>
 #include <string>
+#include <type_traits>
 
 #include <rocprim/rocprim.hpp>
 
@@ -72,14 +73,23 @@ static inline uint32_t blocks_for(size_t n, int bs) {
     return (uint32_t)((n + bs - 1) / bs);
 }
 
-// ============================ MSM plan ============================
+// ============================ MSM plans ============================
+// One templated plan implementation serves both curves:
+//   Bn254G1: 64-B points, ark from_be_bytes_mod_order scalar reduction
+//   BlsG1:   96-B points (canonical + subgroup-checked), raw 256-bit scalars
+// The C ABI exposes separate opaque types (em_msm_plan / em_bls_msm_plan).
 
-struct em_msm_plan {
+template <typename C>
+struct msm_plan_t {
+    using F = typename C::F;
+    static constexpr int PB = 2 * F::W64 * 8;  // affine point bytes
+    static constexpr int JB = 3 * F::W64 * 8;  // Jacobian partial bytes
+    static constexpr int AB = PB;              // affine out bytes
     size_t n;
-    g1a9 *d_pts = nullptr;
+    g1aT<C> *d_pts = nullptr;
     uint8_t *d_inf = nullptr;
     fe4 *d_scalars = nullptr;
-    uint8_t *d_scratch = nullptr;     // 64n bytes: point/scalar byte staging
+    uint8_t *d_scratch = nullptr;     // PB*n bytes: point/scalar byte staging
     uint32_t *d_keys = nullptr;       // 16n
     uint32_t *d_vals = nullptr;
     uint32_t *d_keys_out = nullptr;
@@ -87,12 +97,12 @@ struct em_msm_plan {
     void *d_sort_tmp = nullptr;
     size_t sort_tmp_bytes = 0;
     uint32_t *d_offsets = nullptr;    // NBUCKET_TOTAL + 1
-    g1j9 *d_buckets = nullptr;         // NBUCKET_TOTAL
-    g1j9 *d_seg_sum = nullptr;         // NWIN*NSEG
-    g1j9 *d_seg_wsum = nullptr;
-    g1j9 *d_partials = nullptr;        // NWIN*NBLK_PER_WIN
-    g1j9 *d_windows = nullptr;         // NWIN
-    uint8_t *d_out = nullptr;         // 96 B
+    g1jT<C> *d_buckets = nullptr;     // NBUCKET_TOTAL
+    g1jT<C> *d_seg_sum = nullptr;     // NWIN*NSEG
+    g1jT<C> *d_seg_wsum = nullptr;
+    g1jT<C> *d_partials = nullptr;    // NWIN*NBLK_PER_WIN
+    g1jT<C> *d_windows = nullptr;     // NWIN
+    uint8_t *d_out = nullptr;         // JB
     uint32_t *d_err = nullptr;
     bool have_scalars = false;
     bool have_points = false;
@@ -100,50 +110,8 @@ struct em_msm_plan {
     double last_ms[5] = {0, 0, 0, 0, 0};
 };
 
-extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
-    if (!plan || n == 0) return EM_ERR_INPUT;
-    int rc = require_gpu();
-    if (rc) return rc;
-    em_msm_plan *p = new em_msm_plan();
-    p->n = n;
-    size_t total = n * MSM_NWIN;
-    hipError_t e = hipSuccess;
-    auto mal = [&](void **ptr, size_t bytes) {
-        if (e == hipSuccess) e = hipMalloc(ptr, bytes);
-    };
-    mal((void **)&p->d_pts, n * sizeof(g1a9));
-    mal((void **)&p->d_inf, n);
-    mal((void **)&p->d_scalars, n * sizeof(fe4));
-    mal((void **)&p->d_scratch, n * 64);
-    mal((void **)&p->d_keys, total * 4);
-    mal((void **)&p->d_vals, total * 4);
-    mal((void **)&p->d_keys_out, total * 4);
-    mal((void **)&p->d_vals_out, total * 4);
-    mal((void **)&p->d_offsets, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
-    mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1j9));
-    mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1j9));
-    mal((void **)&p->d_seg_wsum, MSM_NWIN * MSM_NSEG * sizeof(g1j9));
-    mal((void **)&p->d_partials, MSM_NWIN * MSM_NBLK_PER_WIN * sizeof(g1j9));
-    mal((void **)&p->d_windows, MSM_NWIN * sizeof(g1j9));
-    mal((void **)&p->d_out, 96);
-    mal((void **)&p->d_err, 4);
-    if (e == hipSuccess) {
-        // rocPRIM temp-storage size query
-        e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
-                                      p->d_keys_out, p->d_vals, p->d_vals_out,
-                                      total, 0, MSM_SORT_BITS);
-        if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
-    }
-    for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
-    if (e != hipSuccess) {
-        ethrex_mi355_msm_plan_destroy(p);
-        return hip_fail(e, "msm_plan_create");
-    }
-    *plan = p;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
+template <typename C>
+static int msm_destroy_t(msm_plan_t<C> *p) {
     if (!p) return EM_ERR_INPUT;
     hipFree(p->d_pts);
     hipFree(p->d_inf);
@@ -166,49 +134,131 @@ extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
     return EM_OK;
 }
 
-extern "C" int ethrex_mi355_msm_upload_points(em_msm_plan *p,
-                                              const uint8_t *points64) {
-    if (!p || !points64) return EM_ERR_INPUT;
+template <typename C>
+static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
+    if (!plan || n == 0) return EM_ERR_INPUT;
+    int rc = require_gpu();
+    if (rc) return rc;
+    auto *p = new msm_plan_t<C>();
+    p->n = n;
+    size_t total = n * MSM_NWIN;
+    hipError_t e = hipSuccess;
+    auto mal = [&](void **ptr, size_t bytes) {
+        if (e == hipSuccess) e = hipMalloc(ptr, bytes);
+    };
+    mal((void **)&p->d_pts, n * sizeof(g1aT<C>));
+    mal((void **)&p->d_inf, n);
+    mal((void **)&p->d_scalars, n * sizeof(fe4));
+    mal((void **)&p->d_scratch, n * (size_t)msm_plan_t<C>::PB);
+    mal((void **)&p->d_keys, total * 4);
+    mal((void **)&p->d_vals, total * 4);
+    mal((void **)&p->d_keys_out, total * 4);
+    mal((void **)&p->d_vals_out, total * 4);
+    mal((void **)&p->d_offsets, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
+    mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1jT<C>));
+    mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1jT<C>));
+    mal((void **)&p->d_seg_wsum, MSM_NWIN * MSM_NSEG * sizeof(g1jT<C>));
+    mal((void **)&p->d_partials, MSM_NWIN * MSM_NBLK_PER_WIN * sizeof(g1jT<C>));
+    mal((void **)&p->d_windows, MSM_NWIN * sizeof(g1jT<C>));
+    mal((void **)&p->d_out, msm_plan_t<C>::JB);
+    mal((void **)&p->d_err, 4);
+    if (e == hipSuccess) {
+        e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
+                                      p->d_keys_out, p->d_vals, p->d_vals_out,
+                                      total, 0, MSM_SORT_BITS);
+        if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
+    }
+    for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
+    if (e != hipSuccess) {
+        msm_destroy_t(p);
+        return hip_fail(e, "msm_plan_create");
+    }
+    *plan = p;
+    return EM_OK;
+}
+
+template <typename C>
+static int msm_upload_points_t(msm_plan_t<C> *p, const uint8_t *points) {
+    if (!p || !points) return EM_ERR_INPUT;
+    constexpr int PB = msm_plan_t<C>::PB;
     HIP_TRY(hipMemset(p->d_err, 0, 4));
-    HIP_TRY(hipMemcpy(p->d_scratch, points64, p->n * 64, hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_parse_points, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       p->d_scratch, p->d_pts, p->d_inf, p->n, p->d_err);
+    HIP_TRY(hipMemcpy(p->d_scratch, points, p->n * PB, hipMemcpyHostToDevice));
+    if constexpr (std::is_same_v<C, BlsG1>) {
+        hipLaunchKernelGGL(k_bls_parse_points, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_scratch, p->d_pts, p->d_inf,
+                           p->n, p->d_err);
+    } else {
+        hipLaunchKernelGGL(k_parse_points, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_scratch, p->d_pts, p->d_inf,
+                           p->n, p->d_err);
+    }
     uint32_t err = 0;
     HIP_TRY(hipMemcpy(&err, p->d_err, 4, hipMemcpyDeviceToHost));
+    if (err & 2u) {
+        g_last_err = "non-canonical coordinate";
+        return EM_ERR_INPUT;
+    }
+    if (err & 4u) {
+        g_last_err = "G1 point not in subgroup";
+        return EM_ERR_POINT;
+    }
     if (err) return EM_ERR_POINT;
     p->have_points = true;
     return EM_OK;
 }
 
-extern "C" int ethrex_mi355_msm_gen_points(em_msm_plan *p, uint64_t start) {
+template <typename C>
+static int msm_gen_points_t(msm_plan_t<C> *p, uint64_t start) {
     if (!p) return EM_ERR_INPUT;
-    hipLaunchKernelGGL(k_gen_points, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       p->d_pts, p->d_inf, p->n, start);
+    if constexpr (std::is_same_v<C, BlsG1>) {
+        hipLaunchKernelGGL(k_bls_gen_points, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_pts, p->d_inf, p->n, start);
+    } else {
+        hipLaunchKernelGGL(k_gen_points, dim3(blocks_for(p->n, 256)), dim3(256),
+                           0, 0, p->d_pts, p->d_inf, p->n, start);
+    }
     HIP_TRY(hipDeviceSynchronize());
     p->have_points = true;
     return EM_OK;
 }
 
-extern "C" int ethrex_mi355_msm_download_points(em_msm_plan *p, uint8_t *out64) {
-    if (!p || !out64 || !p->have_points) return EM_ERR_INPUT;
-    hipLaunchKernelGGL(k_points_to_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       p->d_pts, p->d_inf, p->d_scratch, p->n);
-    HIP_TRY(hipMemcpy(out64, p->d_scratch, p->n * 64, hipMemcpyDeviceToHost));
+template <typename C>
+static int msm_download_points_t(msm_plan_t<C> *p, uint8_t *out) {
+    if (!p || !out || !p->have_points) return EM_ERR_INPUT;
+    constexpr int PB = msm_plan_t<C>::PB;
+    if constexpr (std::is_same_v<C, BlsG1>) {
+        hipLaunchKernelGGL(k_bls_points_to_be, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_pts, p->d_inf, p->d_scratch,
+                           p->n);
+    } else {
+        hipLaunchKernelGGL(k_points_to_be, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_pts, p->d_inf, p->d_scratch,
+                           p->n);
+    }
+    HIP_TRY(hipMemcpy(out, p->d_scratch, p->n * PB, hipMemcpyDeviceToHost));
     return EM_OK;
 }
 
-extern "C" int ethrex_mi355_msm_upload_scalars(em_msm_plan *p,
-                                               const uint8_t *scalars32) {
+template <typename C>
+static int msm_upload_scalars_t(msm_plan_t<C> *p, const uint8_t *scalars32) {
     if (!p || !scalars32) return EM_ERR_INPUT;
     HIP_TRY(hipMemcpy(p->d_scratch, scalars32, p->n * 32, hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_parse_scalars, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       p->d_scratch, p->d_scalars, p->n);
+    if constexpr (std::is_same_v<C, BlsG1>) {
+        // raw 256-bit scalars, no reduction (blst SCALAR_BITS = 256)
+        hipLaunchKernelGGL(k_bls_parse_scalars, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_scratch, p->d_scalars, p->n);
+    } else {
+        // ark from_be_bytes_mod_order reduction
+        hipLaunchKernelGGL(k_parse_scalars, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_scratch, p->d_scalars, p->n);
+    }
     HIP_TRY(hipDeviceSynchronize());
     p->have_scalars = true;
     return EM_OK;
 }
 
-static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
+template <typename C>
+static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     if (!p || !out) return EM_ERR_INPUT;
     if (!p->have_points || !p->have_scalars) {
         g_last_err = "msm_run: points/scalars not uploaded";
@@ -216,39 +266,36 @@ static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
     }
     size_t total = p->n * MSM_NWIN;
     HIP_TRY(hipEventRecord(p->ev[0], 0));
-    // digits
     hipLaunchKernelGGL(k_digits, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
                        p->d_scalars, p->d_inf, p->d_keys, p->d_vals, p->n);
-    // sort on 20 key bits
     size_t tmp = p->sort_tmp_bytes;
     hipError_t e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
                                              p->d_keys_out, p->d_vals,
-                                             p->d_vals_out, total, 0, MSM_SORT_BITS);
+                                             p->d_vals_out, total, 0,
+                                             MSM_SORT_BITS);
     if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs");
-    // offsets
     hipLaunchKernelGGL(k_offsets,
                        dim3(blocks_for((size_t)MSM_NBUCKET_TOTAL + 1, 256)),
                        dim3(256), 0, 0, p->d_keys_out, total, p->d_offsets);
     HIP_TRY(hipEventRecord(p->ev[1], 0));
-    // bucket accumulation (hot)
-    hipLaunchKernelGGL(k_bucket_acc, dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)),
+    hipLaunchKernelGGL(k_bucket_acc<C>, dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)),
                        dim3(256), 0, 0, p->d_pts, p->d_vals_out, p->d_offsets,
                        p->d_buckets);
     HIP_TRY(hipEventRecord(p->ev[2], 0));
-    // reductions
-    hipLaunchKernelGGL(k_segment_reduce,
-                       dim3(blocks_for(MSM_NWIN * MSM_NSEG, 256)), dim3(256), 0, 0,
-                       p->d_buckets, p->d_seg_sum, p->d_seg_wsum);
-    hipLaunchKernelGGL(k_weighted_reduce,
+    hipLaunchKernelGGL(k_segment_reduce<C>,
+                       dim3(blocks_for(MSM_NWIN * MSM_NSEG, 256)), dim3(256), 0,
+                       0, p->d_buckets, p->d_seg_sum, p->d_seg_wsum);
+    hipLaunchKernelGGL(k_weighted_reduce<C>,
                        dim3(MSM_NWIN * MSM_NBLK_PER_WIN), dim3(MSM_RED_BLOCK), 0,
                        0, p->d_seg_sum, p->d_seg_wsum, p->d_partials);
-    hipLaunchKernelGGL(k_window_sum, dim3(1), dim3(64), 0, 0, p->d_partials,
+    hipLaunchKernelGGL(k_window_sum<C>, dim3(1), dim3(64), 0, 0, p->d_partials,
                        p->d_windows);
     HIP_TRY(hipEventRecord(p->ev[3], 0));
-    hipLaunchKernelGGL(k_final_combine, dim3(1), dim3(64), 0, 0, p->d_windows,
+    hipLaunchKernelGGL(k_final_combine<C>, dim3(1), dim3(64), 0, 0, p->d_windows,
                        p->d_out, out_mode);
     HIP_TRY(hipEventRecord(p->ev[4], 0));
-    HIP_TRY(hipMemcpy(out, p->d_out, out_mode == 0 ? 64 : 96,
+    HIP_TRY(hipMemcpy(out, p->d_out,
+                      out_mode == 0 ? msm_plan_t<C>::AB : msm_plan_t<C>::JB,
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipDeviceSynchronize());
     float ms;
@@ -265,21 +312,79 @@ static int msm_run_inner(em_msm_plan *p, uint8_t *out, int out_mode) {
     return EM_OK;
 }
 
+// opaque ABI types
+struct em_msm_plan : msm_plan_t<Bn254G1> {};
+struct em_bls_msm_plan : msm_plan_t<BlsG1> {};
+
+extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
+    return msm_create_t(n, (msm_plan_t<Bn254G1> **)plan);
+}
+extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
+    return msm_destroy_t((msm_plan_t<Bn254G1> *)p);
+}
+extern "C" int ethrex_mi355_msm_upload_points(em_msm_plan *p,
+                                              const uint8_t *points64) {
+    return msm_upload_points_t((msm_plan_t<Bn254G1> *)p, points64);
+}
+extern "C" int ethrex_mi355_msm_gen_points(em_msm_plan *p, uint64_t start) {
+    return msm_gen_points_t((msm_plan_t<Bn254G1> *)p, start);
+}
+extern "C" int ethrex_mi355_msm_download_points(em_msm_plan *p, uint8_t *out64) {
+    return msm_download_points_t((msm_plan_t<Bn254G1> *)p, out64);
+}
+extern "C" int ethrex_mi355_msm_upload_scalars(em_msm_plan *p,
+                                               const uint8_t *scalars32) {
+    return msm_upload_scalars_t((msm_plan_t<Bn254G1> *)p, scalars32);
+}
 extern "C" int ethrex_mi355_msm_run(em_msm_plan *p, uint8_t out[64]) {
-    return msm_run_inner(p, out, 0);
+    return msm_run_inner_t((msm_plan_t<Bn254G1> *)p, out, 0);
 }
-
 extern "C" int ethrex_mi355_msm_run_partial(em_msm_plan *p, uint8_t out[96]) {
-    return msm_run_inner(p, out, 1);
+    return msm_run_inner_t((msm_plan_t<Bn254G1> *)p, out, 1);
 }
-
 extern "C" int ethrex_mi355_msm_last_times(em_msm_plan *p, double times_ms[5]) {
     if (!p || !times_ms) return EM_ERR_INPUT;
     memcpy(times_ms, p->last_ms, sizeof p->last_ms);
     return EM_OK;
 }
 
-// ============================ one-shot MSM ============================
+extern "C" int ethrex_mi355_bls_msm_plan_create(size_t n, em_bls_msm_plan **plan) {
+    return msm_create_t(n, (msm_plan_t<BlsG1> **)plan);
+}
+extern "C" int ethrex_mi355_bls_msm_plan_destroy(em_bls_msm_plan *p) {
+    return msm_destroy_t((msm_plan_t<BlsG1> *)p);
+}
+extern "C" int ethrex_mi355_bls_msm_upload_points(em_bls_msm_plan *p,
+                                                  const uint8_t *points96) {
+    return msm_upload_points_t((msm_plan_t<BlsG1> *)p, points96);
+}
+extern "C" int ethrex_mi355_bls_msm_gen_points(em_bls_msm_plan *p,
+                                               uint64_t start) {
+    return msm_gen_points_t((msm_plan_t<BlsG1> *)p, start);
+}
+extern "C" int ethrex_mi355_bls_msm_download_points(em_bls_msm_plan *p,
+                                                    uint8_t *out96) {
+    return msm_download_points_t((msm_plan_t<BlsG1> *)p, out96);
+}
+extern "C" int ethrex_mi355_bls_msm_upload_scalars(em_bls_msm_plan *p,
+                                                   const uint8_t *scalars32) {
+    return msm_upload_scalars_t((msm_plan_t<BlsG1> *)p, scalars32);
+}
+extern "C" int ethrex_mi355_bls_msm_run(em_bls_msm_plan *p, uint8_t out[96]) {
+    return msm_run_inner_t((msm_plan_t<BlsG1> *)p, out, 0);
+}
+extern "C" int ethrex_mi355_bls_msm_run_partial(em_bls_msm_plan *p,
+                                                uint8_t out[144]) {
+    return msm_run_inner_t((msm_plan_t<BlsG1> *)p, out, 1);
+}
+extern "C" int ethrex_mi355_bls_msm_last_times(em_bls_msm_plan *p,
+                                               double times_ms[5]) {
+    if (!p || !times_ms) return EM_ERR_INPUT;
+    memcpy(times_ms, p->last_ms, sizeof p->last_ms);
+    return EM_OK;
+}
+
+// ============================ one-shot MSMs ============================
 
 extern "C" int ethrex_mi355_bn254_g1_msm(const uint8_t *points64,
                                          const uint8_t *scalars32, size_t n,
@@ -295,53 +400,72 @@ extern "C" int ethrex_mi355_bn254_g1_msm(const uint8_t *points64,
     return rc;
 }
 
+extern "C" int ethrex_mi355_bls12381_g1_msm(const uint8_t *points96,
+                                            const uint8_t *scalars32, size_t n,
+                                            uint8_t out[96]) {
+    if (!points96 || !scalars32 || !out || n == 0) return EM_ERR_INPUT;
+    em_bls_msm_plan *p = nullptr;
+    int rc = ethrex_mi355_bls_msm_plan_create(n, &p);
+    if (rc) return rc;
+    rc = ethrex_mi355_bls_msm_upload_points(p, points96);
+    if (!rc) rc = ethrex_mi355_bls_msm_upload_scalars(p, scalars32);
+    if (!rc) rc = ethrex_mi355_bls_msm_run(p, out);
+    ethrex_mi355_bls_msm_plan_destroy(p);
+    return rc;
+}
+
 // ============================ single ops ============================
 
-extern "C" int ethrex_mi355_bn254_g1_add(const uint8_t p1[64],
-                                         const uint8_t p2[64], uint8_t out[64]) {
-    if (!p1 || !p2 || !out) return EM_ERR_INPUT;
+// generic 1-thread op runner: in_bytes staged, out_bytes copied back
+template <typename K>
+static int run_single(K kern, const uint8_t *a, size_t la, const uint8_t *b,
+                      size_t lb, uint8_t *out, size_t lo) {
     int rc = require_gpu();
     if (rc) return rc;
     uint8_t *d_in, *d_out;
     uint32_t *d_err;
-    HIP_TRY(hipMalloc(&d_in, 128));
-    HIP_TRY(hipMalloc(&d_out, 64));
+    HIP_TRY(hipMalloc(&d_in, la + lb));
+    HIP_TRY(hipMalloc(&d_out, lo));
     HIP_TRY(hipMalloc(&d_err, 4));
     HIP_TRY(hipMemset(d_err, 0, 4));
-    HIP_TRY(hipMemcpy(d_in, p1, 64, hipMemcpyHostToDevice));
-    HIP_TRY(hipMemcpy(d_in + 64, p2, 64, hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_g1_add_single, dim3(1), dim3(64), 0, 0, d_in, d_out, d_err);
+    HIP_TRY(hipMemcpy(d_in, a, la, hipMemcpyHostToDevice));
+    if (lb) HIP_TRY(hipMemcpy(d_in + la, b, lb, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(kern, dim3(1), dim3(64), 0, 0, d_in, d_out, d_err);
     uint32_t err;
     HIP_TRY(hipMemcpy(&err, d_err, 4, hipMemcpyDeviceToHost));
-    if (!err) HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
+    if (!err) HIP_TRY(hipMemcpy(out, d_out, lo, hipMemcpyDeviceToHost));
     hipFree(d_in);
     hipFree(d_out);
     hipFree(d_err);
+    if (err & 2u) return EM_ERR_INPUT;
     return err ? EM_ERR_POINT : EM_OK;
+}
+
+extern "C" int ethrex_mi355_bn254_g1_add(const uint8_t p1[64],
+                                         const uint8_t p2[64], uint8_t out[64]) {
+    if (!p1 || !p2 || !out) return EM_ERR_INPUT;
+    return run_single(k_g1_add_single, p1, 64, p2, 64, out, 64);
 }
 
 extern "C" int ethrex_mi355_bn254_g1_mul(const uint8_t point[64],
                                          const uint8_t scalar[32],
                                          uint8_t out[64]) {
     if (!point || !scalar || !out) return EM_ERR_INPUT;
-    int rc = require_gpu();
-    if (rc) return rc;
-    uint8_t *d_in, *d_out;
-    uint32_t *d_err;
-    HIP_TRY(hipMalloc(&d_in, 96));
-    HIP_TRY(hipMalloc(&d_out, 64));
-    HIP_TRY(hipMalloc(&d_err, 4));
-    HIP_TRY(hipMemset(d_err, 0, 4));
-    HIP_TRY(hipMemcpy(d_in, point, 64, hipMemcpyHostToDevice));
-    HIP_TRY(hipMemcpy(d_in + 64, scalar, 32, hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_g1_mul_single, dim3(1), dim3(64), 0, 0, d_in, d_out, d_err);
-    uint32_t err;
-    HIP_TRY(hipMemcpy(&err, d_err, 4, hipMemcpyDeviceToHost));
-    if (!err) HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
-    hipFree(d_in);
-    hipFree(d_out);
-    hipFree(d_err);
-    return err ? EM_ERR_POINT : EM_OK;
+    return run_single(k_g1_mul_single, point, 64, scalar, 32, out, 64);
+}
+
+extern "C" int ethrex_mi355_bls12381_g1_add(const uint8_t p1[96],
+                                            const uint8_t p2[96],
+                                            uint8_t out[96]) {
+    if (!p1 || !p2 || !out) return EM_ERR_INPUT;
+    return run_single(k_bls_g1_add_single, p1, 96, p2, 96, out, 96);
+}
+
+extern "C" int ethrex_mi355_bls12381_g1_mul(const uint8_t point[96],
+                                            const uint8_t scalar[32],
+                                            uint8_t out[96]) {
+    if (!point || !scalar || !out) return EM_ERR_INPUT;
+    return run_single(k_bls_g1_mul_single, point, 96, scalar, 32, out, 96);
 }
 
 extern "C" int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96,
@@ -353,8 +477,26 @@ extern "C" int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96,
     HIP_TRY(hipMalloc(&d_in, 96 * count));
     HIP_TRY(hipMalloc(&d_out, 64));
     HIP_TRY(hipMemcpy(d_in, jacobians96, 96 * count, hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_g1_combine, dim3(1), dim3(64), 0, 0, d_in, count, d_out);
+    hipLaunchKernelGGL(k_g1_combine<Bn254G1>, dim3(1), dim3(64), 0, 0, d_in,
+                       count, d_out);
     HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
+    hipFree(d_in);
+    hipFree(d_out);
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_bls12381_g1_combine(const uint8_t *jacobians144,
+                                                size_t count, uint8_t out[96]) {
+    if (!jacobians144 || !out || count == 0) return EM_ERR_INPUT;
+    int rc = require_gpu();
+    if (rc) return rc;
+    uint8_t *d_in, *d_out;
+    HIP_TRY(hipMalloc(&d_in, 144 * count));
+    HIP_TRY(hipMalloc(&d_out, 96));
+    HIP_TRY(hipMemcpy(d_in, jacobians144, 144 * count, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_g1_combine<BlsG1>, dim3(1), dim3(64), 0, 0, d_in,
+                       count, d_out);
+    HIP_TRY(hipMemcpy(out, d_out, 96, hipMemcpyDeviceToHost));
     hipFree(d_in);
     hipFree(d_out);
     return EM_OK;
@@ -634,6 +776,25 @@ extern "C" void ethrex_mi355_gen_fr(uint64_t seed, size_t n, uint8_t *out) {
             s.v[1] = xosh_next(g);
             s.v[2] = xosh_next(g);
             s.v[3] = xosh_next(g) & 0x3fffffffffffffffull;
+        } while (fe_geq(s, rmod));
+        fe_to_be(out + 32 * i, s);
+    }
+}
+
+// same scheme for BLS12-381 Fr (255-bit mask, reject >= r_bls)
+extern "C" void ethrex_mi355_bls_gen_fr(uint64_t seed, size_t n, uint8_t *out) {
+    Xosh g;
+    uint64_t sm = seed;
+    for (int i = 0; i < 4; i++) g.s[i] = splitmix64_next(sm);
+    const fe4 rmod{{bn254::FrB::MOD[0], bn254::FrB::MOD[1], bn254::FrB::MOD[2],
+                    bn254::FrB::MOD[3]}};
+    for (size_t i = 0; i < n; i++) {
+        fe4 s;
+        do {
+            s.v[0] = xosh_next(g);
+            s.v[1] = xosh_next(g);
+            s.v[2] = xosh_next(g);
+            s.v[3] = xosh_next(g) & 0x7fffffffffffffffull;
         } while (fe_geq(s, rmod));
         fe_to_be(out + 32 * i, s);
     }

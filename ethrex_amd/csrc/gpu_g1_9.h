@@ -1,193 +1,254 @@
 // ============================================================================
-// BN254 G1 on the fe9 (9x29-bit) field core — XYZZ coordinates, Montgomery
-// 2^261.  XYZZ (x = X/ZZ, y = Y/ZZZ with ZZ = z^2, ZZZ = z^3) gives the
-// cheapest mixed add on this field core: 10 muls vs 11 for Jacobian
-// (EFD madd-2008-s / add-2008-s / dbl-2008-s shapes, a = 0).
+// Short-Weierstrass G1 (y^2 = x^3 + b) on the 29-bit-limb field core —
+// XYZZ coordinates (x = X/ZZ, y = Y/ZZZ), cheapest mixed add here:
+// 10 muls (EFD madd-2008-s / add-2008-s / dbl-2008-s, a = 0).
 //
-// Reference semantics unchanged (provider.rs:247-318; (0,0) identity);
-// byte-level results identical.  Bound audit per gpu_field9.h contracts is
-// annotated at each sub call site.  infinity <=> ZZ ≡ 0 (mod p).
+// Curve instances:
+//   Bn254G1 (Fq9T,  b=3, G=(1,2))          — provider.rs:247-318 semantics
+//   BlsG1   (FpB14T, b=4, standard G)      — bls_blst.rs EIP-2537 semantics
+// Byte-level results identical to the respective references; bound audit per
+// gpu_field9.h contracts annotated at each sub call site.
+// infinity <=> ZZ ≡ 0 (mod p).
+//
+// NOTE (L=14 column budget): lazy add9 results must NOT feed muls on the
+// BLS field — the doubling formula below therefore uses add9_n for U.
 // ============================================================================
 #pragma once
 #include "gpu_field9.h"
 
 namespace em {
 
-struct g1a9 {
-    fe9 x, y;  // affine, Montgomery form, norm2p
+struct Bn254G1 {
+    using F = Fq9T;
+    static constexpr const u32 (&CB)[9] = bn254::FQ9_B3;   // b in Montgomery
+    static constexpr const u32 (&GX)[9] = bn254::FQ9_GX;
+    static constexpr const u32 (&GY)[9] = bn254::FQ9_GY;
+};
+struct BlsG1 {
+    using F = FpB14T;
+    static constexpr const u32 (&CB)[14] = bn254::FPB_B4;
+    static constexpr const u32 (&GX)[14] = bn254::FPB_GX;
+    static constexpr const u32 (&GY)[14] = bn254::FPB_GY;
 };
 
-struct g1j9 {               // name kept for kernel compatibility; XYZZ layout
-    fe9 x, y, zz, zzz;
+template <typename C>
+struct g1aT {
+    feL<C::F::L> x, y;  // affine, Montgomery form, norm2p
 };
 
-__device__ __forceinline__ g1j9 g1_inf9() {
-    g1j9 p;
-    p.x = fe9_load(bn254::FQ9_ONE);
-    p.y = fe9_load(bn254::FQ9_ONE);
-    p.zz = fe9_zero();
-    p.zzz = fe9_zero();
+template <typename C>
+struct g1jT {
+    feL<C::F::L> x, y, zz, zzz;  // XYZZ
+};
+
+using g1a9 = g1aT<Bn254G1>;
+using g1j9 = g1jT<Bn254G1>;
+using g1aB = g1aT<BlsG1>;
+using g1jB = g1jT<BlsG1>;
+
+template <typename C = Bn254G1>
+__device__ __forceinline__ g1jT<C> g1_inf9() {
+    g1jT<C> p;
+    p.x = fe9_load<C::F::L>(C::F::ONE);
+    p.y = fe9_load<C::F::L>(C::F::ONE);
+    p.zz = fe9z<C::F::L>();
+    p.zzz = fe9z<C::F::L>();
     return p;
 }
 
-__device__ __forceinline__ bool g1_is_inf9(const g1j9 &p) {
-    return fe9_is_zero_modp(p.zz);
+template <typename C>
+__device__ __forceinline__ bool g1_is_inf9(const g1jT<C> &p) {
+    return fe9_is_zero_modp<typename C::F>(p.zz);
 }
 
 // doubling (dbl-2008-s, a = 0)
-__device__ __forceinline__ g1j9 g1_dbl9(const g1j9 &p) {
+template <typename C>
+__device__ __forceinline__ g1jT<C> g1_dbl9(const g1jT<C> &p) {
+    using F = typename C::F;
     if (g1_is_inf9(p)) return p;
-    fe9 U = add9(p.y, p.y);                    // lazy <=2^30
-    fe9 V = mont_mul9(U, U);                   // (2Y)^2
-    fe9 W = mont_mul9(U, V);                   // (2Y)^3
-    fe9 S = mont_mul9(p.x, V);
-    fe9 A = mont_sqr9(p.x);
-    fe9 M = add9_n(add9(A, A), A);             // 3X^2, norm2p
-    g1j9 o;
-    o.x = subm9(subm9(mont_sqr9(M), S), S);    // M^2 - 2S
-    o.y = subm9(mont_mul9(M, subn9(S, o.x)), mont_mul9(W, p.y));
-    o.zz = mont_mul9(V, p.zz);
-    o.zzz = mont_mul9(W, p.zzz);
+    feL<F::L> U = add9_n<F>(p.y, p.y);         // 2Y, norm2p (L=14 mul rule)
+    feL<F::L> V = mont_mul9<F>(U, U);
+    feL<F::L> W = mont_mul9<F>(U, V);
+    feL<F::L> S = mont_mul9<F>(p.x, V);
+    feL<F::L> A = mont_sqr9<F>(p.x);
+    feL<F::L> M = add9_n<F>(add9_n<F>(A, A), A);  // 3X^2, norm2p
+    g1jT<C> o;
+    o.x = subm9<F>(subm9<F>(mont_sqr9<F>(M), S), S);
+    o.y = subm9<F>(mont_mul9<F>(M, subn9<F>(S, o.x)), mont_mul9<F>(W, p.y));
+    o.zz = mont_mul9<F>(V, p.zz);
+    o.zzz = mont_mul9<F>(W, p.zzz);
     return o;
 }
 
 // full XYZZ + XYZZ (add-2008-s)
-__device__ __forceinline__ g1j9 g1_add9(const g1j9 &p, const g1j9 &q) {
+template <typename C>
+__device__ __forceinline__ g1jT<C> g1_add9(const g1jT<C> &p, const g1jT<C> &q) {
+    using F = typename C::F;
     if (g1_is_inf9(p)) return q;
     if (g1_is_inf9(q)) return p;
-    fe9 u1 = mont_mul9(p.x, q.zz);
-    fe9 u2 = mont_mul9(q.x, p.zz);
-    fe9 s1 = mont_mul9(p.y, q.zzz);
-    fe9 s2 = mont_mul9(q.y, p.zzz);
-    fe9 P = subm9(u2, u1);                     // b=u1 mul-out
-    fe9 R = subm9(s2, s1);
-    if (__builtin_expect(fe9_is_zero_modp(P), 0)) {
-        if (fe9_is_zero_modp(R)) return g1_dbl9(p);
-        return g1_inf9();
+    feL<F::L> u1 = mont_mul9<F>(p.x, q.zz);
+    feL<F::L> u2 = mont_mul9<F>(q.x, p.zz);
+    feL<F::L> s1 = mont_mul9<F>(p.y, q.zzz);
+    feL<F::L> s2 = mont_mul9<F>(q.y, p.zzz);
+    feL<F::L> P = subm9<F>(u2, u1);            // b=u1 mul-out
+    feL<F::L> R = subm9<F>(s2, s1);
+    if (__builtin_expect(fe9_is_zero_modp<F>(P), 0)) {
+        if (fe9_is_zero_modp<F>(R)) return g1_dbl9(p);
+        return g1_inf9<C>();
     }
-    fe9 PP = mont_sqr9(P);
-    fe9 PPP = mont_mul9(P, PP);
-    fe9 Q = mont_mul9(u1, PP);
-    g1j9 o;
-    o.x = subm9(subm9(subm9(mont_sqr9(R), PPP), Q), Q);
-    o.y = subm9(mont_mul9(R, subn9(Q, o.x)), mont_mul9(s1, PPP));
-    o.zz = mont_mul9(mont_mul9(p.zz, q.zz), PP);
-    o.zzz = mont_mul9(mont_mul9(p.zzz, q.zzz), PPP);
+    feL<F::L> PP = mont_sqr9<F>(P);
+    feL<F::L> PPP = mont_mul9<F>(P, PP);
+    feL<F::L> Q = mont_mul9<F>(u1, PP);
+    g1jT<C> o;
+    o.x = subm9<F>(subm9<F>(subm9<F>(mont_sqr9<F>(R), PPP), Q), Q);
+    o.y = subm9<F>(mont_mul9<F>(R, subn9<F>(Q, o.x)), mont_mul9<F>(s1, PPP));
+    o.zz = mont_mul9<F>(mont_mul9<F>(p.zz, q.zz), PP);
+    o.zzz = mont_mul9<F>(mont_mul9<F>(p.zzz, q.zzz), PPP);
     return o;
 }
 
 // mixed add (madd-2008-s): q affine, not infinity
-__device__ __forceinline__ g1j9 g1_add_affine9(const g1j9 &p, const g1a9 &q) {
+template <typename C>
+__device__ __forceinline__ g1jT<C> g1_add_affine9(const g1jT<C> &p,
+                                                  const g1aT<C> &q) {
+    using F = typename C::F;
     if (__builtin_expect(g1_is_inf9(p), 0)) {
-        g1j9 o;
+        g1jT<C> o;
         o.x = q.x;
         o.y = q.y;
-        o.zz = fe9_load(bn254::FQ9_ONE);
-        o.zzz = fe9_load(bn254::FQ9_ONE);
+        o.zz = fe9_load<F::L>(F::ONE);
+        o.zzz = fe9_load<F::L>(F::ONE);
         return o;
     }
-    fe9 u2 = mont_mul9(q.x, p.zz);
-    fe9 s2 = mont_mul9(q.y, p.zzz);
-    fe9 P = subn9(u2, p.x);                    // b=X1 norm2p
-    fe9 R = subn9(s2, p.y);
-    if (__builtin_expect(fe9_is_zero_modp(P), 0)) {
-        if (fe9_is_zero_modp(R)) return g1_dbl9(p);
-        return g1_inf9();
+    feL<F::L> u2 = mont_mul9<F>(q.x, p.zz);
+    feL<F::L> s2 = mont_mul9<F>(q.y, p.zzz);
+    feL<F::L> P = subn9<F>(u2, p.x);           // b=X1 norm2p
+    feL<F::L> R = subn9<F>(s2, p.y);
+    if (__builtin_expect(fe9_is_zero_modp<F>(P), 0)) {
+        if (fe9_is_zero_modp<F>(R)) return g1_dbl9(p);
+        return g1_inf9<C>();
     }
-    fe9 PP = mont_sqr9(P);
-    fe9 PPP = mont_mul9(P, PP);
-    fe9 Q = mont_mul9(p.x, PP);
-    g1j9 o;
-    o.x = subm9(subm9(subm9(mont_sqr9(R), PPP), Q), Q);
-    o.y = subm9(mont_mul9(R, subn9(Q, o.x)), mont_mul9(p.y, PPP));
-    o.zz = mont_mul9(p.zz, PP);
-    o.zzz = mont_mul9(p.zzz, PPP);
+    feL<F::L> PP = mont_sqr9<F>(P);
+    feL<F::L> PPP = mont_mul9<F>(P, PP);
+    feL<F::L> Q = mont_mul9<F>(p.x, PP);
+    g1jT<C> o;
+    o.x = subm9<F>(subm9<F>(subm9<F>(mont_sqr9<F>(R), PPP), Q), Q);
+    o.y = subm9<F>(mont_mul9<F>(R, subn9<F>(Q, o.x)), mont_mul9<F>(p.y, PPP));
+    o.zz = mont_mul9<F>(p.zz, PP);
+    o.zzz = mont_mul9<F>(p.zzz, PPP);
     return o;
 }
 
-// y^2 == x^3 + 3 (mod p), inputs norm2p
-__device__ __forceinline__ bool g1a9_on_curve(const g1a9 &p) {
-    fe9 l = mont_sqr9(p.y);
-    fe9 r = mont_mul9(mont_sqr9(p.x), p.x);
-    r = add9_n(r, fe9_load(bn254::FQ9_B3));
-    return fe9_eq_modp(l, r);
+// y^2 == x^3 + b (mod p), inputs norm2p
+template <typename C>
+__device__ __forceinline__ bool g1a9_on_curve(const g1aT<C> &p) {
+    using F = typename C::F;
+    feL<F::L> l = mont_sqr9<F>(p.y);
+    feL<F::L> r = mont_mul9<F>(mont_sqr9<F>(p.x), p.x);
+    r = add9_n<F>(r, fe9_load<F::L>(C::CB));
+    return fe9_eq_modp<F>(l, r);
 }
 
-__device__ __forceinline__ g1a9 g1_generator9() {
-    g1a9 g;
-    g.x = fe9_load(bn254::FQ9_GX);
-    g.y = fe9_load(bn254::FQ9_GY);
+template <typename C = Bn254G1>
+__device__ __forceinline__ g1aT<C> g1_generator9() {
+    g1aT<C> g;
+    g.x = fe9_load<C::F::L>(C::GX);
+    g.y = fe9_load<C::F::L>(C::GY);
     return g;
 }
 
-// scalar mul, k canonical 4x64, p affine non-infinity
-__device__ __forceinline__ g1j9 g1_scalar_mul9(const g1a9 &p, const u64 k[4]) {
-    g1j9 acc = g1_inf9();
-    for (int i = 255; i >= 0; i--) {
+// scalar mul, k canonical u64 words (nwords*64 bits scanned), p affine
+template <typename C>
+__device__ __forceinline__ g1jT<C> g1_scalar_mul9(const g1aT<C> &p,
+                                                  const u64 *k, int nwords = 4) {
+    g1jT<C> acc = g1_inf9<C>();
+    for (int i = 64 * nwords - 1; i >= 0; i--) {
         acc = g1_dbl9(acc);
         if ((k[i >> 6] >> (i & 63)) & 1) acc = g1_add_affine9(acc, p);
     }
     return acc;
 }
 
-// big-endian byte output helpers (canonical form)
-__device__ __forceinline__ void fe9_to_be(uint8_t *b, const fe9 &canon) {
-    u64 w[4];
-    fe9_to_u64x4(w, canon);
+// big-endian byte IO of canonical field elements (T::W64 * 8 bytes)
+template <typename T>
+__device__ __forceinline__ void feT_to_be(uint8_t *b, const feL<T::L> &canon) {
+    u64 w[T::W64];
+    fe9_to_u64<T>(w, canon);
     u64 *o = (u64 *)b;
-    o[0] = __builtin_bswap64(w[3]);
-    o[1] = __builtin_bswap64(w[2]);
-    o[2] = __builtin_bswap64(w[1]);
-    o[3] = __builtin_bswap64(w[0]);
+#pragma unroll
+    for (int i = 0; i < T::W64; i++)
+        o[i] = __builtin_bswap64(w[T::W64 - 1 - i]);
 }
 
-__device__ __forceinline__ fe9 fe9_from_be(const uint8_t *b) {
+template <typename T>
+__device__ __forceinline__ feL<T::L> feT_from_be(const uint8_t *b) {
     const u64 *w = (const u64 *)b;
-    u64 v[4] = {__builtin_bswap64(w[3]), __builtin_bswap64(w[2]),
-                __builtin_bswap64(w[1]), __builtin_bswap64(w[0])};
-    return fe9_from_u64x4(v);
+    u64 v[T::W64];
+#pragma unroll
+    for (int i = 0; i < T::W64; i++)
+        v[i] = __builtin_bswap64(w[T::W64 - 1 - i]);
+    return fe9_from_u64<T>(v);
+}
+
+// legacy bn254 names
+__device__ __forceinline__ void fe9_to_be(uint8_t *b, const fe9 &canon) {
+    feT_to_be<Fq9T>(b, canon);
+}
+__device__ __forceinline__ fe9 fe9_from_be(const uint8_t *b) {
+    return feT_from_be<Fq9T>(b);
 }
 
 // XYZZ -> affine (x = X/ZZ, y = Y/ZZZ): one inversion + 3 muls
-__device__ __forceinline__ g1a9 g1_to_affine9(const g1j9 &p) {
-    fe9 t = mont_inv9(mont_mul9(p.zz, p.zzz));  // 1/(ZZ*ZZZ)
-    g1a9 a;
-    a.x = fe9_csub2p(mont_mul9(p.x, mont_mul9(t, p.zzz)));
-    a.y = fe9_csub2p(mont_mul9(p.y, mont_mul9(t, p.zz)));
+template <typename C>
+__device__ __forceinline__ g1aT<C> g1_to_affine9(const g1jT<C> &p) {
+    using F = typename C::F;
+    feL<F::L> t = mont_inv9<F>(mont_mul9<F>(p.zz, p.zzz));
+    g1aT<C> a;
+    a.x = fe9_csub2p<F>(mont_mul9<F>(p.x, mont_mul9<F>(t, p.zzz)));
+    a.y = fe9_csub2p<F>(mont_mul9<F>(p.y, mont_mul9<F>(t, p.zz)));
     return a;
 }
 
-// XYZZ -> affine 64-byte BE; infinity -> zeros
-__device__ __forceinline__ void g1_to_affine_be9(uint8_t *out, const g1j9 &p) {
+// XYZZ -> affine BE bytes (2 coords); infinity -> zeros
+template <typename C>
+__device__ __forceinline__ void g1_to_affine_be9(uint8_t *out, const g1jT<C> &p) {
+    using F = typename C::F;
+    constexpr int NB = F::W64 * 8;
     if (g1_is_inf9(p)) {
-        for (int i = 0; i < 8; i++) ((u64 *)out)[i] = 0;
+        for (int i = 0; i < 2 * F::W64; i++) ((u64 *)out)[i] = 0;
         return;
     }
-    g1a9 a = g1_to_affine9(p);
-    fe9_to_be(out, from_mont9(a.x));
-    fe9_to_be(out + 32, from_mont9(a.y));
+    g1aT<C> a = g1_to_affine9(p);
+    feT_to_be<F>(out, from_mont9<F>(a.x));
+    feT_to_be<F>(out + NB, from_mont9<F>(a.y));
 }
 
 // XYZZ -> Jacobian (X_j, Y_j, Z_j) with Z_j = ZZ*ZZZ (no inversion):
 //   X_j = x*Z_j^2 = X*ZZ*ZZZ^2,  Y_j = y*Z_j^3 = Y*ZZ^3*ZZZ^2
-__device__ __forceinline__ void g1_xyzz_to_jacobian9(fe9 &X, fe9 &Y, fe9 &Z,
-                                                     const g1j9 &p) {
-    fe9 zzz2 = mont_sqr9(p.zzz);
-    fe9 zz2 = mont_sqr9(p.zz);
-    X = mont_mul9(mont_mul9(p.x, p.zz), zzz2);
-    Y = mont_mul9(mont_mul9(p.y, mont_mul9(zz2, p.zz)), zzz2);
-    Z = mont_mul9(p.zz, p.zzz);
+template <typename C>
+__device__ __forceinline__ void g1_xyzz_to_jacobian9(feL<C::F::L> &X,
+                                                     feL<C::F::L> &Y,
+                                                     feL<C::F::L> &Z,
+                                                     const g1jT<C> &p) {
+    using F = typename C::F;
+    feL<F::L> zzz2 = mont_sqr9<F>(p.zzz);
+    feL<F::L> zz2 = mont_sqr9<F>(p.zz);
+    X = mont_mul9<F>(mont_mul9<F>(p.x, p.zz), zzz2);
+    Y = mont_mul9<F>(mont_mul9<F>(p.y, mont_mul9<F>(zz2, p.zz)), zzz2);
+    Z = mont_mul9<F>(p.zz, p.zzz);
 }
 
-// Jacobian (X, Y, Z) -> XYZZ: ZZ = Z^2, ZZZ = Z^3; adjust X,Y? no —
-// Jacobian x = X/Z^2 = X/ZZ, y = Y/Z^3 = Y/ZZZ: same numerators.
-__device__ __forceinline__ g1j9 g1_jacobian_to_xyzz9(const fe9 &X, const fe9 &Y,
-                                                     const fe9 &Z) {
-    g1j9 p;
+// Jacobian (X, Y, Z) -> XYZZ: same numerators, ZZ = Z^2, ZZZ = Z^3
+template <typename C>
+__device__ __forceinline__ g1jT<C> g1_jacobian_to_xyzz9(const feL<C::F::L> &X,
+                                                        const feL<C::F::L> &Y,
+                                                        const feL<C::F::L> &Z) {
+    using F = typename C::F;
+    g1jT<C> p;
     p.x = X;
     p.y = Y;
-    p.zz = mont_sqr9(Z);
-    p.zzz = mont_mul9(p.zz, Z);
+    p.zz = mont_sqr9<F>(Z);
+    p.zzz = mont_mul9<F>(p.zz, Z);
     return p;
 }
 

@@ -146,23 +146,24 @@ __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total
 
 // ---- bucket accumulation (the hot kernel) ----
 // one thread per bucket id; digit-0 buckets are skipped (never read later).
+template <typename C>
 __global__ void __launch_bounds__(256)
-k_bucket_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
-             const uint32_t *__restrict__ offsets, g1j9 *__restrict__ buckets) {
+k_bucket_acc(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
+             const uint32_t *__restrict__ offsets, g1jT<C> *__restrict__ buckets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= MSM_NBUCKET_TOTAL) return;
     if ((b & MSM_DMASK) == 0) return;  // digit 0
     uint32_t lo = offsets[b], hi = offsets[b + 1];
-    g1j9 acc = g1_inf9();
+    g1jT<C> acc = g1_inf9<C>();
     if (lo >= hi) {
         buckets[b] = acc;
         return;
     }
     // software pipeline: issue the NEXT point's gather before the long mixed
     // add so the dependent idx->point load chain overlaps the VALU work.
-    g1a9 p = pts[vals[lo]];
+    g1aT<C> p = pts[vals[lo]];
     for (uint32_t t = lo; t < hi; t++) {
-        g1a9 cur = p;
+        g1aT<C> cur = p;
         uint32_t nxt = t + 1 < hi ? t + 1 : t;
         p = pts[vals[nxt]];
         acc = g1_add_affine9(acc, cur);
@@ -174,19 +175,20 @@ k_bucket_acc(const g1a9 *__restrict__ pts, const uint32_t *__restrict__ vals,
 // level 1: per (window, 32-bucket segment): from the top digit down,
 //   run  += B_d           (=> run  = sum of segment buckets)
 //   wsum += run           (=> wsum = sum (d - lo + 1) * B_d)
+template <typename C>
 __global__ void __launch_bounds__(256)
-k_segment_reduce(const g1j9 *__restrict__ buckets,
-                 g1j9 *__restrict__ seg_sum,
-                 g1j9 *__restrict__ seg_wsum) {
+k_segment_reduce(const g1jT<C> *__restrict__ buckets,
+                 g1jT<C> *__restrict__ seg_sum,
+                 g1jT<C> *__restrict__ seg_wsum) {
     // the second accumulator lives in LDS: two register XYZZ accumulators
     // plus mul temporaries spill 232 B/lane to scratch otherwise
-    __shared__ g1j9 lds_wsum[256];
+    __shared__ g1jT<C> lds_wsum[256];
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;  // [0, 16*2048)
     if (t >= MSM_NWIN * MSM_NSEG) return;
     uint32_t w = t / MSM_NSEG, seg = t % MSM_NSEG;
     uint32_t lo = seg * MSM_SEG;
-    g1j9 run = g1_inf9();
-    lds_wsum[threadIdx.x] = g1_inf9();
+    g1jT<C> run = g1_inf9<C>();
+    lds_wsum[threadIdx.x] = g1_inf9<C>();
     for (int32_t d = (int32_t)lo + MSM_SEG - 1; d >= (int32_t)lo; d--) {
         // digit-0 bucket is unused: skip the add but keep the wsum step so
         // segment 0 carries the same (d - lo + 1) weights
@@ -201,23 +203,24 @@ k_segment_reduce(const g1j9 *__restrict__ buckets,
 // level 2: fully parallel weighted combine + LDS tree reduction.
 //   W_w = sum_j [ wsum_j + (j*SEG - 1) * sum_j ]   (j=0 term: -sum_0),
 // then scaled by 2^(16w) (the doubling chains run SIMD-wide here).
+template <typename C>
 __global__ void __launch_bounds__(MSM_RED_BLOCK)
-k_weighted_reduce(const g1j9 *__restrict__ seg_sum,
-                  const g1j9 *__restrict__ seg_wsum,
-                  g1j9 *__restrict__ partials /* NWIN*NBLK_PER_WIN */) {
-    __shared__ g1j9 lds[MSM_RED_BLOCK];
+k_weighted_reduce(const g1jT<C> *__restrict__ seg_sum,
+                  const g1jT<C> *__restrict__ seg_wsum,
+                  g1jT<C> *__restrict__ partials /* NWIN*NBLK_PER_WIN */) {
+    __shared__ g1jT<C> lds[MSM_RED_BLOCK];
     uint32_t t = blockIdx.x * MSM_RED_BLOCK + threadIdx.x;
     uint32_t j = t % MSM_NSEG;
-    g1j9 ws = seg_wsum[t];
-    g1j9 ss = seg_sum[t];
-    g1j9 val;
+    g1jT<C> ws = seg_wsum[t];
+    g1jT<C> ss = seg_sum[t];
+    g1jT<C> val;
     if (j == 0) {
         // weight -1: subtract sum_0
-        if (!g1_is_inf9(ss)) ss.y = neg9(ss.y);
+        if (!g1_is_inf9(ss)) ss.y = neg9<typename C::F>(ss.y);
         val = g1_add9(ws, ss);
     } else {
         uint32_t weight = j * MSM_SEG - 1;  // < 2^C
-        g1j9 acc = g1_inf9();
+        g1jT<C> acc = g1_inf9<C>();
         for (int b = MSM_C; b >= 0; b--) {
             acc = g1_dbl9(acc);
             if ((weight >> b) & 1) acc = g1_add9(acc, ss);
@@ -233,8 +236,8 @@ k_weighted_reduce(const g1j9 *__restrict__ seg_sum,
     __syncthreads();
     for (int s = MSM_RED_BLOCK / 2; s > 0; s >>= 1) {
         if (threadIdx.x < (uint32_t)s) {
-            g1j9 o = lds[threadIdx.x + s];
-            g1j9 m = g1_add9(lds[threadIdx.x], o);
+            g1jT<C> o = lds[threadIdx.x + s];
+            g1jT<C> m = g1_add9(lds[threadIdx.x], o);
             lds[threadIdx.x] = m;
         }
         __syncthreads();
@@ -243,11 +246,12 @@ k_weighted_reduce(const g1j9 *__restrict__ seg_sum,
 }
 
 // level 3: 16 threads, 8 partials each -> per-window sums (pre-scaled)
-__global__ void k_window_sum(const g1j9 *__restrict__ partials,
-                             g1j9 *__restrict__ windows) {
+template <typename C>
+__global__ void k_window_sum(const g1jT<C> *__restrict__ partials,
+                             g1jT<C> *__restrict__ windows) {
     uint32_t w = blockIdx.x * blockDim.x + threadIdx.x;
     if (w >= MSM_NWIN) return;
-    g1j9 acc = g1_inf9();
+    g1jT<C> acc = g1_inf9<C>();
     for (int b = 0; b < MSM_NBLK_PER_WIN; b++)
         acc = g1_add9(acc, partials[w * MSM_NBLK_PER_WIN + b]);
     windows[w] = acc;
@@ -256,24 +260,27 @@ __global__ void k_window_sum(const g1j9 *__restrict__ partials,
 // ---- final combine + output ----
 // out_mode 0: 64-byte BE affine (infinity -> zeros)
 // out_mode 1: 96-byte BE Jacobian canonical X||Y||Z (Z=0 -> infinity)
-__global__ void k_final_combine(const g1j9 *__restrict__ windows,
+template <typename C>
+__global__ void k_final_combine(const g1jT<C> *__restrict__ windows,
                                 uint8_t *__restrict__ out, int out_mode) {
+    using F = typename C::F;
+    constexpr int NB = F::W64 * 8;
     if (blockIdx.x != 0 || threadIdx.x != 0) return;
     // windows[] arrive pre-scaled by 2^(16w) (k_weighted_reduce)
-    g1j9 acc = windows[MSM_NWIN - 1];
+    g1jT<C> acc = windows[MSM_NWIN - 1];
     for (int w = MSM_NWIN - 2; w >= 0; w--) acc = g1_add9(acc, windows[w]);
     if (out_mode == 0) {
         g1_to_affine_be9(out, acc);
     } else {
-        // 96-B Jacobian exchange payload (ABI): convert XYZZ -> Jacobian
+        // Jacobian exchange payload (ABI): convert XYZZ -> Jacobian
         if (g1_is_inf9(acc)) {
-            for (int j = 0; j < 12; j++) ((u64 *)out)[j] = 0;
+            for (int j = 0; j < 3 * F::W64; j++) ((u64 *)out)[j] = 0;
         } else {
-            fe9 X, Y, Z;
+            feL<F::L> X, Y, Z;
             g1_xyzz_to_jacobian9(X, Y, Z, acc);
-            fe9_to_be(out, from_mont9(X));
-            fe9_to_be(out + 32, from_mont9(Y));
-            fe9_to_be(out + 64, from_mont9(Z));
+            feT_to_be<F>(out, from_mont9<F>(X));
+            feT_to_be<F>(out + NB, from_mont9<F>(Y));
+            feT_to_be<F>(out + 2 * NB, from_mont9<F>(Z));
         }
     }
 }
@@ -322,18 +329,152 @@ __global__ void k_g1_mul_single(const uint8_t *in /* 96 B: point||scalar */,
 }
 
 // combine count Jacobian partials (96-B BE canonical each) -> affine
+template <typename C>
 __global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
                              uint8_t *__restrict__ out) {
+    using F = typename C::F;
+    constexpr int NB = F::W64 * 8;
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
-    g1j9 acc = g1_inf9();
+    g1jT<C> acc = g1_inf9<C>();
     for (size_t i = 0; i < count; i++) {
-        fe9 X = to_mont9(fe9_from_be(in + 96 * i));
-        fe9 Y = to_mont9(fe9_from_be(in + 96 * i + 32));
-        fe9 Z = to_mont9(fe9_from_be(in + 96 * i + 64));
-        if (fe9_is_zero_modp(Z)) continue;  // Z=0 encodes infinity
-        acc = g1_add9(acc, g1_jacobian_to_xyzz9(X, Y, Z));
+        feL<F::L> X = to_mont9<F>(feT_from_be<F>(in + 3 * NB * i));
+        feL<F::L> Y = to_mont9<F>(feT_from_be<F>(in + 3 * NB * i + NB));
+        feL<F::L> Z = to_mont9<F>(feT_from_be<F>(in + 3 * NB * i + 2 * NB));
+        if (fe9_is_zero_modp<F>(Z)) continue;  // Z=0 encodes infinity
+        acc = g1_add9(acc, g1_jacobian_to_xyzz9<C>(X, Y, Z));
     }
     g1_to_affine_be9(out, acc);
+}
+
+// ============================================================================
+// BLS12-381 G1 specifics (bls_blst.rs EIP-2537
+// semantics: 48-B BE canonical coords, (0,0) infinity, on-curve check,
+// r-subgroup check for MSM inputs, scalars as FULL 256-bit integers).
+// ============================================================================
+
+// 96-byte BE affine -> Montgomery(2^406) g1aB + infinity flag.
+// err bits: 1 = non-canonical/off-curve/subgroup (details via last_error)
+__global__ void k_bls_parse_points(const uint8_t *__restrict__ in,
+                                   g1aB *__restrict__ pts,
+                                   uint8_t *__restrict__ inf, size_t n,
+                                   uint32_t *__restrict__ err) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    fe14 xr = feT_from_be<FpB14T>(in + 96 * i);
+    fe14 yr = feT_from_be<FpB14T>(in + 96 * i + 48);
+    if (fe9_geq_raw<14>(xr, bn254::FPB_P) || fe9_geq_raw<14>(yr, bn254::FPB_P)) {
+        atomicOr(err, 2u);  // non-canonical coordinate (EIP-2537 reject)
+        return;
+    }
+    if (fe9_is_zero_raw<14>(xr) && fe9_is_zero_raw<14>(yr)) {
+        inf[i] = 1;
+        pts[i].x = fe9z<14>();
+        pts[i].y = fe9z<14>();
+        return;
+    }
+    g1aB p{to_mont9<FpB14T>(xr), to_mont9<FpB14T>(yr)};
+    inf[i] = 0;
+    if (!g1a9_on_curve(p)) {
+        atomicOr(err, 1u);
+        return;
+    }
+    // subgroup check: r*P == infinity (read_g1_subgroup, bls_blst.rs:215-222)
+    g1jB t = g1_scalar_mul9(p, bn254::FRB_ORDER, 4);
+    if (!g1_is_inf9(t)) atomicOr(err, 4u);
+    pts[i] = p;
+}
+
+// P_i = (start+i+1)*G
+__global__ void k_bls_gen_points(g1aB *__restrict__ pts,
+                                 uint8_t *__restrict__ inf, size_t n,
+                                 uint64_t start) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    u64 k[4] = {start + i + 1, 0, 0, 0};
+    g1aB g = g1_generator9<BlsG1>();
+    g1jB acc = g1_inf9<BlsG1>();
+    for (int b = 63; b >= 0; b--) {
+        acc = g1_dbl9(acc);
+        if ((k[0] >> b) & 1) acc = g1_add_affine9(acc, g);
+    }
+    pts[i] = g1_to_affine9(acc);
+    inf[i] = 0;
+}
+
+__global__ void k_bls_points_to_be(const g1aB *__restrict__ pts,
+                                   const uint8_t *__restrict__ inf,
+                                   uint8_t *__restrict__ out, size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    if (inf[i]) {
+        for (int j = 0; j < 12; j++) ((u64 *)(out + 96 * i))[j] = 0;
+        return;
+    }
+    feT_to_be<FpB14T>(out + 96 * i, from_mont9<FpB14T>(pts[i].x));
+    feT_to_be<FpB14T>(out + 96 * i + 48, from_mont9<FpB14T>(pts[i].y));
+}
+
+// scalars: raw 256-bit big-endian -> u64[4] LE words (NO reduction:
+// blst SCALAR_BITS = 256); windows cover all 256 bits (16 x 16)
+__global__ void k_bls_parse_scalars(const uint8_t *__restrict__ in,
+                                    fe4 *__restrict__ out, size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    const u64 *w = (const u64 *)(in + 32 * i);
+    out[i].v[3] = __builtin_bswap64(w[0]);
+    out[i].v[2] = __builtin_bswap64(w[1]);
+    out[i].v[1] = __builtin_bswap64(w[2]);
+    out[i].v[0] = __builtin_bswap64(w[3]);
+}
+
+// single ops (parity probes, bls_blst.rs g1_add / p1_mult semantics)
+__global__ void k_bls_g1_add_single(const uint8_t *in /* 192 B */, uint8_t *out,
+                                    uint32_t *err) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    g1jB acc = g1_inf9<BlsG1>();
+#pragma unroll
+    for (int k = 0; k < 2; k++) {
+        fe14 xr = feT_from_be<FpB14T>(in + 96 * k);
+        fe14 yr = feT_from_be<FpB14T>(in + 96 * k + 48);
+        if (fe9_geq_raw<14>(xr, bn254::FPB_P) ||
+            fe9_geq_raw<14>(yr, bn254::FPB_P)) {
+            atomicOr(err, 2u);
+            return;
+        }
+        if (fe9_is_zero_raw<14>(xr) && fe9_is_zero_raw<14>(yr)) continue;
+        g1aB p{to_mont9<FpB14T>(xr), to_mont9<FpB14T>(yr)};
+        if (!g1a9_on_curve(p)) {
+            atomicOr(err, 1u);
+            return;
+        }
+        acc = g1_add_affine9(acc, p);  // g1_add has no subgroup requirement
+    }
+    g1_to_affine_be9(out, acc);
+}
+
+__global__ void k_bls_g1_mul_single(const uint8_t *in /* 96+32 B */,
+                                    uint8_t *out, uint32_t *err) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    fe14 xr = feT_from_be<FpB14T>(in);
+    fe14 yr = feT_from_be<FpB14T>(in + 48);
+    if (fe9_geq_raw<14>(xr, bn254::FPB_P) || fe9_geq_raw<14>(yr, bn254::FPB_P)) {
+        atomicOr(err, 2u);
+        return;
+    }
+    if (fe9_is_zero_raw<14>(xr) && fe9_is_zero_raw<14>(yr)) {
+        for (int j = 0; j < 12; j++) ((u64 *)out)[j] = 0;
+        return;
+    }
+    g1aB p{to_mont9<FpB14T>(xr), to_mont9<FpB14T>(yr)};
+    if (!g1a9_on_curve(p)) {
+        atomicOr(err, 1u);
+        return;
+    }
+    // full 256-bit scalar (big-endian -> LE words), no reduction
+    const u64 *w = (const u64 *)(in + 96);
+    u64 k[4] = {__builtin_bswap64(w[3]), __builtin_bswap64(w[2]),
+                __builtin_bswap64(w[1]), __builtin_bswap64(w[0])};
+    g1_to_affine_be9(out, g1_scalar_mul9(p, k, 4));
 }
 
 }  // namespace em

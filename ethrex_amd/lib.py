@@ -199,3 +199,104 @@ class NttPlan:
             self.destroy()
         except Exception:
             pass
+
+
+# ==== BLS12-381 G1 (bls_blst.rs semantics) ====
+
+for _f in ("bls12381_g1_add", "bls12381_g1_mul", "bls12381_g1_msm",
+           "bls12381_g1_combine", "bls_msm_plan_create", "bls_msm_plan_destroy",
+           "bls_msm_upload_points", "bls_msm_gen_points",
+           "bls_msm_download_points", "bls_msm_upload_scalars", "bls_msm_run",
+           "bls_msm_run_partial", "bls_msm_last_times"):
+    getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
+
+
+def bls_g1_add(p1: bytes, p2: bytes):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.ethrex_mi355_bls12381_g1_add(_buf(p1), _buf(p2), out)
+    return rc, bytes(out)
+
+
+def bls_g1_mul(point: bytes, scalar: bytes):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.ethrex_mi355_bls12381_g1_mul(_buf(point), _buf(scalar), out)
+    return rc, bytes(out)
+
+
+def bls_g1_msm(points: bytes, scalars: bytes, n: int):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.ethrex_mi355_bls12381_g1_msm(_buf(points), _buf(scalars),
+                                           ctypes.c_size_t(n), out)
+    return rc, bytes(out)
+
+
+def bls_g1_combine(jacobians: bytes, count: int):
+    out = (ctypes.c_uint8 * 96)()
+    rc = _lib.ethrex_mi355_bls12381_g1_combine(_buf(jacobians),
+                                               ctypes.c_size_t(count), out)
+    return rc, bytes(out)
+
+
+def bls_gen_fr(seed: int, n: int) -> bytes:
+    out = (ctypes.c_uint8 * (32 * n))()
+    _lib.ethrex_mi355_bls_gen_fr(ctypes.c_uint64(seed), ctypes.c_size_t(n), out)
+    return bytes(out)
+
+
+class BlsMsmPlan:
+    """Device-resident BLS12-381 G1 MSM plan (blob-KZG commitment shape)."""
+
+    def __init__(self, n: int):
+        self.n = n
+        self._p = ctypes.c_void_p()
+        _check(_lib.ethrex_mi355_bls_msm_plan_create(ctypes.c_size_t(n),
+                                                     ctypes.byref(self._p)),
+               "bls_msm_plan_create")
+
+    def upload_points(self, points: bytes):
+        _check(_lib.ethrex_mi355_bls_msm_upload_points(self._p, _buf(points)),
+               "bls_msm_upload_points")
+
+    def gen_points(self, start: int = 0):
+        _check(_lib.ethrex_mi355_bls_msm_gen_points(self._p,
+                                                    ctypes.c_uint64(start)),
+               "bls_msm_gen_points")
+
+    def download_points(self) -> bytes:
+        out = (ctypes.c_uint8 * (96 * self.n))()
+        _check(_lib.ethrex_mi355_bls_msm_download_points(self._p, out),
+               "bls_msm_download_points")
+        return bytes(out)
+
+    def upload_scalars(self, scalars: bytes):
+        _check(_lib.ethrex_mi355_bls_msm_upload_scalars(self._p, _buf(scalars)),
+               "bls_msm_upload_scalars")
+
+    def run(self) -> bytes:
+        out = (ctypes.c_uint8 * 96)()
+        _check(_lib.ethrex_mi355_bls_msm_run(self._p, out), "bls_msm_run")
+        return bytes(out)
+
+    def run_partial(self) -> bytes:
+        out = (ctypes.c_uint8 * 144)()
+        _check(_lib.ethrex_mi355_bls_msm_run_partial(self._p, out),
+               "bls_msm_run_partial")
+        return bytes(out)
+
+    def last_times(self):
+        t = (ctypes.c_double * 5)()
+        _check(_lib.ethrex_mi355_bls_msm_last_times(self._p, t),
+               "bls_msm_last_times")
+        return {"digits_sort_ms": t[0], "bucket_acc_ms": t[1],
+                "reduce_ms": t[2], "combine_ms": t[3], "total_ms": t[4]}
+
+    def destroy(self):
+        if self._p:
+            _lib.ethrex_mi355_bls_msm_plan_destroy(self._p)
+            self._p = ctypes.c_void_p()
+
+    def __del__(self):
+        try:
+            self.destroy()
+        except Exception:
+            pass

@@ -101,6 +101,41 @@ int ethrex_mi355_ntt_last_times(em_ntt_plan *plan, double times_ms[3]);
  *      against the oracle's independent restatement) ---- */
 void ethrex_mi355_gen_fr(uint64_t seed, size_t n, uint8_t *out32);
 
+/* ==== BLS12-381 G1 (the blob-KZG commitment MSM
+ *      of crates/common/crypto/kzg.rs:208-230 and the EIP-2537 G1 MSM of
+ *      provider.rs:620-634 / bls_blst.rs).  Semantics = the in-tree blst
+ *      path: 48-byte big-endian CANONICAL coordinates (non-canonical =>
+ *      EM_ERR_INPUT), (0,0) identity, on-curve check, r-subgroup check on
+ *      MSM inputs (=> EM_ERR_POINT), scalars = full 256-bit integers (no
+ *      reduction).  Affine point = 96 B x||y; Jacobian partial = 144 B
+ *      X||Y||Z canonical, Z=0 identity. ==== */
+
+int ethrex_mi355_bls12381_g1_add(const uint8_t p1[96], const uint8_t p2[96],
+                                 uint8_t out[96]);
+int ethrex_mi355_bls12381_g1_mul(const uint8_t point[96],
+                                 const uint8_t scalar[32], uint8_t out[96]);
+int ethrex_mi355_bls12381_g1_msm(const uint8_t *points96,
+                                 const uint8_t *scalars32, size_t n,
+                                 uint8_t out[96]);
+int ethrex_mi355_bls12381_g1_combine(const uint8_t *jacobians144, size_t count,
+                                     uint8_t out[96]);
+
+typedef struct em_bls_msm_plan em_bls_msm_plan;
+int ethrex_mi355_bls_msm_plan_create(size_t n, em_bls_msm_plan **plan);
+int ethrex_mi355_bls_msm_plan_destroy(em_bls_msm_plan *plan);
+int ethrex_mi355_bls_msm_upload_points(em_bls_msm_plan *plan,
+                                       const uint8_t *points96);
+int ethrex_mi355_bls_msm_gen_points(em_bls_msm_plan *plan, uint64_t start);
+int ethrex_mi355_bls_msm_download_points(em_bls_msm_plan *plan, uint8_t *out96);
+int ethrex_mi355_bls_msm_upload_scalars(em_bls_msm_plan *plan,
+                                        const uint8_t *scalars32);
+int ethrex_mi355_bls_msm_run(em_bls_msm_plan *plan, uint8_t out[96]);
+int ethrex_mi355_bls_msm_run_partial(em_bls_msm_plan *plan, uint8_t out[144]);
+int ethrex_mi355_bls_msm_last_times(em_bls_msm_plan *plan, double times_ms[5]);
+
+/* n elements uniform in [0, r_bls) (canonical blob field elements) */
+void ethrex_mi355_bls_gen_fr(uint64_t seed, size_t n, uint8_t *out32);
+
 #ifdef __cplusplus
 }
 #endif

@@ -406,11 +406,25 @@ static int bls_msm_core(const uint8_t *points, const uint8_t *scalars, size_t n,
     bg1a *pts = malloc(n * sizeof(bg1a));
     uint8_t *inf = malloc(n);
     int err = 0;
-    for (size_t i = 0; i < n && !err; i++) {
+#ifdef _OPENMP
+#pragma omp parallel for schedule(dynamic, 16)
+#endif
+    for (size_t i = 0; i < n; i++) {
         int ii, rc;
-        if ((rc = bg1_parse_be(&pts[i], &ii, points + 96 * i))) err = rc;
+        if ((rc = bg1_parse_be(&pts[i], &ii, points + 96 * i))) {
+#ifdef _OPENMP
+#pragma omp atomic write
+#endif
+            err = rc;
+            continue;
+        }
         inf[i] = (uint8_t)ii;
-        if (!err && !ii && !bg1_in_subgroup(&pts[i])) err = BORC_ERR_SUBGROUP;
+        if (!ii && !bg1_in_subgroup(&pts[i])) {
+#ifdef _OPENMP
+#pragma omp atomic write
+#endif
+            err = BORC_ERR_SUBGROUP;
+        }
     }
     if (err) { free(pts); free(inf); return err; }
 

@@ -185,6 +185,112 @@ for n, seed in [(1, 20), (2, 21), (8, 22), (32, 23), (256, 24)]:
         "inv_of_fwd": "".join(x.to_bytes(32, "big").hex() for x in a),
     })
 
+# --- BLS12-381 G1 fixtures (semantics: crates/common/crypto/
+# bls_blst.rs — canonical coords, (0,0) identity, subgroup-checked MSM,
+# full 256-bit scalars) ---
+BP = 0x1A0111EA397FE69A4B1BA7B6434BACD764774B84F38512BF6730D2A0F6B0F6241EABFFFEB153FFFFB9FEFFFFFFFFAAAB
+BR = 0x73EDA753299D7D483339D80809A1D80553BDA402FFFE5BFEFFFFFFFF00000001
+BGX = 0x17F1D3A73197D7942695638C4FA9AC0FC3688C4F9774B905A14E3A3F171BAC586C55E83FF97A1AEFFB3AF00ADB22C6BB
+BGY = 0x08B3F481E3AAA0F1A09E30ED741D8AE4FCF5E095D5D00AF600DB18CB2C04B3EDD03CC744A2888AE40CAA232946C5E7E1
+
+
+def bec_add(Pt, Q):
+    if Pt is None:
+        return Q
+    if Q is None:
+        return Pt
+    x1, y1 = Pt
+    x2, y2 = Q
+    if x1 == x2 and (y1 + y2) % BP == 0:
+        return None
+    if Pt == Q:
+        lam = 3 * x1 * x1 * pow(2 * y1, -1, BP) % BP
+    else:
+        lam = (y2 - y1) * pow(x2 - x1, -1, BP) % BP
+    x3 = (lam * lam - x1 - x2) % BP
+    y3 = (lam * (x1 - x3) - y1) % BP
+    return (x3, y3)
+
+
+def bec_mul(k, Pt):
+    Racc = None
+    while k:
+        if k & 1:
+            Racc = bec_add(Racc, Pt)
+        Pt = bec_add(Pt, Pt)
+        k >>= 1
+    return Racc
+
+
+def benc(Pt):
+    if Pt is None:
+        return "00" * 96
+    return Pt[0].to_bytes(48, "big").hex() + Pt[1].to_bytes(48, "big").hex()
+
+
+BG = (BGX, BGY)
+assert (BGY * BGY - BGX ** 3 - 4) % BP == 0
+
+fx["bls_g1_add"] = [
+    {"name": "G+G", "a": benc(BG), "b": benc(BG), "out": benc(bec_mul(2, BG))},
+    {"name": "G+0", "a": benc(BG), "b": "00" * 96, "out": benc(BG)},
+    {"name": "0+0", "a": "00" * 96, "b": "00" * 96, "out": "00" * 96},
+    {"name": "G+offcurve", "a": benc(BG),
+     "b": (1).to_bytes(48, "big").hex() + (1).to_bytes(48, "big").hex(),
+     "error": "point"},
+    {"name": "G+noncanonical", "a": benc(BG),
+     "b": BP.to_bytes(48, "big").hex() + BGY.to_bytes(48, "big").hex(),
+     "error": "input"},
+]
+
+fx["bls_g1_mul"] = []
+for k in [0, 1, 2, 7, 255, BR - 1, BR, (1 << 256) - 1]:
+    fx["bls_g1_mul"].append({
+        "name": f"G*{hex(k)[:18]}", "point": benc(BG),
+        "scalar": k.to_bytes(32, "big").hex(),
+        "out": benc(bec_mul(k, BG)),  # FULL 256-bit scalar, no reduction
+    })
+
+# a curve point NOT in the r-subgroup (for the MSM rejection test):
+# scan small x until x^3+4 is a QR and r*P != infinity
+def sqrt_mod(a, p):
+    assert p % 4 == 3
+    r = pow(a, (p + 1) // 4, p)
+    return r if r * r % p == a else None
+
+
+_off = None
+for x in range(2, 50):
+    y = sqrt_mod((x ** 3 + 4) % BP, BP)
+    if y is None:
+        continue
+    if bec_mul(BR, (x, y)) is not None:  # not killed by r => outside subgroup
+        _off = (x, y)
+        break
+assert _off is not None
+fx["bls_offsubgroup_point"] = benc(_off)
+
+# small MSM fixtures: points (i+1)G (in-subgroup), scalars incl. edge cases
+fx["bls_msm"] = []
+for n, seed in [(1, 30), (2, 31), (17, 32)]:
+    rngb = random.Random(seed)
+    pts, scs = [], []
+    acc = None
+    Pt = None
+    for i in range(n):
+        Pt = bec_add(Pt, BG)
+        if i == 0 and n >= 17:
+            k = 0
+        elif i == 1 and n >= 17:
+            k = (1 << 256) - 1  # full-width scalar
+        else:
+            k = rngb.randrange(BR)
+        pts.append(benc(Pt))
+        scs.append(k.to_bytes(32, "big").hex())
+        acc = bec_add(acc, bec_mul(k, Pt))
+    fx["bls_msm"].append({"n": n, "points": "".join(pts),
+                          "scalars": "".join(scs), "out": benc(acc)})
+
 out_path = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden.json")
 with open(out_path, "w") as f:
     json.dump(fx, f, indent=1)

@@ -62,3 +62,9 @@ def test_compute_fails_loudly_without_gpu():
     assert rc == ethrex_amd.EM_ERR_HIP
     with pytest.raises(Exception):
         ethrex_amd.MsmPlan(16)
+
+
+def test_bls_gen_fr_parity_with_oracle(oracle_mod):
+    import ethrex_amd
+    for seed, n in [(42, 200), (45, 17)]:
+        assert ethrex_amd.bls_gen_fr(seed, n) == oracle_mod.bls_gen_fr(seed, n)
