@@ -65,6 +65,7 @@ def main():
     ap.add_argument("--ntt-log2", type=int, default=24)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--no-ntt", action="store_true")
+    ap.add_argument("--no-bls", action="store_true")
     ap.add_argument("--check", action="store_true",
                     help="verify the first step's result against the oracle "
                          "via the shard-combine identity (adds oracle time)")
@@ -217,6 +218,29 @@ def main():
 
     phase_ms = {k: round(v, 3) for k, v in plan.last_times().items()}
 
+    # ---- BLS12-381 blob-KZG commitment leg (the
+    # 4096-point G1 MSM the sequencer computes per blob,
+    # crates/common/crypto/kzg.rs:208-230; rank 0, single GPU) ----
+    bls = None
+    if rank == 0 and not args.no_bls:
+        bp = ethrex_amd.BlsMsmPlan(4096)
+        bp.gen_points(0)
+        bp.upload_scalars(ethrex_amd.bls_gen_fr(43, 4096))
+        for _ in range(max(args.warmup, 1)):
+            bp.run()
+        t2 = time.perf_counter()
+        for _ in range(args.steps):
+            bp.run()
+        bls_dt = (time.perf_counter() - t2) / args.steps
+        bls = {
+            "metric": "bls12381_blob_kzg_commitments_per_s",
+            "value": 1.0 / bls_dt,
+            "n_points": 4096,
+            "ms_per_commitment": bls_dt * 1000.0,
+            "phase_ms": {k: round(v, 3) for k, v in bp.last_times().items()},
+        }
+        bp.destroy()
+
     cpu_baseline = None
     if rank == 0 and not args.no_cpu_baseline:
         cpu_baseline = cpu_baseline_leg()
@@ -250,6 +274,7 @@ def main():
             "roofline": roofline,
             "ntt": ntt,
             "cpu_baseline": cpu_baseline,
+            "bls_blob": bls,
             "phase_ms": phase_ms,
         }
         print(json.dumps(result), flush=True)
