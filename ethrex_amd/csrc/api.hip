@@ -85,7 +85,9 @@ struct msm_plan_t {
     static constexpr int PB = 2 * F::W64 * 8;  // affine point bytes
     static constexpr int JB = 3 * F::W64 * 8;  // Jacobian partial bytes
     static constexpr int AB = PB;              // affine out bytes
+    static constexpr int SB = std::is_same_v<C, BlsG1> ? 256 : 254;
     size_t n;
+    int cbits;                        // window config: 8 (small) or 16
     g1aT<C> *d_pts = nullptr;
     uint8_t *d_inf = nullptr;
     fe4 *d_scalars = nullptr;
@@ -141,7 +143,18 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     if (rc) return rc;
     auto *p = new msm_plan_t<C>();
     p->n = n;
-    size_t total = n * MSM_NWIN;
+    // small MSMs (blob-KZG-sized) use c=8: the fixed bucket-reduction tail
+    // shrinks 256x; large MSMs use c=16 (bucket work dominates)
+    p->cbits = n <= 65536 ? 8 : 16;
+    using CFGS = msm_cfg<8, msm_plan_t<C>::SB>;
+    using CFGL = msm_cfg<16, msm_plan_t<C>::SB>;
+    const int nwin = p->cbits == 8 ? CFGS::NWIN : CFGL::NWIN;
+    const uint32_t nbuckets = p->cbits == 8 ? CFGS::NBUCKETS : CFGL::NBUCKETS;
+    const int nseg_tot = p->cbits == 8 ? CFGS::NWIN * CFGS::NSEG
+                                       : CFGL::NWIN * CFGL::NSEG;
+    const int npart = p->cbits == 8 ? CFGS::NPART : CFGL::NPART;
+    const int sort_bits = p->cbits == 8 ? CFGS::SORT_BITS : CFGL::SORT_BITS;
+    size_t total = n * (size_t)nwin;
     hipError_t e = hipSuccess;
     auto mal = [&](void **ptr, size_t bytes) {
         if (e == hipSuccess) e = hipMalloc(ptr, bytes);
@@ -154,18 +167,18 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     mal((void **)&p->d_vals, total * 4);
     mal((void **)&p->d_keys_out, total * 4);
     mal((void **)&p->d_vals_out, total * 4);
-    mal((void **)&p->d_offsets, ((size_t)MSM_NBUCKET_TOTAL + 1) * 4);
-    mal((void **)&p->d_buckets, (size_t)MSM_NBUCKET_TOTAL * sizeof(g1jT<C>));
-    mal((void **)&p->d_seg_sum, MSM_NWIN * MSM_NSEG * sizeof(g1jT<C>));
-    mal((void **)&p->d_seg_wsum, MSM_NWIN * MSM_NSEG * sizeof(g1jT<C>));
-    mal((void **)&p->d_partials, MSM_NWIN * MSM_NBLK_PER_WIN * sizeof(g1jT<C>));
-    mal((void **)&p->d_windows, MSM_NWIN * sizeof(g1jT<C>));
+    mal((void **)&p->d_offsets, ((size_t)nbuckets + 1) * 4);
+    mal((void **)&p->d_buckets, (size_t)nbuckets * sizeof(g1jT<C>));
+    mal((void **)&p->d_seg_sum, (size_t)nseg_tot * sizeof(g1jT<C>));
+    mal((void **)&p->d_seg_wsum, (size_t)nseg_tot * sizeof(g1jT<C>));
+    mal((void **)&p->d_partials, (size_t)npart * sizeof(g1jT<C>));
+    mal((void **)&p->d_windows, (size_t)nwin * sizeof(g1jT<C>));
     mal((void **)&p->d_out, msm_plan_t<C>::JB);
     mal((void **)&p->d_err, 4);
     if (e == hipSuccess) {
         e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
                                       p->d_keys_out, p->d_vals, p->d_vals_out,
-                                      total, 0, MSM_SORT_BITS);
+                                      total, 0, sort_bits);
         if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
     }
     for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
@@ -257,42 +270,38 @@ static int msm_upload_scalars_t(msm_plan_t<C> *p, const uint8_t *scalars32) {
     return EM_OK;
 }
 
-template <typename C>
-static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
-    if (!p || !out) return EM_ERR_INPUT;
-    if (!p->have_points || !p->have_scalars) {
-        g_last_err = "msm_run: points/scalars not uploaded";
-        return EM_ERR_INPUT;
-    }
-    size_t total = p->n * MSM_NWIN;
+template <typename C, typename CFG>
+static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
+    size_t total = p->n * (size_t)CFG::NWIN;
     HIP_TRY(hipEventRecord(p->ev[0], 0));
-    hipLaunchKernelGGL(k_digits, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       p->d_scalars, p->d_inf, p->d_keys, p->d_vals, p->n);
+    hipLaunchKernelGGL((k_digits<CFG>), dim3(blocks_for(p->n, 256)), dim3(256),
+                       0, 0, p->d_scalars, p->d_inf, p->d_keys, p->d_vals, p->n);
     size_t tmp = p->sort_tmp_bytes;
     hipError_t e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
                                              p->d_keys_out, p->d_vals,
                                              p->d_vals_out, total, 0,
-                                             MSM_SORT_BITS);
+                                             CFG::SORT_BITS);
     if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs");
-    hipLaunchKernelGGL(k_offsets,
-                       dim3(blocks_for((size_t)MSM_NBUCKET_TOTAL + 1, 256)),
+    hipLaunchKernelGGL((k_offsets<CFG>),
+                       dim3(blocks_for((size_t)CFG::NBUCKETS + 1, 256)),
                        dim3(256), 0, 0, p->d_keys_out, total, p->d_offsets);
     HIP_TRY(hipEventRecord(p->ev[1], 0));
-    hipLaunchKernelGGL(k_bucket_acc<C>, dim3(blocks_for(MSM_NBUCKET_TOTAL, 256)),
-                       dim3(256), 0, 0, p->d_pts, p->d_vals_out, p->d_offsets,
-                       p->d_buckets);
+    hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
+                       dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, 0,
+                       p->d_pts, p->d_vals_out, p->d_offsets, p->d_buckets);
     HIP_TRY(hipEventRecord(p->ev[2], 0));
-    hipLaunchKernelGGL(k_segment_reduce<C>,
-                       dim3(blocks_for(MSM_NWIN * MSM_NSEG, 256)), dim3(256), 0,
-                       0, p->d_buckets, p->d_seg_sum, p->d_seg_wsum);
-    hipLaunchKernelGGL(k_weighted_reduce<C>,
-                       dim3(MSM_NWIN * MSM_NBLK_PER_WIN), dim3(MSM_RED_BLOCK), 0,
-                       0, p->d_seg_sum, p->d_seg_wsum, p->d_partials);
-    hipLaunchKernelGGL(k_window_sum<C>, dim3(1), dim3(64), 0, 0, p->d_partials,
-                       p->d_windows);
+    hipLaunchKernelGGL((k_segment_reduce<C, CFG>),
+                       dim3(blocks_for(CFG::NWIN * CFG::NSEG, 256)), dim3(256),
+                       0, 0, p->d_buckets, p->d_seg_sum, p->d_seg_wsum);
+    hipLaunchKernelGGL((k_weighted_reduce<C, CFG>),
+                       dim3(blocks_for(CFG::NWIN * CFG::NSEG, CFG::RED_BLOCK)),
+                       dim3(CFG::RED_BLOCK), 0, 0, p->d_seg_sum, p->d_seg_wsum,
+                       p->d_partials);
+    hipLaunchKernelGGL((k_window_sum<C, CFG>), dim3(1), dim3(64), 0, 0,
+                       p->d_partials, p->d_windows);
     HIP_TRY(hipEventRecord(p->ev[3], 0));
-    hipLaunchKernelGGL(k_final_combine<C>, dim3(1), dim3(64), 0, 0, p->d_windows,
-                       p->d_out, out_mode);
+    hipLaunchKernelGGL((k_final_combine<C, CFG>), dim3(1), dim3(64), 0, 0,
+                       p->d_windows, p->d_out, out_mode);
     HIP_TRY(hipEventRecord(p->ev[4], 0));
     HIP_TRY(hipMemcpy(out, p->d_out,
                       out_mode == 0 ? msm_plan_t<C>::AB : msm_plan_t<C>::JB,
@@ -310,6 +319,18 @@ static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[4]));
     p->last_ms[4] = ms;
     return EM_OK;
+}
+
+template <typename C>
+static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
+    if (!p || !out) return EM_ERR_INPUT;
+    if (!p->have_points || !p->have_scalars) {
+        g_last_err = "msm_run: points/scalars not uploaded";
+        return EM_ERR_INPUT;
+    }
+    if (p->cbits == 8)
+        return msm_run_cfg<C, msm_cfg<8, msm_plan_t<C>::SB>>(p, out, out_mode);
+    return msm_run_cfg<C, msm_cfg<16, msm_plan_t<C>::SB>>(p, out, out_mode);
 }
 
 // opaque ABI types

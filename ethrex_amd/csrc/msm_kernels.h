@@ -25,23 +25,45 @@
 
 namespace em {
 
-constexpr int MSM_C = 16;                      // window bits
-constexpr int MSM_NWIN = 16;                   // ceil(254/16)
-constexpr uint32_t MSM_DMASK = (1u << MSM_C) - 1;
-constexpr uint32_t MSM_NBUCKET_TOTAL = (uint32_t)MSM_NWIN << MSM_C;  // 1M ids
-constexpr int MSM_SORT_BITS = 20;              // 16 digit + 4 window bits
-constexpr int MSM_SEG = 32;                    // buckets per reduction segment
-constexpr int MSM_NSEG = (1 << MSM_C) / MSM_SEG;  // 2048 segments per window
-constexpr int MSM_RED_BLOCK = 256;             // threads per level-2 block
-constexpr int MSM_NBLK_PER_WIN = MSM_NSEG / MSM_RED_BLOCK;  // 8
+// Window geometry is a compile-time config: CB = window bits.  Large MSMs
+// use c=16 (bucket work dominates); small (blob-KZG-sized) MSMs use c=8 so
+// the fixed bucket-reduction tail shrinks 256x (measured 7.3 ms -> ~2 ms
+// per 4096-point commitment).  Scalar width: BN254 scalars are reduced mod
+// r (254 bits); BLS12-381 scalars are raw 256-bit integers.
+template <int CB, int SBITS>
+struct msm_cfg {
+    static constexpr int C = CB;
+    static constexpr int NWIN = (SBITS + CB - 1) / CB;
+    static constexpr uint32_t DMASK = (1u << CB) - 1;
+    static constexpr uint32_t NBUCKETS = (uint32_t)NWIN << CB;
+    static constexpr int SORT_BITS = CB + 6;   // digit bits + window bits
+    static constexpr int SEG = (1 << CB) >= 8192 ? 32 : 8;
+    static constexpr int NSEG = (1 << CB) / SEG;        // segments per window
+    static constexpr int RED_BLOCK = 256;
+    // when a 256-thread block spans multiple windows the LDS tree is skipped
+    static constexpr bool TREE = NSEG >= RED_BLOCK;
+    static constexpr int NPART = TREE ? NWIN * (NSEG / RED_BLOCK) : NWIN * NSEG;
+};
 
-// digit w = bits [C*w, C*w+C) of the canonical scalar (spans u64 limbs)
+// the two shipped geometries (both curves)
+using CfgL254 = msm_cfg<16, 254>;  // BN254 large
+using CfgS254 = msm_cfg<8, 254>;   // BN254 small (n <= 2^16)
+using CfgL256 = msm_cfg<16, 256>;  // BLS large
+using CfgS256 = msm_cfg<8, 256>;   // BLS small
+
+// compatibility constants for the plan layer (max over configs)
+constexpr int MSM_NWIN_MAX = 33;                 // ceil(256/8) + 1 margin
+constexpr uint32_t MSM_NBUCKETS_MAX = (uint32_t)16 << 16;  // c=16 dominates
+constexpr int MSM_NPART_MAX = 16 * 8 > 32 * 32 ? 16 * 8 : 32 * 32;
+
+// digit w = bits [CB*w, CB*w+CB) of the scalar (spans u64 limbs)
+template <int CB>
 __device__ __forceinline__ uint32_t msm_digit(const fe4 &k, int w) {
-    int bit = MSM_C * w;
+    int bit = CB * w;
     int limb = bit >> 6, off = bit & 63;
     uint64_t d = k.v[limb] >> off;
-    if (off > 64 - MSM_C && limb < 3) d |= k.v[limb + 1] << (64 - off);
-    return (uint32_t)d & MSM_DMASK;
+    if (off > 64 - CB && limb < 3) d |= k.v[limb + 1] << (64 - off);
+    return (uint32_t)d & ((1u << CB) - 1);
 }
 
 // ---- input parsing ----
@@ -107,6 +129,7 @@ __global__ void k_parse_scalars(const uint8_t *__restrict__ in,
 }
 
 // ---- digit extraction ----
+template <typename CFG>
 __global__ void k_digits(const fe4 *__restrict__ scalars,
                          const uint8_t *__restrict__ inf,
                          uint32_t *__restrict__ keys, uint32_t *__restrict__ vals,
@@ -116,20 +139,21 @@ __global__ void k_digits(const fe4 *__restrict__ scalars,
     fe4 k = scalars[i];
     bool skip = inf[i];
 #pragma unroll
-    for (int w = 0; w < MSM_NWIN; w++) {
-        uint32_t d = msm_digit(k, w);
+    for (int w = 0; w < CFG::NWIN; w++) {
+        uint32_t d = msm_digit<CFG::C>(k, w);
         if (skip) d = 0;  // identity points contribute nothing: park in bucket 0
-        keys[(size_t)w * n + i] = ((uint32_t)w << MSM_C) | d;
+        keys[(size_t)w * n + i] = ((uint32_t)w << CFG::C) | d;
         vals[(size_t)w * n + i] = (uint32_t)i;
     }
 }
 
 // ---- bucket segment offsets: lower_bound of each bucket id ----
+template <typename CFG>
 __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total,
                           uint32_t *__restrict__ offsets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
-    if (b > MSM_NBUCKET_TOTAL) return;
-    if (b == MSM_NBUCKET_TOTAL) {
+    if (b > CFG::NBUCKETS) return;
+    if (b == CFG::NBUCKETS) {
         offsets[b] = (uint32_t)total;
         return;
     }
@@ -146,13 +170,13 @@ __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total
 
 // ---- bucket accumulation (the hot kernel) ----
 // one thread per bucket id; digit-0 buckets are skipped (never read later).
-template <typename C>
+template <typename C, typename CFG>
 __global__ void __launch_bounds__(256)
 k_bucket_acc(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
              const uint32_t *__restrict__ offsets, g1jT<C> *__restrict__ buckets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
-    if (b >= MSM_NBUCKET_TOTAL) return;
-    if ((b & MSM_DMASK) == 0) return;  // digit 0
+    if (b >= CFG::NBUCKETS) return;
+    if ((b & CFG::DMASK) == 0) return;  // digit 0
     uint32_t lo = offsets[b], hi = offsets[b + 1];
     g1jT<C> acc = g1_inf9<C>();
     if (lo >= hi) {
@@ -175,7 +199,7 @@ k_bucket_acc(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
 // level 1: per (window, 32-bucket segment): from the top digit down,
 //   run  += B_d           (=> run  = sum of segment buckets)
 //   wsum += run           (=> wsum = sum (d - lo + 1) * B_d)
-template <typename C>
+template <typename C, typename CFG>
 __global__ void __launch_bounds__(256)
 k_segment_reduce(const g1jT<C> *__restrict__ buckets,
                  g1jT<C> *__restrict__ seg_sum,
@@ -183,17 +207,17 @@ k_segment_reduce(const g1jT<C> *__restrict__ buckets,
     // the second accumulator lives in LDS: two register XYZZ accumulators
     // plus mul temporaries spill 232 B/lane to scratch otherwise
     __shared__ g1jT<C> lds_wsum[256];
-    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;  // [0, 16*2048)
-    if (t >= MSM_NWIN * MSM_NSEG) return;
-    uint32_t w = t / MSM_NSEG, seg = t % MSM_NSEG;
-    uint32_t lo = seg * MSM_SEG;
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= CFG::NWIN * CFG::NSEG) return;
+    uint32_t w = t / CFG::NSEG, seg = t % CFG::NSEG;
+    uint32_t lo = seg * CFG::SEG;
     g1jT<C> run = g1_inf9<C>();
     lds_wsum[threadIdx.x] = g1_inf9<C>();
-    for (int32_t d = (int32_t)lo + MSM_SEG - 1; d >= (int32_t)lo; d--) {
+    for (int32_t d = (int32_t)lo + CFG::SEG - 1; d >= (int32_t)lo; d--) {
         // digit-0 bucket is unused: skip the add but keep the wsum step so
         // segment 0 carries the same (d - lo + 1) weights
         if (d != 0)
-            run = g1_add9(run, buckets[((uint32_t)w << MSM_C) | (uint32_t)d]);
+            run = g1_add9(run, buckets[((uint32_t)w << CFG::C) | (uint32_t)d]);
         lds_wsum[threadIdx.x] = g1_add9(lds_wsum[threadIdx.x], run);
     }
     seg_sum[t] = run;
@@ -203,72 +227,81 @@ k_segment_reduce(const g1jT<C> *__restrict__ buckets,
 // level 2: fully parallel weighted combine + LDS tree reduction.
 //   W_w = sum_j [ wsum_j + (j*SEG - 1) * sum_j ]   (j=0 term: -sum_0),
 // then scaled by 2^(16w) (the doubling chains run SIMD-wide here).
-template <typename C>
-__global__ void __launch_bounds__(MSM_RED_BLOCK)
+template <typename C, typename CFG>
+__global__ void __launch_bounds__(CFG::RED_BLOCK)
 k_weighted_reduce(const g1jT<C> *__restrict__ seg_sum,
                   const g1jT<C> *__restrict__ seg_wsum,
-                  g1jT<C> *__restrict__ partials /* NWIN*NBLK_PER_WIN */) {
-    __shared__ g1jT<C> lds[MSM_RED_BLOCK];
-    uint32_t t = blockIdx.x * MSM_RED_BLOCK + threadIdx.x;
-    uint32_t j = t % MSM_NSEG;
-    g1jT<C> ws = seg_wsum[t];
-    g1jT<C> ss = seg_sum[t];
-    g1jT<C> val;
-    if (j == 0) {
-        // weight -1: subtract sum_0
-        if (!g1_is_inf9(ss)) ss.y = neg9<typename C::F>(ss.y);
-        val = g1_add9(ws, ss);
-    } else {
-        uint32_t weight = j * MSM_SEG - 1;  // < 2^C
-        g1jT<C> acc = g1_inf9<C>();
-        for (int b = MSM_C; b >= 0; b--) {
-            acc = g1_dbl9(acc);
-            if ((weight >> b) & 1) acc = g1_add9(acc, ss);
+                  g1jT<C> *__restrict__ partials /* CFG::NPART */) {
+    __shared__ g1jT<C> lds[CFG::RED_BLOCK];
+    uint32_t t = blockIdx.x * CFG::RED_BLOCK + threadIdx.x;
+    bool live = t < (uint32_t)(CFG::NWIN * CFG::NSEG);
+    uint32_t j = t % CFG::NSEG;
+    g1jT<C> val = g1_inf9<C>();
+    if (live) {
+        g1jT<C> ws = seg_wsum[t];
+        g1jT<C> ss = seg_sum[t];
+        if (j == 0) {
+            // weight -1: subtract sum_0
+            if (!g1_is_inf9(ss)) ss.y = neg9<typename C::F>(ss.y);
+            val = g1_add9(ws, ss);
+        } else {
+            uint32_t weight = j * CFG::SEG - 1;  // < 2^C
+            g1jT<C> acc = g1_inf9<C>();
+            for (int b = CFG::C; b >= 0; b--) {
+                acc = g1_dbl9(acc);
+                if ((weight >> b) & 1) acc = g1_add9(acc, ss);
+            }
+            val = g1_add9(ws, acc);
         }
-        val = g1_add9(ws, acc);
+        // fold the window factor 2^(C*w) HERE: the doubling chains run
+        // SIMD-wide across all threads
+        uint32_t w = t / CFG::NSEG;
+        for (uint32_t d = 0; d < (uint32_t)CFG::C * w; d++) val = g1_dbl9(val);
     }
-    // fold the window factor 2^(C*w) HERE: all 32K threads double in
-    // parallel (SIMD-wide); a per-block post-tree chain serializes on one
-    // lane per CU and measured ~3 ms vs ~2 ms for this layout
-    uint32_t w = t / MSM_NSEG;
-    for (uint32_t d = 0; d < (uint32_t)MSM_C * w; d++) val = g1_dbl9(val);
-    lds[threadIdx.x] = val;
-    __syncthreads();
-    for (int s = MSM_RED_BLOCK / 2; s > 0; s >>= 1) {
-        if (threadIdx.x < (uint32_t)s) {
-            g1jT<C> o = lds[threadIdx.x + s];
-            g1jT<C> m = g1_add9(lds[threadIdx.x], o);
-            lds[threadIdx.x] = m;
-        }
+    if constexpr (CFG::TREE) {
+        // one window per block: LDS tree -> one partial per block
+        lds[threadIdx.x] = val;
         __syncthreads();
+        for (int s = CFG::RED_BLOCK / 2; s > 0; s >>= 1) {
+            if (threadIdx.x < (uint32_t)s) {
+                g1jT<C> o = lds[threadIdx.x + s];
+                g1jT<C> m = g1_add9(lds[threadIdx.x], o);
+                lds[threadIdx.x] = m;
+            }
+            __syncthreads();
+        }
+        if (threadIdx.x == 0) partials[blockIdx.x] = lds[0];
+    } else {
+        // small config: blocks span windows; write per-segment partials
+        if (live) partials[t] = val;
     }
-    if (threadIdx.x == 0) partials[blockIdx.x] = lds[0];
 }
 
 // level 3: 16 threads, 8 partials each -> per-window sums (pre-scaled)
-template <typename C>
+template <typename C, typename CFG>
 __global__ void k_window_sum(const g1jT<C> *__restrict__ partials,
                              g1jT<C> *__restrict__ windows) {
+    constexpr int PER_WIN = CFG::NPART / CFG::NWIN;
     uint32_t w = blockIdx.x * blockDim.x + threadIdx.x;
-    if (w >= MSM_NWIN) return;
+    if (w >= CFG::NWIN) return;
     g1jT<C> acc = g1_inf9<C>();
-    for (int b = 0; b < MSM_NBLK_PER_WIN; b++)
-        acc = g1_add9(acc, partials[w * MSM_NBLK_PER_WIN + b]);
+    for (int b = 0; b < PER_WIN; b++)
+        acc = g1_add9(acc, partials[w * PER_WIN + b]);
     windows[w] = acc;
 }
 
 // ---- final combine + output ----
 // out_mode 0: 64-byte BE affine (infinity -> zeros)
 // out_mode 1: 96-byte BE Jacobian canonical X||Y||Z (Z=0 -> infinity)
-template <typename C>
+template <typename C, typename CFG>
 __global__ void k_final_combine(const g1jT<C> *__restrict__ windows,
                                 uint8_t *__restrict__ out, int out_mode) {
     using F = typename C::F;
     constexpr int NB = F::W64 * 8;
     if (blockIdx.x != 0 || threadIdx.x != 0) return;
-    // windows[] arrive pre-scaled by 2^(16w) (k_weighted_reduce)
-    g1jT<C> acc = windows[MSM_NWIN - 1];
-    for (int w = MSM_NWIN - 2; w >= 0; w--) acc = g1_add9(acc, windows[w]);
+    // windows[] arrive pre-scaled by 2^(C*w) (k_weighted_reduce)
+    g1jT<C> acc = windows[CFG::NWIN - 1];
+    for (int w = CFG::NWIN - 2; w >= 0; w--) acc = g1_add9(acc, windows[w]);
     if (out_mode == 0) {
         g1_to_affine_be9(out, acc);
     } else {
