@@ -294,14 +294,27 @@ __global__ void k_window_sum(const g1jT<C> *__restrict__ partials,
 // out_mode 0: 64-byte BE affine (infinity -> zeros)
 // out_mode 1: 96-byte BE Jacobian canonical X||Y||Z (Z=0 -> infinity)
 template <typename C, typename CFG>
-__global__ void k_final_combine(const g1jT<C> *__restrict__ windows,
-                                uint8_t *__restrict__ out, int out_mode) {
+__global__ void __launch_bounds__(64)
+k_final_combine(const g1jT<C> *__restrict__ windows,
+                uint8_t *__restrict__ out, int out_mode) {
     using F = typename C::F;
     constexpr int NB = F::W64 * 8;
-    if (blockIdx.x != 0 || threadIdx.x != 0) return;
-    // windows[] arrive pre-scaled by 2^(C*w) (k_weighted_reduce)
-    g1jT<C> acc = windows[CFG::NWIN - 1];
-    for (int w = CFG::NWIN - 2; w >= 0; w--) acc = g1_add9(acc, windows[w]);
+    // windows[] arrive pre-scaled by 2^(C*w); 64-lane LDS tree for the
+    // window sum (a serial 32-add chain costs ~1 ms on the 14-limb field)
+    __shared__ g1jT<C> lds[64];
+    uint32_t t = threadIdx.x;
+    lds[t] = t < (uint32_t)CFG::NWIN ? windows[t] : g1_inf9<C>();
+    __syncthreads();
+    for (int sh = 32; sh > 0; sh >>= 1) {
+        if (t < (uint32_t)sh) {
+            g1jT<C> o = lds[t + sh];
+            g1jT<C> m = g1_add9(lds[t], o);
+            lds[t] = m;
+        }
+        __syncthreads();
+    }
+    if (t != 0) return;
+    g1jT<C> acc = lds[0];
     if (out_mode == 0) {
         g1_to_affine_be9(out, acc);
     } else {
