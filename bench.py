@@ -111,9 +111,7 @@ def main():
             return plan.run()
         part = plan.run_partial()
         allparts = allgather_partials(part, dist, device="cuda")
-        rc, out = ethrex_amd.g1_combine(allparts, world)
-        assert rc == 0
-        return out
+        return plan.combine(allparts, world)
     # warmup
     first = None
     for _ in range(max(args.warmup, 1)):

@@ -369,6 +369,21 @@ extern "C" int ethrex_mi355_msm_last_times(em_msm_plan *p, double times_ms[5]) {
     return EM_OK;
 }
 
+// combine Jacobian partials reusing the plan's buffers (no per-call
+// hipMalloc: the N>1 exchange runs this every step)
+extern "C" int ethrex_mi355_msm_combine(em_msm_plan *p,
+                                        const uint8_t *jacobians96,
+                                        size_t count, uint8_t out[64]) {
+    if (!p || !jacobians96 || !out || count == 0 || count * 96 > p->n * 64)
+        return EM_ERR_INPUT;
+    HIP_TRY(hipMemcpy(p->d_scratch, jacobians96, 96 * count,
+                      hipMemcpyHostToDevice));
+    hipLaunchKernelGGL((k_g1_combine<Bn254G1>), dim3(1), dim3(64), 0, 0,
+                       p->d_scratch, count, p->d_out);
+    HIP_TRY(hipMemcpy(out, p->d_out, 64, hipMemcpyDeviceToHost));
+    return EM_OK;
+}
+
 extern "C" int ethrex_mi355_bls_msm_plan_create(size_t n, em_bls_msm_plan **plan) {
     return msm_create_t(n, (msm_plan_t<BlsG1> **)plan);
 }

@@ -37,7 +37,7 @@ struct msm_cfg {
     static constexpr uint32_t DMASK = (1u << CB) - 1;
     static constexpr uint32_t NBUCKETS = (uint32_t)NWIN << CB;
     static constexpr int SORT_BITS = CB + 6;   // digit bits + window bits
-    static constexpr int SEG = (1 << CB) >= 8192 ? 32 : 8;
+    static constexpr int SEG = (1 << CB) >= 8192 ? 16 : 8;
     static constexpr int NSEG = (1 << CB) / SEG;        // segments per window
     static constexpr int RED_BLOCK = 256;
     // when a 256-thread block spans multiple windows the LDS tree is skipped

@@ -28,7 +28,7 @@ for _f in ("device_count", "set_device", "bn254_g1_add", "bn254_g1_mul",
            "bn254_g1_msm", "bn254_fr_ntt", "bn254_g1_combine",
            "msm_plan_create", "msm_plan_destroy", "msm_upload_points",
            "msm_gen_points", "msm_download_points", "msm_upload_scalars",
-           "msm_run", "msm_run_partial", "msm_last_times",
+           "msm_run", "msm_run_partial", "msm_last_times", "msm_combine",
            "ntt_plan_create", "ntt_plan_destroy", "ntt_upload", "ntt_run",
            "ntt_download", "ntt_last_times"):
     getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
@@ -150,6 +150,14 @@ class MsmPlan:
         _check(_lib.ethrex_mi355_msm_last_times(self._p, t), "msm_last_times")
         return {"digits_sort_ms": t[0], "bucket_acc_ms": t[1],
                 "reduce_ms": t[2], "combine_ms": t[3], "total_ms": t[4]}
+
+    def combine(self, jacobians: bytes, count: int) -> bytes:
+        """combine Jacobian partials reusing this plan's device buffers"""
+        out = (ctypes.c_uint8 * 64)()
+        _check(_lib.ethrex_mi355_msm_combine(self._p, _buf(jacobians),
+                                             ctypes.c_size_t(count), out),
+               "msm_combine")
+        return bytes(out)
 
     def destroy(self):
         if self._p:

@@ -87,6 +87,9 @@ int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96, size_t count,
  * [0]=digits+sort, [1]=bucket accumulation, [2]=bucket reduction,
  * [3]=window combine + affine, [4]=total */
 int ethrex_mi355_msm_last_times(em_msm_plan *plan, double times_ms[5]);
+/* combine partials reusing the plan's device buffers (per-step exchange) */
+int ethrex_mi355_msm_combine(em_msm_plan *plan, const uint8_t *jacobians96,
+                             size_t count, uint8_t out[64]);
 
 int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan);
 int ethrex_mi355_ntt_plan_destroy(em_ntt_plan *plan);

@@ -193,3 +193,21 @@ def test_gpu_msm_parity_2_20_direct(gpu, oracle_mod):
     plan.destroy()
     rc, want = oracle_mod.g1_msm(pts, scs, n)
     assert rc == 0 and got == want
+
+
+def test_gpu_plan_combine_matches_standalone(gpu, oracle_mod):
+    """plan-attached combine (the per-step N>1 exchange path) matches the
+    standalone combine and the oracle."""
+    n = 1 << 12
+    parts = b""
+    for s in range(4):
+        plan = gpu.MsmPlan(n // 4)
+        plan.gen_points(s * (n // 4))
+        plan.upload_scalars(gpu.gen_fr(42 + s, n // 4))
+        parts += plan.run_partial()
+        plan.destroy()
+    plan = gpu.MsmPlan(n)
+    got = plan.combine(parts, 4)
+    plan.destroy()
+    rc, want = oracle_mod.g1_combine_jacobian(parts, 4)
+    assert rc == 0 and got == want
