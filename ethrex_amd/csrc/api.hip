@@ -98,7 +98,11 @@ struct msm_plan_t {
     uint32_t *d_vals_out = nullptr;
     void *d_sort_tmp = nullptr;
     size_t sort_tmp_bytes = 0;
-    uint32_t *d_offsets = nullptr;    // NBUCKET_TOTAL + 1
+    uint32_t *d_offsets = nullptr;    // NBUCKETS + 1
+    uint32_t *d_blen = nullptr;       // bucket lengths (schedule keys)
+    uint32_t *d_blen_out = nullptr;
+    uint32_t *d_bids = nullptr;       // bucket ids
+    uint32_t *d_sched = nullptr;      // length-sorted bucket ids
     g1jT<C> *d_buckets = nullptr;     // NBUCKET_TOTAL
     g1jT<C> *d_seg_sum = nullptr;     // NWIN*NSEG
     g1jT<C> *d_seg_wsum = nullptr;
@@ -125,6 +129,10 @@ static int msm_destroy_t(msm_plan_t<C> *p) {
     hipFree(p->d_vals_out);
     hipFree(p->d_sort_tmp);
     hipFree(p->d_offsets);
+    hipFree(p->d_blen);
+    hipFree(p->d_blen_out);
+    hipFree(p->d_bids);
+    hipFree(p->d_sched);
     hipFree(p->d_buckets);
     hipFree(p->d_seg_sum);
     hipFree(p->d_seg_wsum);
@@ -168,6 +176,10 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     mal((void **)&p->d_keys_out, total * 4);
     mal((void **)&p->d_vals_out, total * 4);
     mal((void **)&p->d_offsets, ((size_t)nbuckets + 1) * 4);
+    mal((void **)&p->d_blen, (size_t)nbuckets * 4);
+    mal((void **)&p->d_blen_out, (size_t)nbuckets * 4);
+    mal((void **)&p->d_bids, (size_t)nbuckets * 4);
+    mal((void **)&p->d_sched, (size_t)nbuckets * 4);
     mal((void **)&p->d_buckets, (size_t)nbuckets * sizeof(g1jT<C>));
     mal((void **)&p->d_seg_sum, (size_t)nseg_tot * sizeof(g1jT<C>));
     mal((void **)&p->d_seg_wsum, (size_t)nseg_tot * sizeof(g1jT<C>));
@@ -285,10 +297,22 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     hipLaunchKernelGGL((k_offsets<CFG>),
                        dim3(blocks_for((size_t)CFG::NBUCKETS + 1, 256)),
                        dim3(256), 0, 0, p->d_keys_out, total, p->d_offsets);
+    // schedule buckets by run length (kills wave divergence in the hot kernel)
+    hipLaunchKernelGGL((k_bucket_lengths<CFG>),
+                       dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, 0,
+                       p->d_offsets, p->d_blen, p->d_bids);
+    {
+        size_t tmp2 = p->sort_tmp_bytes;
+        hipError_t e2 = rocprim::radix_sort_pairs(
+            p->d_sort_tmp, tmp2, p->d_blen, p->d_blen_out, p->d_bids,
+            p->d_sched, (size_t)CFG::NBUCKETS, 0, 32);
+        if (e2 != hipSuccess) return hip_fail(e2, "bucket length sort");
+    }
     HIP_TRY(hipEventRecord(p->ev[1], 0));
     hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
                        dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, 0,
-                       p->d_pts, p->d_vals_out, p->d_offsets, p->d_buckets);
+                       p->d_pts, p->d_vals_out, p->d_offsets, p->d_sched,
+                       p->d_buckets);
     HIP_TRY(hipEventRecord(p->ev[2], 0));
     hipLaunchKernelGGL((k_segment_reduce<C, CFG>),
                        dim3(blocks_for(CFG::NWIN * CFG::NSEG, 256)), dim3(256),

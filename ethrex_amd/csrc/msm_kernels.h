@@ -168,14 +168,29 @@ __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total
     offsets[b] = (uint32_t)lo;
 }
 
+// ---- bucket-length schedule: sort bucket ids by run length so each wave
+// processes similar-length runs (a wave pays max-of-64 Poisson run lengths
+// otherwise: measured ~1.2-1.5x divergence loss) ----
+template <typename CFG>
+__global__ void k_bucket_lengths(const uint32_t *__restrict__ offsets,
+                                 uint32_t *__restrict__ len,
+                                 uint32_t *__restrict__ ids) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= CFG::NBUCKETS) return;
+    len[b] = (b & CFG::DMASK) == 0 ? 0u : offsets[b + 1] - offsets[b];
+    ids[b] = b;
+}
+
 // ---- bucket accumulation (the hot kernel) ----
-// one thread per bucket id; digit-0 buckets are skipped (never read later).
+// one thread per SCHEDULED bucket id; digit-0 buckets skipped.
 template <typename C, typename CFG>
 __global__ void __launch_bounds__(256)
 k_bucket_acc(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
-             const uint32_t *__restrict__ offsets, g1jT<C> *__restrict__ buckets) {
-    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
-    if (b >= CFG::NBUCKETS) return;
+             const uint32_t *__restrict__ offsets,
+             const uint32_t *__restrict__ sched, g1jT<C> *__restrict__ buckets) {
+    uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    if (tid >= CFG::NBUCKETS) return;
+    uint32_t b = sched[tid];
     if ((b & CFG::DMASK) == 0) return;  // digit 0
     uint32_t lo = offsets[b], hi = offsets[b + 1];
     g1jT<C> acc = g1_inf9<C>();
