@@ -191,6 +191,12 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
         e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
                                       p->d_keys_out, p->d_vals, p->d_vals_out,
                                       total, 0, sort_bits);
+        size_t tmp_len = 0;
+        if (e == hipSuccess)
+            e = rocprim::radix_sort_pairs(nullptr, tmp_len, p->d_blen,
+                                          p->d_blen_out, p->d_bids, p->d_sched,
+                                          (size_t)nbuckets, 0, 32);
+        if (tmp_len > p->sort_tmp_bytes) p->sort_tmp_bytes = tmp_len;
         if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
     }
     for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
