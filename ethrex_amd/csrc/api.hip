@@ -89,6 +89,8 @@ struct msm_plan_t {
     size_t n;
     int cbits;                        // window config: 8 (small) or 16
     g1aT<C> *d_pts = nullptr;
+    g1aT<C> *d_pts_ext = nullptr;     // fixed-base table (FB_NWIN * n)
+    bool fixed_base = false;
     uint8_t *d_inf = nullptr;
     fe4 *d_scalars = nullptr;
     uint8_t *d_scratch = nullptr;     // PB*n bytes: point/scalar byte staging
@@ -120,6 +122,7 @@ template <typename C>
 static int msm_destroy_t(msm_plan_t<C> *p) {
     if (!p) return EM_ERR_INPUT;
     hipFree(p->d_pts);
+    hipFree(p->d_pts_ext);
     hipFree(p->d_inf);
     hipFree(p->d_scalars);
     hipFree(p->d_scratch);
@@ -235,6 +238,7 @@ static int msm_upload_points_t(msm_plan_t<C> *p, const uint8_t *points) {
     }
     if (err) return EM_ERR_POINT;
     p->have_points = true;
+    p->fixed_base = false;
     return EM_OK;
 }
 
@@ -250,6 +254,7 @@ static int msm_gen_points_t(msm_plan_t<C> *p, uint64_t start) {
     }
     HIP_TRY(hipDeviceSynchronize());
     p->have_points = true;
+    p->fixed_base = false;
     return EM_OK;
 }
 
@@ -267,6 +272,29 @@ static int msm_download_points_t(msm_plan_t<C> *p, uint8_t *out) {
                            p->n);
     }
     HIP_TRY(hipMemcpy(out, p->d_scratch, p->n * PB, hipMemcpyDeviceToHost));
+    return EM_OK;
+}
+
+// build the fixed-base table (blob-KZG mode: points fixed across blobs).
+// requires the small-config plan (n <= 65536); ~5 ms once per setup.
+template <typename C>
+static int msm_precompute_t(msm_plan_t<C> *p) {
+    if (!p || !p->have_points) return EM_ERR_INPUT;
+    if (p->cbits != 8) {
+        g_last_err = "fixed-base precompute requires n <= 65536";
+        return EM_ERR_INPUT;
+    }
+    if (!p->d_pts_ext) {
+        hipError_t e =
+            hipMalloc((void **)&p->d_pts_ext,
+                      (size_t)FB_NWIN * p->n * sizeof(g1aT<C>));
+        if (e != hipSuccess) return hip_fail(e, "fb precompute alloc");
+    }
+    hipLaunchKernelGGL((k_fb_precompute<C>),
+                       dim3(blocks_for(p->n * FB_NWIN, 256)), dim3(256), 0, 0,
+                       p->d_pts, p->d_inf, p->n, p->d_pts_ext);
+    HIP_TRY(hipDeviceSynchronize());
+    p->fixed_base = true;
     return EM_OK;
 }
 
@@ -290,10 +318,19 @@ static int msm_upload_scalars_t(msm_plan_t<C> *p, const uint8_t *scalars32) {
 
 template <typename C, typename CFG>
 static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
-    size_t total = p->n * (size_t)CFG::NWIN;
+    constexpr bool FB = std::is_same_v<CFG, CfgFB>;
+    size_t total = FB ? p->n * (size_t)FB_NWIN : p->n * (size_t)CFG::NWIN;
+    const g1aT<C> *pts = FB ? p->d_pts_ext : p->d_pts;
     HIP_TRY(hipEventRecord(p->ev[0], 0));
-    hipLaunchKernelGGL((k_digits<CFG>), dim3(blocks_for(p->n, 256)), dim3(256),
-                       0, 0, p->d_scalars, p->d_inf, p->d_keys, p->d_vals, p->n);
+    if constexpr (FB) {
+        hipLaunchKernelGGL(k_fb_digits, dim3(blocks_for(total, 256)), dim3(256),
+                           0, 0, p->d_scalars, p->d_inf, p->d_keys, p->d_vals,
+                           p->n);
+    } else {
+        hipLaunchKernelGGL((k_digits<CFG>), dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_scalars, p->d_inf, p->d_keys,
+                           p->d_vals, p->n);
+    }
     size_t tmp = p->sort_tmp_bytes;
     hipError_t e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
                                              p->d_keys_out, p->d_vals,
@@ -317,7 +354,7 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     HIP_TRY(hipEventRecord(p->ev[1], 0));
     hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
                        dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, 0,
-                       p->d_pts, p->d_vals_out, p->d_offsets, p->d_sched,
+                       pts, p->d_vals_out, p->d_offsets, p->d_sched,
                        p->d_buckets);
     HIP_TRY(hipEventRecord(p->ev[2], 0));
     hipLaunchKernelGGL((k_segment_reduce<C, CFG>),
@@ -327,7 +364,7 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
                        dim3(blocks_for(CFG::NWIN * CFG::NSEG, CFG::RED_BLOCK)),
                        dim3(CFG::RED_BLOCK), 0, 0, p->d_seg_sum, p->d_seg_wsum,
                        p->d_partials);
-    hipLaunchKernelGGL((k_window_sum<C, CFG>), dim3(1), dim3(64), 0, 0,
+    hipLaunchKernelGGL((k_window_sum<C, CFG>), dim3(CFG::NWIN), dim3(64), 0, 0,
                        p->d_partials, p->d_windows);
     HIP_TRY(hipEventRecord(p->ev[3], 0));
     hipLaunchKernelGGL((k_final_combine<C, CFG>), dim3(1), dim3(64), 0, 0,
@@ -358,6 +395,7 @@ static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
         g_last_err = "msm_run: points/scalars not uploaded";
         return EM_ERR_INPUT;
     }
+    if (p->fixed_base) return msm_run_cfg<C, CfgFB>(p, out, out_mode);
     if (p->cbits == 8)
         return msm_run_cfg<C, msm_cfg<8, msm_plan_t<C>::SB>>(p, out, out_mode);
     return msm_run_cfg<C, msm_cfg<16, msm_plan_t<C>::SB>>(p, out, out_mode);
@@ -442,6 +480,9 @@ extern "C" int ethrex_mi355_bls_msm_run(em_bls_msm_plan *p, uint8_t out[96]) {
 extern "C" int ethrex_mi355_bls_msm_run_partial(em_bls_msm_plan *p,
                                                 uint8_t out[144]) {
     return msm_run_inner_t((msm_plan_t<BlsG1> *)p, out, 1);
+}
+extern "C" int ethrex_mi355_bls_msm_precompute(em_bls_msm_plan *p) {
+    return msm_precompute_t((msm_plan_t<BlsG1> *)p);
 }
 extern "C" int ethrex_mi355_bls_msm_last_times(em_bls_msm_plan *p,
                                                double times_ms[5]) {

@@ -37,7 +37,8 @@ struct msm_cfg {
     static constexpr uint32_t DMASK = (1u << CB) - 1;
     static constexpr uint32_t NBUCKETS = (uint32_t)NWIN << CB;
     static constexpr int SORT_BITS = CB + 6;   // digit bits + window bits
-    static constexpr int SEG = (1 << CB) >= 8192 ? 16 : 8;
+    static constexpr int SEG =
+        (1 << CB) >= 8192 ? 16 : ((1 << CB) >= 4096 ? 32 : 8);
     static constexpr int NSEG = (1 << CB) / SEG;        // segments per window
     static constexpr int RED_BLOCK = 256;
     // when a 256-thread block spans multiple windows the LDS tree is skipped
@@ -293,16 +294,30 @@ k_weighted_reduce(const g1jT<C> *__restrict__ seg_sum,
 }
 
 // level 3: 16 threads, 8 partials each -> per-window sums (pre-scaled)
+// level 3: one 64-lane block per window (serial per-window sums cost up to
+// ~1 ms when PER_WIN is large)
 template <typename C, typename CFG>
-__global__ void k_window_sum(const g1jT<C> *__restrict__ partials,
-                             g1jT<C> *__restrict__ windows) {
+__global__ void __launch_bounds__(64)
+k_window_sum(const g1jT<C> *__restrict__ partials,
+             g1jT<C> *__restrict__ windows) {
     constexpr int PER_WIN = CFG::NPART / CFG::NWIN;
-    uint32_t w = blockIdx.x * blockDim.x + threadIdx.x;
-    if (w >= CFG::NWIN) return;
+    __shared__ g1jT<C> lds[64];
+    uint32_t w = blockIdx.x;
+    uint32_t t = threadIdx.x;
     g1jT<C> acc = g1_inf9<C>();
-    for (int b = 0; b < PER_WIN; b++)
+    for (uint32_t b = t; b < (uint32_t)PER_WIN; b += 64)
         acc = g1_add9(acc, partials[w * PER_WIN + b]);
-    windows[w] = acc;
+    lds[t] = acc;
+    __syncthreads();
+    for (int sh = 32; sh > 0; sh >>= 1) {
+        if (t < (uint32_t)sh) {
+            g1jT<C> o = lds[t + sh];
+            g1jT<C> m = g1_add9(lds[t], o);
+            lds[t] = m;
+        }
+        __syncthreads();
+    }
+    if (t == 0) windows[w] = lds[0];
 }
 
 // ---- final combine + output ----
@@ -405,6 +420,52 @@ __global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
         acc = g1_add9(acc, g1_jacobian_to_xyzz9<C>(X, Y, Z));
     }
     g1_to_affine_be9(out, acc);
+}
+
+// ---- fixed-base mode (blob-KZG: the 4096 setup points are FIXED across
+// blobs, so precompute 2^(12w)*P_i once — the window dimension collapses
+// into the point table and the per-blob MSM runs one 4096-bucket window;
+// the same idea as c-kzg's KZG_PRECOMPUTE fixed-base tables) ----
+
+constexpr int FB_C = 12;
+constexpr int FB_NWIN = 22;  // ceil(256 / 12), raw 256-bit scalars
+using CfgFB = msm_cfg<FB_C, FB_C>;  // NWIN=1: single merged window space
+
+// P_ext[w*n + i] = 2^(12w) * P_i (affine); inf entries follow the base flag
+template <typename C>
+__global__ void k_fb_precompute(const g1aT<C> *__restrict__ pts,
+                                const uint8_t *__restrict__ inf, size_t n,
+                                g1aT<C> *__restrict__ ext) {
+    size_t e = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (e >= n * FB_NWIN) return;
+    size_t i = e % n;
+    int w = (int)(e / n);
+    if (inf[i]) {
+        ext[e] = pts[i];
+        return;
+    }
+    g1jT<C> acc;
+    acc.x = pts[i].x;
+    acc.y = pts[i].y;
+    acc.zz = fe9_load<C::F::L>(C::F::ONE);
+    acc.zzz = fe9_load<C::F::L>(C::F::ONE);
+    for (int d = 0; d < FB_C * w; d++) acc = g1_dbl9(acc);
+    ext[e] = g1_to_affine9(acc);
+}
+
+// digits for fixed-base: entry e = (w, i); key = digit only (single window
+// space), value = index into the precomputed table
+__global__ void k_fb_digits(const fe4 *__restrict__ scalars,
+                            const uint8_t *__restrict__ inf,
+                            uint32_t *__restrict__ keys,
+                            uint32_t *__restrict__ vals, size_t n) {
+    size_t e = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (e >= n * FB_NWIN) return;
+    size_t i = e % n;
+    int w = (int)(e / n);
+    uint32_t d = inf[i] ? 0 : msm_digit<FB_C>(scalars[i], w);
+    keys[e] = d;
+    vals[e] = (uint32_t)e;
 }
 
 // ============================================================================

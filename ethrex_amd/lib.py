@@ -215,7 +215,7 @@ for _f in ("bls12381_g1_add", "bls12381_g1_mul", "bls12381_g1_msm",
            "bls12381_g1_combine", "bls_msm_plan_create", "bls_msm_plan_destroy",
            "bls_msm_upload_points", "bls_msm_gen_points",
            "bls_msm_download_points", "bls_msm_upload_scalars", "bls_msm_run",
-           "bls_msm_run_partial", "bls_msm_last_times"):
+           "bls_msm_run_partial", "bls_msm_last_times", "bls_msm_precompute"):
     getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
 
 
@@ -279,6 +279,11 @@ class BlsMsmPlan:
     def upload_scalars(self, scalars: bytes):
         _check(_lib.ethrex_mi355_bls_msm_upload_scalars(self._p, _buf(scalars)),
                "bls_msm_upload_scalars")
+
+    def precompute(self):
+        """fixed-base table (setup points fixed across blobs)"""
+        _check(_lib.ethrex_mi355_bls_msm_precompute(self._p),
+               "bls_msm_precompute")
 
     def run(self) -> bytes:
         out = (ctypes.c_uint8 * 96)()

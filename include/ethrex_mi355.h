@@ -132,6 +132,10 @@ int ethrex_mi355_bls_msm_gen_points(em_bls_msm_plan *plan, uint64_t start);
 int ethrex_mi355_bls_msm_download_points(em_bls_msm_plan *plan, uint8_t *out96);
 int ethrex_mi355_bls_msm_upload_scalars(em_bls_msm_plan *plan,
                                         const uint8_t *scalars32);
+/* build the fixed-base table (2^(12w)*P_i, c-kzg KZG_PRECOMPUTE-style):
+ * the setup points are fixed across blobs, so subsequent runs collapse the
+ * window dimension.  Requires n <= 65536; invalidated by new points. */
+int ethrex_mi355_bls_msm_precompute(em_bls_msm_plan *plan);
 int ethrex_mi355_bls_msm_run(em_bls_msm_plan *plan, uint8_t out[96]);
 int ethrex_mi355_bls_msm_run_partial(em_bls_msm_plan *plan, uint8_t out[144]);
 int ethrex_mi355_bls_msm_last_times(em_bls_msm_plan *plan, double times_ms[5]);

@@ -70,6 +70,23 @@ def test_gpu_bls_msm_blob_4096(gpu, oracle_mod):
     assert t["total_ms"] > 0
 
 
+def test_gpu_bls_fixed_base_blob(gpu, oracle_mod):
+    """fixed-base (precomputed) blob commitment path: bit-exact vs the
+    oracle and vs the non-precomputed path across multiple scalar sets."""
+    n = 4096
+    plan = gpu.BlsMsmPlan(n)
+    plan.gen_points(0)
+    pts = plan.download_points()
+    plan.precompute()
+    for seed in (42, 99):
+        scs = gpu.bls_gen_fr(seed, n)
+        plan.upload_scalars(scs)
+        got = plan.run()
+        rc, want = oracle_mod.bls_g1_msm(pts, scs, n)
+        assert rc == 0 and got == want, seed
+    plan.destroy()
+
+
 def test_gpu_bls_msm_parity_2_14(gpu, oracle_mod):
     n = 1 << 14
     plan = gpu.BlsMsmPlan(n)
