@@ -427,8 +427,11 @@ __global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
 // into the point table and the per-blob MSM runs one 4096-bucket window;
 // the same idea as c-kzg's KZG_PRECOMPUTE fixed-base tables) ----
 
-constexpr int FB_C = 12;
-constexpr int FB_NWIN = 22;  // ceil(256 / 12), raw 256-bit scalars
+// FB_C = 13 so the TOP window spans bits 247..255 (scalars < 2^255): its
+// digits spread over ~2^8 buckets instead of piling 4096 entries into a
+// handful (measured 7 ms of latency-bound straggler buckets at FB_C=12).
+constexpr int FB_C = 13;
+constexpr int FB_NWIN = 20;  // ceil(256 / 13), raw 256-bit scalars
 using CfgFB = msm_cfg<FB_C, FB_C>;  // NWIN=1: single merged window space
 
 // P_ext[w*n + i] = 2^(12w) * P_i (affine); inf entries follow the base flag
