@@ -1,22 +1,23 @@
 // ============================================================================
-// Pippenger bucket MSM over BN254 G1 — gfx950 kernels.
+// Pippenger bucket MSM — gfx950 kernels, curve- and geometry-templated
+// (BN254 G1 and BLS12-381 G1 on the 29-bit-limb field core).
 //
-// Window size c = 16 => 16 windows over the 254-bit scalar, 65535 live
-// buckets per window.  Pipeline per run:
-//   1. parse/reduce scalars (ark from_be_bytes_mod_order semantics; Fr stays
-//      on the 4x64 path — digit extraction wants packed u64 limbs)
-//   2. digit extraction -> (key = window<<16 | digit, value = point index)
-//   3. device radix sort of the 16n pairs on 20 key bits (rocPRIM)
-//   4. per-bucket segment offsets by binary search in the sorted keys
-//   5. bucket accumulation: one thread per bucket walks its run of sorted
-//      point indices with Jacobian+affine mixed adds — the VALU-bound hot
-//      kernel, on the 9x29-bit field core (gpu_field9.h)
-//   6. segment running sums -> weighted LDS-tree reduce (folds the 2^(16w)
-//      window factor) -> window sums -> 16-way combine + affine conversion.
+// Pipeline per run (window bits CB from msm_cfg; c=16 for large MSMs,
+// c=8 for n <= 2^16, FB_C=13 fixed-base for blob-KZG):
+//   1. parse scalars (BN254: ark from_be_bytes_mod_order reduction;
+//      BLS: raw 256-bit per blst — both as packed u64[4] for digits)
+//   2. digit extraction -> (key = window<<CB | digit, value = point index)
+//   3. device radix sort of the NWIN*n pairs (rocPRIM)
+//   4. per-bucket offsets by binary search + LENGTH-SORTED bucket schedule
+//      (waves then process similar-length runs: no Poisson divergence)
+//   5. bucket accumulation: one thread per scheduled bucket walks its run
+//      with XYZZ+affine mixed adds — the VALU-bound hot kernel
+//   6. segment running sums -> weighted reduce (folds the 2^(CB*w) window
+//      factor SIMD-wide) -> per-window LDS trees -> combine + affine.
 //
-// Work shape: ~16*(n + 2*65536) mixed adds; HBM traffic is only the
-// gathered 72-B points + sorted pairs => VALU-bound, so
-// there is deliberately no MFMA anywhere here.
+// Work shape: ~NWIN*(n + 2^(CB+1)) mixed adds; HBM traffic is only the
+// gathered points + sorted pairs => VALU-bound, so there
+// is deliberately no MFMA anywhere here.
 // ============================================================================
 #pragma once
 #include <hip/hip_runtime.h>
@@ -51,11 +52,6 @@ using CfgL254 = msm_cfg<16, 254>;  // BN254 large
 using CfgS254 = msm_cfg<8, 254>;   // BN254 small (n <= 2^16)
 using CfgL256 = msm_cfg<16, 256>;  // BLS large
 using CfgS256 = msm_cfg<8, 256>;   // BLS small
-
-// compatibility constants for the plan layer (max over configs)
-constexpr int MSM_NWIN_MAX = 33;                 // ceil(256/8) + 1 margin
-constexpr uint32_t MSM_NBUCKETS_MAX = (uint32_t)16 << 16;  // c=16 dominates
-constexpr int MSM_NPART_MAX = 16 * 8 > 32 * 32 ? 16 * 8 : 32 * 32;
 
 // digit w = bits [CB*w, CB*w+CB) of the scalar (spans u64 limbs)
 template <int CB>
