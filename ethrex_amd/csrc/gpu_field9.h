@@ -54,7 +54,6 @@ struct Fq9T {
     static constexpr const u32 (&ONE)[9] = bn254::FQ9_ONE;
     static constexpr const u64 (&MOD64)[4] = bn254::Fq::MOD;
     static constexpr u32 N0INV = bn254::FQ9_N0INV;
-    static constexpr u64 N0INV58 = bn254::FQ9_N0INV58;
 };
 struct Fr9T {
     static constexpr int L = 9;
@@ -67,7 +66,6 @@ struct Fr9T {
     static constexpr const u32 (&ONE)[9] = bn254::FR9_ONE;
     static constexpr const u64 (&MOD64)[4] = bn254::Fr::MOD;
     static constexpr u32 N0INV = bn254::FR9_N0INV;
-    static constexpr u64 N0INV58 = bn254::FR9_N0INV58;
 };
 struct FpB14T {
     static constexpr int L = 14;
@@ -80,7 +78,6 @@ struct FpB14T {
     static constexpr const u32 (&ONE)[14] = bn254::FPB_ONE;
     static constexpr const u64 (&MOD64)[6] = bn254::FPB_MOD64;
     static constexpr u32 N0INV = bn254::FPB_N0INV;
-    static constexpr u64 N0INV58 = bn254::FPB_N0INV58;
 };
 
 template <int LN>
@@ -237,31 +234,12 @@ __device__ __host__ __forceinline__ feL<T::L> mont_mul9(const feL<T::L> &A,
 #pragma unroll
         for (int j = 0; j < LN; j++) t[i + j] += (u64)A.v[i] * B.v[j];
     }
-    // Montgomery reduction in radix-2^58 double-rounds: one 58-bit quotient
-    // clears TWO limb positions per round, halving the serial m->fold->carry
-    // chain that dominates issue-stall (PMC: 57% WAIT_INST on the hot
-    // kernel).  Column bound: mul-phase products + <=2 quotient streams per
-    // column stay under 2^64 for the audited input ranges (header).
 #pragma unroll
-    for (int k = 0; k + 1 < LN; k += 2) {
-        u64 low58 = (t[k] + (t[k + 1] << 29)) & 0x3ffffffffffffffull;
-        u64 m58 = (low58 * T::N0INV58) & 0x3ffffffffffffffull;
-        u64 qlo = m58 & bn254::FQ9_MASK;
-        u64 qhi = m58 >> 29;
-#pragma unroll
-        for (int j = 0; j < LN; j++) {
-            t[k + j] += qlo * T::P[j];
-            t[k + j + 1] += qhi * T::P[j];
-        }
-        t[k + 1] += t[k] >> 29;        // low 58 bits ≡ 0 now
-        t[k + 2] += t[k + 1] >> 29;
-    }
-    if constexpr (LN & 1) {            // final single 29-bit round
-        constexpr int k = LN - 1;
+    for (int k = 0; k < LN; k++) {
         u32 m = ((u32)t[k] * T::N0INV) & bn254::FQ9_MASK;
 #pragma unroll
         for (int j = 0; j < LN; j++) t[k + j] += (u64)m * T::P[j];
-        t[k + 1] += t[k] >> 29;
+        t[k + 1] += t[k] >> 29;        // t[k] ≡ 0 mod 2^29 now
     }
     feL<LN> r;
     u64 c = 0;
