@@ -13,6 +13,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <string>
 #include <type_traits>
@@ -112,6 +113,15 @@ struct msm_plan_t {
     g1jT<C> *d_windows = nullptr;     // NWIN
     uint8_t *d_out = nullptr;         // JB
     uint32_t *d_err = nullptr;
+    // batch-affine pairing tree (large BN254 MSMs only; see msm_kernels.h)
+    bool use_tree = false;
+    int tree_levels = 0;
+    g1aT<C> *d_lvl[2] = {nullptr, nullptr};   // level ping/pong buffers
+    feL<F::L> *d_aux = nullptr;               // denominator prefix products
+    uint32_t *d_loff[2] = {nullptr, nullptr}; // level offsets ping/pong
+    uint32_t *d_cnt = nullptr;                // per-bucket pair counts
+    void *d_scan_tmp = nullptr;
+    size_t scan_tmp_bytes = 0;
     bool have_scalars = false;
     bool have_points = false;
     hipEvent_t ev[6];
@@ -143,6 +153,13 @@ static int msm_destroy_t(msm_plan_t<C> *p) {
     hipFree(p->d_windows);
     hipFree(p->d_out);
     hipFree(p->d_err);
+    hipFree(p->d_lvl[0]);
+    hipFree(p->d_lvl[1]);
+    hipFree(p->d_aux);
+    hipFree(p->d_loff[0]);
+    hipFree(p->d_loff[1]);
+    hipFree(p->d_cnt);
+    hipFree(p->d_scan_tmp);
     delete p;
     return EM_OK;
 }
@@ -201,6 +218,60 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
                                           (size_t)nbuckets, 0, 32);
         if (tmp_len > p->sort_tmp_bytes) p->sort_tmp_bytes = tmp_len;
         if (e == hipSuccess) e = hipMalloc(&p->d_sort_tmp, p->sort_tmp_bytes);
+    }
+    // batch-affine pairing tree: EXPERIMENTAL alternative bucket walk for
+    // large BN254 MSMs (opt-in via EM_MSM_TREE=1).  Parity-green, but the
+    // per-round bisect (profiles/r01_summary.md) measured it at 60 ms vs
+    // 29.8 ms for the flat XYZZ walk at 2^24: the two-pass level structure
+    // has strictly worse memory-level parallelism than the single prefetched
+    // XYZZ stream, and the batched-inversion chains serialize.  Kept for
+    // round-2 experiments; the product path stays XYZZ.
+    if constexpr (std::is_same_v<C, Bn254G1>) {
+        size_t avg = total / CFGL::NBUCKETS;
+        if (p->cbits == 16 && avg >= 8 && e == hipSuccess &&
+            std::getenv("EM_MSM_TREE") != nullptr) {
+            p->use_tree = true;
+            int lg = 0;
+            while ((avg >> lg) > 1) lg++;  // floor log2(avg run length)
+            p->tree_levels = lg - 2;
+            if (p->tree_levels < 1) p->tree_levels = 1;
+            if (p->tree_levels > 8) p->tree_levels = 8;
+            size_t c0 = total / 2 + CFGL::NBUCKETS + 1;
+            size_t c1 = total / 4 + CFGL::NBUCKETS + 1;
+            mal((void **)&p->d_lvl[0], c0 * sizeof(g1aT<C>));
+            mal((void **)&p->d_lvl[1], c1 * sizeof(g1aT<C>));
+            mal((void **)&p->d_aux,
+                (size_t)AUX_PLANES * c0 * sizeof(feL<msm_plan_t<C>::F::L>));
+            mal((void **)&p->d_loff[0], ((size_t)CFGL::NBUCKETS + 1) * 4);
+            mal((void **)&p->d_loff[1], ((size_t)CFGL::NBUCKETS + 1) * 4);
+            mal((void **)&p->d_cnt, ((size_t)CFGL::NBUCKETS + 1) * 4);
+            if (e == hipSuccess) {
+                e = rocprim::exclusive_scan(nullptr, p->scan_tmp_bytes,
+                                            p->d_cnt, p->d_loff[0], 0u,
+                                            (size_t)CFGL::NBUCKETS + 1);
+                if (e == hipSuccess)
+                    e = hipMalloc(&p->d_scan_tmp, p->scan_tmp_bytes);
+            }
+            if (e == hipErrorOutOfMemory) {
+                // tree buffers don't fit (very large n): fall back to the
+                // XYZZ bucket walk rather than failing plan creation
+                hipFree(p->d_lvl[0]);
+                hipFree(p->d_lvl[1]);
+                hipFree(p->d_aux);
+                hipFree(p->d_loff[0]);
+                hipFree(p->d_loff[1]);
+                hipFree(p->d_cnt);
+                hipFree(p->d_scan_tmp);
+                p->d_lvl[0] = p->d_lvl[1] = nullptr;
+                p->d_aux = nullptr;
+                p->d_loff[0] = p->d_loff[1] = nullptr;
+                p->d_cnt = nullptr;
+                p->d_scan_tmp = nullptr;
+                p->use_tree = false;
+                e = hipSuccess;
+                (void)hipGetLastError();
+            }
+        }
     }
     for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
     if (e != hipSuccess) {
@@ -352,10 +423,57 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
         if (e2 != hipSuccess) return hip_fail(e2, "bucket length sort");
     }
     HIP_TRY(hipEventRecord(p->ev[1], 0));
-    hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
-                       dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, 0,
-                       pts, p->d_vals_out, p->d_offsets, p->d_sched,
-                       p->d_buckets);
+    bool tree_done = false;
+    if constexpr (std::is_same_v<C, Bn254G1> && !FB && CFG::C == 16) {
+        if (p->use_tree) {
+            // batch-affine pairing tree, then XYZZ cleanup of the short tails
+            constexpr uint32_t NB = CFG::NBUCKETS;
+            uint32_t *o_in = p->d_offsets;
+            size_t bound = total;
+            for (int l = 0; l < p->tree_levels; l++) {
+                uint32_t *o_out = p->d_loff[l & 1];
+                hipLaunchKernelGGL(k_pair_counts,
+                                   dim3(blocks_for((size_t)NB + 1, 256)),
+                                   dim3(256), 0, 0, o_in, p->d_cnt, NB,
+                                   CFG::DMASK, l == 0 ? 1 : 0);
+                size_t st = p->scan_tmp_bytes;
+                hipError_t es = rocprim::exclusive_scan(
+                    p->d_scan_tmp, st, p->d_cnt, o_out, 0u, (size_t)NB + 1);
+                if (es != hipSuccess) return hip_fail(es, "pair scan");
+                size_t pbound = bound / 2 + NB;
+                size_t nthreads = (pbound + PAIR_K - 1) / PAIR_K;
+                g1aT<C> *dst = p->d_lvl[l & 1];
+                uint32_t cap =
+                    (uint32_t)(total / 2 + CFG::NBUCKETS + 1);
+                if (l == 0) {
+                    hipLaunchKernelGGL((k_pair_level<C, true>),
+                                       dim3(blocks_for(nthreads, 256)),
+                                       dim3(256), 0, 0, p->d_pts,
+                                       p->d_vals_out, o_in, o_out, p->d_aux,
+                                       cap, dst, NB);
+                } else {
+                    hipLaunchKernelGGL((k_pair_level<C, false>),
+                                       dim3(blocks_for(nthreads, 256)),
+                                       dim3(256), 0, 0, p->d_lvl[(l & 1) ^ 1],
+                                       nullptr, o_in, o_out, p->d_aux, cap,
+                                       dst, NB);
+                }
+                o_in = o_out;
+                bound = pbound;
+            }
+            hipLaunchKernelGGL((k_bucket_acc<C, CFG, false>),
+                               dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256),
+                               0, 0, p->d_lvl[(p->tree_levels - 1) & 1],
+                               nullptr, o_in, p->d_sched, p->d_buckets);
+            tree_done = true;
+        }
+    }
+    if (!tree_done) {
+        hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
+                           dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0,
+                           0, pts, p->d_vals_out, p->d_offsets, p->d_sched,
+                           p->d_buckets);
+    }
     HIP_TRY(hipEventRecord(p->ev[2], 0));
     hipLaunchKernelGGL((k_segment_reduce<C, CFG>),
                        dim3(blocks_for(CFG::NWIN * CFG::NSEG, 256)), dim3(256),

@@ -180,7 +180,9 @@ __global__ void k_bucket_lengths(const uint32_t *__restrict__ offsets,
 
 // ---- bucket accumulation (the hot kernel) ----
 // one thread per SCHEDULED bucket id; digit-0 buckets skipped.
-template <typename C, typename CFG>
+// GATHER=true reads pts[vals[t]] (sorted-pair gather); GATHER=false reads
+// the pairing-tree level buffer directly (vals may be null then).
+template <typename C, typename CFG, bool GATHER = true>
 __global__ void __launch_bounds__(256)
 k_bucket_acc(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
              const uint32_t *__restrict__ offsets,
@@ -197,14 +199,211 @@ k_bucket_acc(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
     }
     // software pipeline: issue the NEXT point's gather before the long mixed
     // add so the dependent idx->point load chain overlaps the VALU work.
-    g1aT<C> p = pts[vals[lo]];
+    g1aT<C> p = GATHER ? pts[vals[lo]] : pts[lo];
     for (uint32_t t = lo; t < hi; t++) {
         g1aT<C> cur = p;
         uint32_t nxt = t + 1 < hi ? t + 1 : t;
-        p = pts[vals[nxt]];
+        p = GATHER ? pts[vals[nxt]] : pts[nxt];
+        if constexpr (!GATHER) {
+            // tree outputs may be the identity (P + (-P)): encoded (0,0)
+            if (fe9_is_zero_raw<C::F::L>(cur.x) &&
+                fe9_is_zero_raw<C::F::L>(cur.y))
+                continue;
+        }
         acc = g1_add_affine9(acc, cur);
     }
     buckets[b] = acc;
+}
+
+// ---- batch-affine pairing tree (large BN254 MSMs) ----
+// The XYZZ mixed add above costs ~13 Montgomery muls per point add.  For
+// large MSMs the bucket runs are long (avg n/2^16 = 256 at 2^24), so the
+// bucket sums can instead be built as a level-synchronized pairing tree of
+// AFFINE adds: lambda = (y2-y1)/(x2-x1), x3 = lambda^2-x1-x2,
+// y3 = lambda(x1-x3)-y1 — ~6 muls per add once the inversion is batched.
+// Each thread processes PAIR_K consecutive pair slots per level: a forward
+// pass chains denominator prefix-products (Montgomery's trick, products
+// stored to an aux buffer), ONE Fermat inversion per thread (amortized
+// ~1.2 muls/add at PAIR_K=256, and fully parallel across threads), then a
+// backward pass peels per-pair inverses and writes the sums.  Identity /
+// doubling / P+(-P) / odd-singleton cases ride the same batch with an
+// identity denominator (doubling contributes its real 2y denominator, so
+// it needs no separate inversion).  After the tree levels shrink runs to
+// <= ~4, the XYZZ kernel (GATHER=false) finishes against the level buffer.
+#ifndef EM_PAIR_K
+#define EM_PAIR_K 256
+#endif
+constexpr int PAIR_K = EM_PAIR_K;  // pair slots per thread per level
+
+// per-bucket pair counts for the next level: ceil(len/2); digit-0 buckets
+// contribute nothing (skip_d0 set at level 0; empty thereafter).
+__global__ void k_pair_counts(const uint32_t *__restrict__ off_in,
+                              uint32_t *__restrict__ cnt, uint32_t nbuckets,
+                              uint32_t dmask, int skip_d0) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b > nbuckets) return;
+    if (b == nbuckets) {
+        cnt[b] = 0;
+        return;
+    }
+    if (skip_d0 && (b & dmask) == 0) {
+        cnt[b] = 0;
+        return;
+    }
+    cnt[b] = (off_in[b + 1] - off_in[b] + 1) >> 1;
+}
+
+// classification of one pair slot; d/n are only meaningful when kind==0
+template <typename C>
+struct pair_case_t {
+    feL<C::F::L> d, n;   // lambda = n / d
+    g1aT<C> copy;        // kind==1 result
+    int kind;            // 0 = lambda path, 1 = copy/identity result
+};
+
+template <typename C>
+__device__ __forceinline__ bool g1a_is_inf(const g1aT<C> &p) {
+    using T = typename C::F;
+    return fe9_is_zero_modp<T>(p.x) && fe9_is_zero_modp<T>(p.y);
+}
+
+// classify pair (A,B); sgl means B absent (odd tail)
+template <typename C>
+__device__ __forceinline__ pair_case_t<C> pair_classify(const g1aT<C> &A,
+                                                        const g1aT<C> &B,
+                                                        bool sgl) {
+    using T = typename C::F;
+    pair_case_t<C> r;
+    if (sgl || g1a_is_inf<C>(B)) {
+        r.kind = 1;
+        r.copy = A;
+        return r;
+    }
+    if (g1a_is_inf<C>(A)) {
+        r.kind = 1;
+        r.copy = B;
+        return r;
+    }
+    if (fe9_eq_modp<T>(A.x, B.x)) {
+        if (fe9_eq_modp<T>(A.y, B.y) && !fe9_is_zero_modp<T>(A.y)) {
+            // doubling: lambda = 3 x1^2 / 2 y1
+            r.kind = 0;
+            r.d = add9_n<T>(A.y, A.y);
+            feL<C::F::L> s = mont_sqr9<T>(A.x);
+            r.n = add9_n<T>(add9_n<T>(s, s), s);
+            return r;
+        }
+        r.kind = 1;  // B = -A (or 2-torsion): identity
+        r.copy.x = fe9z<C::F::L>();
+        r.copy.y = fe9z<C::F::L>();
+        return r;
+    }
+    r.kind = 0;
+    r.d = subn9<T>(B.x, A.x);
+    r.n = subn9<T>(B.y, A.y);
+    return r;
+}
+
+// largest b with off[b] <= j (off non-decreasing, off[0]=0)
+__device__ __forceinline__ uint32_t bucket_of(const uint32_t *__restrict__ off,
+                                              uint32_t nbuckets, uint32_t j) {
+    uint32_t lo = 0, hi = nbuckets;
+    while (lo < hi) {
+        uint32_t mid = (lo + hi + 1) >> 1;
+        if (off[mid] <= j)
+            lo = mid;
+        else
+            hi = mid - 1;
+    }
+    return lo;
+}
+
+// one tree level.  L0 gathers pts[vals[.]] (the sorted-pair view of the
+// input points); deeper levels read the previous level buffer directly.
+//
+// Pair assignment is LANE-STRIDED: a wave's 64 lanes cover 64 consecutive
+// pair slots per step (lane l owns pairs base + i*64 + l), so level-buffer
+// reads, SoA stores and out[] writes all coalesce — the per-thread
+// contiguous-chunk version measured 44% of wave cycles parked on memory
+// waits.  The forward pass classifies, chains the denominator prefix
+// products and stores an SoA record (d, n, s=x1+x2, x1, y1) per lambda
+// pair; copy-like pairs (identity operands, P+(-P), odd singleton) write
+// out[] immediately and flag d=0.  The backward pass then needs NO point
+// gathers and no re-classification: it peels inverses from the aux plane
+// and finishes x3 = lam^2 - s, y3 = lam(x1-x3) - y1.
+enum { AUX_PP = 0, AUX_D, AUX_N, AUX_S, AUX_X1, AUX_Y1, AUX_PLANES };
+
+template <typename C, bool L0>
+__global__ void __launch_bounds__(256)
+k_pair_level(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
+             const uint32_t *__restrict__ off_in,
+             const uint32_t *__restrict__ off_out,
+             feL<C::F::L> *__restrict__ aux, uint32_t cap,
+             g1aT<C> *__restrict__ out, uint32_t nbuckets) {
+    using T = typename C::F;
+    constexpr int LN = C::F::L;
+    uint32_t total = off_out[nbuckets];
+    uint32_t lane = threadIdx.x & 63u;
+    uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    uint32_t base = wave * 64u * (uint32_t)PAIR_K;
+    if (base >= total) return;
+    feL<LN> *PPp = aux + (size_t)AUX_PP * cap;
+    feL<LN> *Dp = aux + (size_t)AUX_D * cap;
+    feL<LN> *Np = aux + (size_t)AUX_N * cap;
+    feL<LN> *Sp = aux + (size_t)AUX_S * cap;
+    feL<LN> *X1p = aux + (size_t)AUX_X1 * cap;
+    feL<LN> *Y1p = aux + (size_t)AUX_Y1 * cap;
+    uint32_t jf = base + lane;
+    uint32_t bb = bucket_of(off_out, nbuckets, jf < total ? jf : total - 1);
+    uint32_t ob = off_out[bb], oe = off_out[bb + 1];
+    uint32_t ib = off_in[bb], ie = off_in[bb + 1];
+    // forward: classify, store SoA records, chain prefix products
+    feL<LN> PP = fe9_load<LN>(T::ONE);
+    for (int i = 0; i < PAIR_K; i++) {
+        uint32_t j = base + (uint32_t)i * 64u + lane;
+        if (j >= total) break;
+        while (j >= oe) {
+            bb++;
+            ob = oe;
+            oe = off_out[bb + 1];
+            ib = ie;
+            ie = off_in[bb + 1];
+        }
+        uint32_t s0 = ib + 2 * (j - ob);
+        bool sgl = s0 + 1 >= ie;
+        g1aT<C> A = L0 ? pts[vals[s0]] : pts[s0];
+        g1aT<C> B;
+        if (!sgl) B = L0 ? pts[vals[s0 + 1]] : pts[s0 + 1];
+        pair_case_t<C> pc = pair_classify<C>(A, B, sgl);
+        if (pc.kind == 0) {
+            PP = mont_mul9<T>(PP, pc.d);
+            Dp[j] = pc.d;
+            Np[j] = pc.n;
+            Sp[j] = add9_n<T>(A.x, B.x);
+            X1p[j] = A.x;
+            Y1p[j] = A.y;
+        } else {
+            out[j] = pc.copy;
+            Dp[j] = fe9z<LN>();  // d = 0 flags "done in forward"
+        }
+        PPp[j] = PP;
+    }
+    feL<LN> inv = mont_inv9<T>(PP);
+    // backward: peel per-pair inverses from the SoA planes, emit sums
+    for (int i = PAIR_K; i-- > 0;) {
+        uint32_t j = base + (uint32_t)i * 64u + lane;
+        if (j >= total) continue;
+        feL<LN> d = Dp[j];
+        if (fe9_is_zero_raw<LN>(d)) continue;
+        feL<LN> invd = i > 0 ? mont_mul9<T>(inv, PPp[j - 64]) : inv;
+        feL<LN> lam = mont_mul9<T>(Np[j], invd);
+        feL<LN> x3 = subn9<T>(mont_sqr9<T>(lam), Sp[j]);
+        feL<LN> y3 =
+            subn9<T>(mont_mul9<T>(lam, subn9<T>(X1p[j], x3)), Y1p[j]);
+        out[j].x = x3;
+        out[j].y = y3;
+        inv = mont_mul9<T>(inv, d);
+    }
 }
 
 // ---- two-level running-sum reduction ----

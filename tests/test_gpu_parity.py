@@ -195,6 +195,45 @@ def test_gpu_msm_parity_2_20_direct(gpu, oracle_mod):
     assert rc == 0 and got == want
 
 
+def test_gpu_msm_tree_edge_cases(gpu, oracle_mod, monkeypatch):
+    """Exercises the EXPERIMENTAL batch-affine pairing tree (EM_MSM_TREE=1;
+    active for n > 2^16 with avg bucket runs >= 8): odd n, heavy duplicate
+    chains (doubling denominators in the batch), identity points and zero
+    scalars.  The same inputs are then re-run on the default XYZZ path."""
+    monkeypatch.setenv("EM_MSM_TREE", "1")
+    n = (1 << 19) + 5
+    base_n = 8
+    plan0 = gpu.MsmPlan(base_n)
+    plan0.gen_points(7)
+    base = plan0.download_points()
+    plan0.destroy()
+    pts = bytearray()
+    for i in range(n):
+        if i % 4099 == 0:
+            pts += b"\x00" * 64  # identity point
+        else:
+            j = (i * i) % base_n  # long duplicate runs within buckets
+            pts += base[64 * j:64 * j + 64]
+    scs = bytearray(oracle_mod.gen_fr(77, n))
+    for i in range(0, n, 2048):
+        scs[32 * i:32 * i + 32] = b"\x00" * 32  # zero scalars
+    rc, want = oracle_mod.g1_msm(bytes(pts), bytes(scs), n)
+    assert rc == 0
+    plan = gpu.MsmPlan(n)  # EM_MSM_TREE=1: batch-affine tree path
+    plan.upload_points(bytes(pts))
+    plan.upload_scalars(bytes(scs))
+    got_tree = plan.run()
+    plan.destroy()
+    assert got_tree == want
+    monkeypatch.delenv("EM_MSM_TREE")
+    plan = gpu.MsmPlan(n)  # default XYZZ path, same inputs
+    plan.upload_points(bytes(pts))
+    plan.upload_scalars(bytes(scs))
+    got_xyzz = plan.run()
+    plan.destroy()
+    assert got_xyzz == want
+
+
 def test_gpu_plan_combine_matches_standalone(gpu, oracle_mod):
     """plan-attached combine (the per-step N>1 exchange path) matches the
     standalone combine and the oracle."""
