@@ -402,11 +402,29 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
                            dim3(256), 0, 0, p->d_scalars, p->d_inf, p->d_keys,
                            p->d_vals, p->n);
     }
-    size_t tmp = p->sort_tmp_bytes;
-    hipError_t e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
-                                             p->d_keys_out, p->d_vals,
-                                             p->d_vals_out, total, 0,
-                                             CFG::SORT_BITS);
+    // The window bits of a key are constant within each window's segment
+    // [w*n, (w+1)*n), so for large MSMs NWIN per-window sorts over just the
+    // digit bits replace one global sort over digit+window bits — one fewer
+    // radix pass over the full 16n pair array.  Small MSMs keep the single
+    // sort (per-call overhead dominates at small n).
+    hipError_t e = hipSuccess;
+    constexpr int DBITS = CFG::C;  // digit bits (FB: k_fb_digits uses FB_C)
+    const int nwin = FB ? FB_NWIN : CFG::NWIN;
+    if (!FB && p->n >= (1u << 18)) {
+        for (int w = 0; w < nwin && e == hipSuccess; w++) {
+            size_t tmp = p->sort_tmp_bytes;
+            size_t off = (size_t)w * p->n;
+            e = rocprim::radix_sort_pairs(
+                p->d_sort_tmp, tmp, p->d_keys + off, p->d_keys_out + off,
+                p->d_vals + off, p->d_vals_out + off, p->n, 0, DBITS);
+        }
+    } else {
+        size_t tmp = p->sort_tmp_bytes;
+        e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys,
+                                      p->d_keys_out, p->d_vals,
+                                      p->d_vals_out, total, 0,
+                                      CFG::SORT_BITS);
+    }
     if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs");
     hipLaunchKernelGGL((k_offsets<CFG>),
                        dim3(blocks_for((size_t)CFG::NBUCKETS + 1, 256)),
