@@ -126,11 +126,34 @@ struct msm_plan_t {
     bool have_points = false;
     hipEvent_t ev[6];
     double last_ms[5] = {0, 0, 0, 0, 0};
+    // async pipelined path: the sort chain (HBM-heavy) of step k+1 runs on
+    // s_sort concurrently with the compute chain (VALU-heavy bucket walk +
+    // reduction) of step k on s_comp.  Sort outputs are double-buffered by
+    // step parity; depth-2 backpressure via ev_comp_done.
+    hipStream_t s_sort = nullptr, s_comp = nullptr;
+    uint32_t *d_keys_out2 = nullptr, *d_vals_out2 = nullptr;
+    uint32_t *d_offsets2 = nullptr, *d_sched2 = nullptr;
+    uint8_t *d_out2 = nullptr;
+    uint8_t *h_out[2] = {nullptr, nullptr};   // pinned result staging
+    hipEvent_t ev_sort_done[2], ev_comp_done[2];
+    struct {
+        uint8_t *dest;
+        int bytes;
+        bool valid;
+    } pend[2] = {{nullptr, 0, false}, {nullptr, 0, false}};
+    int apar = 0;
 };
+
+template <typename C>
+static int msm_sync_t(msm_plan_t<C> *p);
 
 template <typename C>
 static int msm_destroy_t(msm_plan_t<C> *p) {
     if (!p) return EM_ERR_INPUT;
+    if (p->s_comp) {
+        (void)hipStreamSynchronize(p->s_sort);
+        (void)hipStreamSynchronize(p->s_comp);
+    }
     hipFree(p->d_pts);
     hipFree(p->d_pts_ext);
     hipFree(p->d_inf);
@@ -160,6 +183,15 @@ static int msm_destroy_t(msm_plan_t<C> *p) {
     hipFree(p->d_loff[1]);
     hipFree(p->d_cnt);
     hipFree(p->d_scan_tmp);
+    hipFree(p->d_keys_out2);
+    hipFree(p->d_vals_out2);
+    hipFree(p->d_offsets2);
+    hipFree(p->d_sched2);
+    hipFree(p->d_out2);
+    if (p->h_out[0]) hipHostFree(p->h_out[0]);
+    if (p->h_out[1]) hipHostFree(p->h_out[1]);
+    if (p->s_sort) hipStreamDestroy(p->s_sort);
+    if (p->s_comp) hipStreamDestroy(p->s_comp);
     delete p;
     return EM_OK;
 }
@@ -274,6 +306,28 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
         }
     }
     for (int i = 0; i < 6 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
+    // async pipelined path: second sort-output buffer set + streams
+    mal((void **)&p->d_keys_out2, total * 4);
+    mal((void **)&p->d_vals_out2, total * 4);
+    mal((void **)&p->d_offsets2, ((size_t)nbuckets + 1) * 4);
+    mal((void **)&p->d_sched2, (size_t)nbuckets * 4);
+    mal((void **)&p->d_out2, msm_plan_t<C>::JB);
+    if (e == hipSuccess)
+        e = hipStreamCreateWithFlags(&p->s_sort, hipStreamNonBlocking);
+    if (e == hipSuccess)
+        e = hipStreamCreateWithFlags(&p->s_comp, hipStreamNonBlocking);
+    for (int i = 0; i < 2 && e == hipSuccess; i++) {
+        if (e == hipSuccess)
+            e = hipHostMalloc((void **)&p->h_out[i], msm_plan_t<C>::JB);
+        if (e == hipSuccess)
+            e = hipEventCreateWithFlags(&p->ev_sort_done[i],
+                                        hipEventDisableTiming);
+        if (e == hipSuccess)
+            e = hipEventCreateWithFlags(&p->ev_comp_done[i],
+                                        hipEventDisableTiming);
+        // record once so the first hipStreamWaitEvent sees a signaled event
+        if (e == hipSuccess) e = hipEventRecord(p->ev_comp_done[i], 0);
+    }
     if (e != hipSuccess) {
         msm_destroy_t(p);
         return hip_fail(e, "msm_plan_create");
@@ -285,6 +339,8 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
 template <typename C>
 static int msm_upload_points_t(msm_plan_t<C> *p, const uint8_t *points) {
     if (!p || !points) return EM_ERR_INPUT;
+    int rc0 = msm_sync_t(p);
+    if (rc0) return rc0;
     constexpr int PB = msm_plan_t<C>::PB;
     HIP_TRY(hipMemset(p->d_err, 0, 4));
     HIP_TRY(hipMemcpy(p->d_scratch, points, p->n * PB, hipMemcpyHostToDevice));
@@ -372,6 +428,8 @@ static int msm_precompute_t(msm_plan_t<C> *p) {
 template <typename C>
 static int msm_upload_scalars_t(msm_plan_t<C> *p, const uint8_t *scalars32) {
     if (!p || !scalars32) return EM_ERR_INPUT;
+    int rc0 = msm_sync_t(p);
+    if (rc0) return rc0;
     HIP_TRY(hipMemcpy(p->d_scratch, scalars32, p->n * 32, hipMemcpyHostToDevice));
     if constexpr (std::is_same_v<C, BlsG1>) {
         // raw 256-bit scalars, no reduction (blst SCALAR_BITS = 256)
@@ -524,6 +582,127 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     return EM_OK;
 }
 
+// deliver a completed pipelined result to its caller's buffer
+template <typename C>
+static int msm_deliver(msm_plan_t<C> *p, int par) {
+    if (!p->pend[par].valid) return EM_OK;
+    HIP_TRY(hipEventSynchronize(p->ev_comp_done[par]));
+    memcpy(p->pend[par].dest, p->h_out[par], p->pend[par].bytes);
+    p->pend[par].valid = false;
+    return EM_OK;
+}
+
+// drain the pipeline (also called before any sync-path operation)
+template <typename C>
+static int msm_sync_t(msm_plan_t<C> *p) {
+    if (!p) return EM_ERR_INPUT;
+    if (!p->s_comp) return EM_OK;
+    HIP_TRY(hipStreamSynchronize(p->s_sort));
+    HIP_TRY(hipStreamSynchronize(p->s_comp));
+    int rc = msm_deliver(p, p->apar ^ 1);
+    if (rc) return rc;
+    return msm_deliver(p, p->apar);
+}
+
+// one pipelined step: sort chain on s_sort (buffers chosen by step parity),
+// compute chain on s_comp ordered behind it by event.  Returns after
+// ENQUEUE; the result lands in `out` by the time msm_sync (or the depth-2
+// backpressure of a later run_async) returns.  The proving loop runs many
+// MSMs back-to-back, so steady-state cost = max(sort chain, compute chain)
+// instead of their sum.
+template <typename C, typename CFG>
+static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
+    constexpr bool FB = std::is_same_v<CFG, CfgFB>;
+    size_t total = FB ? p->n * (size_t)FB_NWIN : p->n * (size_t)CFG::NWIN;
+    const g1aT<C> *pts = FB ? p->d_pts_ext : p->d_pts;
+    int par = p->apar;
+    int rc = msm_deliver(p, par);  // free this parity's slots (step k-2)
+    if (rc) return rc;
+    uint32_t *KO = par ? p->d_keys_out2 : p->d_keys_out;
+    uint32_t *VO = par ? p->d_vals_out2 : p->d_vals_out;
+    uint32_t *OFF = par ? p->d_offsets2 : p->d_offsets;
+    uint32_t *SCH = par ? p->d_sched2 : p->d_sched;
+    uint8_t *DOUT = par ? p->d_out2 : p->d_out;
+    hipStream_t ss = p->s_sort, sc = p->s_comp;
+    // ---- sort chain ----
+    HIP_TRY(hipStreamWaitEvent(ss, p->ev_comp_done[par], 0));
+    if constexpr (FB) {
+        hipLaunchKernelGGL(k_fb_digits, dim3(blocks_for(total, 256)),
+                           dim3(256), 0, ss, p->d_scalars, p->d_inf,
+                           p->d_keys, p->d_vals, p->n);
+    } else {
+        hipLaunchKernelGGL((k_digits<CFG>), dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, ss, p->d_scalars, p->d_inf,
+                           p->d_keys, p->d_vals, p->n);
+    }
+    hipError_t e = hipSuccess;
+    if (!FB && p->n >= (1u << 18)) {
+        for (int w = 0; w < CFG::NWIN && e == hipSuccess; w++) {
+            size_t tmp = p->sort_tmp_bytes;
+            size_t off = (size_t)w * p->n;
+            e = rocprim::radix_sort_pairs(
+                p->d_sort_tmp, tmp, p->d_keys + off, KO + off,
+                p->d_vals + off, VO + off, p->n, 0, CFG::C, ss);
+        }
+    } else {
+        size_t tmp = p->sort_tmp_bytes;
+        e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys, KO,
+                                      p->d_vals, VO, total, 0,
+                                      CFG::SORT_BITS, ss);
+    }
+    if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs (async)");
+    hipLaunchKernelGGL((k_offsets<CFG>),
+                       dim3(blocks_for((size_t)CFG::NBUCKETS + 1, 256)),
+                       dim3(256), 0, ss, KO, total, OFF);
+    hipLaunchKernelGGL((k_bucket_lengths<CFG>),
+                       dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, ss,
+                       OFF, p->d_blen, p->d_bids);
+    {
+        size_t tmp2 = p->sort_tmp_bytes;
+        e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp2, p->d_blen,
+                                      p->d_blen_out, p->d_bids, SCH,
+                                      (size_t)CFG::NBUCKETS, 0, 32, ss);
+        if (e != hipSuccess) return hip_fail(e, "bucket length sort (async)");
+    }
+    HIP_TRY(hipEventRecord(p->ev_sort_done[par], ss));
+    // ---- compute chain ----
+    HIP_TRY(hipStreamWaitEvent(sc, p->ev_sort_done[par], 0));
+    hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
+                       dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, sc,
+                       pts, VO, OFF, SCH, p->d_buckets);
+    hipLaunchKernelGGL((k_segment_reduce<C, CFG>),
+                       dim3(blocks_for(CFG::NWIN * CFG::NSEG, 256)), dim3(256),
+                       0, sc, p->d_buckets, p->d_seg_sum, p->d_seg_wsum);
+    hipLaunchKernelGGL((k_weighted_reduce<C, CFG>),
+                       dim3(blocks_for(CFG::NWIN * CFG::NSEG, CFG::RED_BLOCK)),
+                       dim3(CFG::RED_BLOCK), 0, sc, p->d_seg_sum,
+                       p->d_seg_wsum, p->d_partials);
+    hipLaunchKernelGGL((k_window_sum<C, CFG>), dim3(CFG::NWIN), dim3(64), 0,
+                       sc, p->d_partials, p->d_windows);
+    hipLaunchKernelGGL((k_final_combine<C, CFG>), dim3(1), dim3(64), 0, sc,
+                       p->d_windows, DOUT, out_mode);
+    int bytes = out_mode == 0 ? msm_plan_t<C>::AB : msm_plan_t<C>::JB;
+    HIP_TRY(hipMemcpyAsync(p->h_out[par], DOUT, bytes,
+                           hipMemcpyDeviceToHost, sc));
+    HIP_TRY(hipEventRecord(p->ev_comp_done[par], sc));
+    p->pend[par] = {out, bytes, true};
+    p->apar ^= 1;
+    return EM_OK;
+}
+
+template <typename C>
+static int msm_run_async_t(msm_plan_t<C> *p, uint8_t *out) {
+    if (!p || !out) return EM_ERR_INPUT;
+    if (!p->have_points || !p->have_scalars) {
+        g_last_err = "msm_run_async: points/scalars not uploaded";
+        return EM_ERR_INPUT;
+    }
+    if (p->fixed_base) return msm_run_async_cfg<C, CfgFB>(p, out, 0);
+    if (p->cbits == 8)
+        return msm_run_async_cfg<C, msm_cfg<8, msm_plan_t<C>::SB>>(p, out, 0);
+    return msm_run_async_cfg<C, msm_cfg<16, msm_plan_t<C>::SB>>(p, out, 0);
+}
+
 template <typename C>
 static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     if (!p || !out) return EM_ERR_INPUT;
@@ -531,6 +710,8 @@ static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
         g_last_err = "msm_run: points/scalars not uploaded";
         return EM_ERR_INPUT;
     }
+    int rc = msm_sync_t(p);  // drain any pipelined steps first
+    if (rc) return rc;
     if (p->fixed_base) return msm_run_cfg<C, CfgFB>(p, out, out_mode);
     if (p->cbits == 8)
         return msm_run_cfg<C, msm_cfg<8, msm_plan_t<C>::SB>>(p, out, out_mode);
@@ -563,6 +744,12 @@ extern "C" int ethrex_mi355_msm_upload_scalars(em_msm_plan *p,
 }
 extern "C" int ethrex_mi355_msm_run(em_msm_plan *p, uint8_t out[64]) {
     return msm_run_inner_t((msm_plan_t<Bn254G1> *)p, out, 0);
+}
+extern "C" int ethrex_mi355_msm_run_async(em_msm_plan *p, uint8_t out[64]) {
+    return msm_run_async_t((msm_plan_t<Bn254G1> *)p, out);
+}
+extern "C" int ethrex_mi355_msm_sync(em_msm_plan *p) {
+    return msm_sync_t((msm_plan_t<Bn254G1> *)p);
 }
 extern "C" int ethrex_mi355_msm_run_partial(em_msm_plan *p, uint8_t out[96]) {
     return msm_run_inner_t((msm_plan_t<Bn254G1> *)p, out, 1);
@@ -612,6 +799,13 @@ extern "C" int ethrex_mi355_bls_msm_upload_scalars(em_bls_msm_plan *p,
 }
 extern "C" int ethrex_mi355_bls_msm_run(em_bls_msm_plan *p, uint8_t out[96]) {
     return msm_run_inner_t((msm_plan_t<BlsG1> *)p, out, 0);
+}
+extern "C" int ethrex_mi355_bls_msm_run_async(em_bls_msm_plan *p,
+                                              uint8_t out[96]) {
+    return msm_run_async_t((msm_plan_t<BlsG1> *)p, out);
+}
+extern "C" int ethrex_mi355_bls_msm_sync(em_bls_msm_plan *p) {
+    return msm_sync_t((msm_plan_t<BlsG1> *)p);
 }
 extern "C" int ethrex_mi355_bls_msm_run_partial(em_bls_msm_plan *p,
                                                 uint8_t out[144]) {

@@ -28,7 +28,8 @@ for _f in ("device_count", "set_device", "bn254_g1_add", "bn254_g1_mul",
            "bn254_g1_msm", "bn254_fr_ntt", "bn254_g1_combine",
            "msm_plan_create", "msm_plan_destroy", "msm_upload_points",
            "msm_gen_points", "msm_download_points", "msm_upload_scalars",
-           "msm_run", "msm_run_partial", "msm_last_times", "msm_combine",
+           "msm_run", "msm_run_partial", "msm_run_async", "msm_sync",
+           "msm_last_times", "msm_combine",
            "ntt_plan_create", "ntt_plan_destroy", "ntt_upload", "ntt_run",
            "ntt_download", "ntt_last_times"):
     getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
@@ -113,6 +114,7 @@ class MsmPlan:
     def __init__(self, n: int):
         self.n = n
         self._p = ctypes.c_void_p()
+        self._pending = []
         _check(_lib.ethrex_mi355_msm_plan_create(ctypes.c_size_t(n),
                                                  ctypes.byref(self._p)),
                "msm_plan_create")
@@ -150,6 +152,20 @@ class MsmPlan:
         _check(_lib.ethrex_mi355_msm_last_times(self._p, t), "msm_last_times")
         return {"digits_sort_ms": t[0], "bucket_acc_ms": t[1],
                 "reduce_ms": t[2], "combine_ms": t[3], "total_ms": t[4]}
+
+    def run_async(self):
+        """Enqueue one pipelined MSM step (sort chain of the next step
+        overlaps this step's compute chain).  Result delivered at sync()."""
+        buf = (ctypes.c_uint8 * 64)()
+        _check(_lib.ethrex_mi355_msm_run_async(self._p, buf), "msm_run_async")
+        self._pending.append(buf)
+
+    def sync(self) -> bytes:
+        """Drain the pipeline; returns the LAST pipelined result."""
+        _check(_lib.ethrex_mi355_msm_sync(self._p), "msm_sync")
+        outs = [bytes(b) for b in self._pending]
+        self._pending = []
+        return outs[-1] if outs else b""
 
     def combine(self, jacobians: bytes, count: int) -> bytes:
         """combine Jacobian partials reusing this plan's device buffers"""
@@ -215,6 +231,7 @@ for _f in ("bls12381_g1_add", "bls12381_g1_mul", "bls12381_g1_msm",
            "bls12381_g1_combine", "bls_msm_plan_create", "bls_msm_plan_destroy",
            "bls_msm_upload_points", "bls_msm_gen_points",
            "bls_msm_download_points", "bls_msm_upload_scalars", "bls_msm_run",
+           "bls_msm_run_async", "bls_msm_sync",
            "bls_msm_run_partial", "bls_msm_last_times", "bls_msm_precompute"):
     getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
 
@@ -257,6 +274,7 @@ class BlsMsmPlan:
     def __init__(self, n: int):
         self.n = n
         self._p = ctypes.c_void_p()
+        self._pending = []
         _check(_lib.ethrex_mi355_bls_msm_plan_create(ctypes.c_size_t(n),
                                                      ctypes.byref(self._p)),
                "bls_msm_plan_create")
@@ -295,6 +313,18 @@ class BlsMsmPlan:
         _check(_lib.ethrex_mi355_bls_msm_run_partial(self._p, out),
                "bls_msm_run_partial")
         return bytes(out)
+
+    def run_async(self):
+        buf = (ctypes.c_uint8 * 96)()
+        _check(_lib.ethrex_mi355_bls_msm_run_async(self._p, buf),
+               "bls_msm_run_async")
+        self._pending.append(buf)
+
+    def sync(self) -> bytes:
+        _check(_lib.ethrex_mi355_bls_msm_sync(self._p), "bls_msm_sync")
+        outs = [bytes(b) for b in self._pending]
+        self._pending = []
+        return outs[-1] if outs else b""
 
     def last_times(self):
         t = (ctypes.c_double * 5)()

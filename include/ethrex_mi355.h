@@ -80,6 +80,14 @@ int ethrex_mi355_msm_upload_scalars(em_msm_plan *plan, const uint8_t *scalars32)
 int ethrex_mi355_msm_run(em_msm_plan *plan, uint8_t out[64]);
 /* shard partial -> Jacobian 96 B (multi-GPU: AllGather these, then combine) */
 int ethrex_mi355_msm_run_partial(em_msm_plan *plan, uint8_t out[96]);
+/* pipelined MSM step: enqueues the step and returns; the sort chain of the
+ * NEXT run_async overlaps this step's bucket/reduction chain on a second
+ * HIP stream (the proving loop runs many MSMs back-to-back).  `out` is
+ * filled by the time ethrex_mi355_msm_sync — or the depth-2 backpressure of
+ * a later run_async — returns; keep it valid until then. */
+int ethrex_mi355_msm_run_async(em_msm_plan *plan, uint8_t out[64]);
+/* drain all pipelined steps and deliver their results */
+int ethrex_mi355_msm_sync(em_msm_plan *plan);
 /* combine Jacobian partials on the GPU -> affine result */
 int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96, size_t count,
                                   uint8_t out[64]);
@@ -138,6 +146,8 @@ int ethrex_mi355_bls_msm_upload_scalars(em_bls_msm_plan *plan,
 int ethrex_mi355_bls_msm_precompute(em_bls_msm_plan *plan);
 int ethrex_mi355_bls_msm_run(em_bls_msm_plan *plan, uint8_t out[96]);
 int ethrex_mi355_bls_msm_run_partial(em_bls_msm_plan *plan, uint8_t out[144]);
+int ethrex_mi355_bls_msm_run_async(em_bls_msm_plan *plan, uint8_t out[96]);
+int ethrex_mi355_bls_msm_sync(em_bls_msm_plan *plan);
 int ethrex_mi355_bls_msm_last_times(em_bls_msm_plan *plan, double times_ms[5]);
 
 /* n elements uniform in [0, r_bls) (canonical blob field elements) */

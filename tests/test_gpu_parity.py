@@ -234,6 +234,31 @@ def test_gpu_msm_tree_edge_cases(gpu, oracle_mod, monkeypatch):
     assert got_xyzz == want
 
 
+def test_gpu_msm_async_pipeline_matches_sync(gpu, oracle_mod):
+    """The pipelined path (run_async x4 + sync: sort chain of step k+1
+    overlapped with compute chain of step k) returns bit-identical results
+    to the synchronous run, including after a scalar re-upload mid-stream
+    (upload drains the pipeline first)."""
+    n = 1 << 18
+    plan = gpu.MsmPlan(n)
+    plan.gen_points(0)
+    plan.upload_scalars(gpu.gen_fr(42, n))
+    want = plan.run()
+    for _ in range(4):
+        plan.run_async()
+    got = plan.sync()
+    assert got == want
+    # re-upload new scalars while idle-piped, then pipeline again
+    plan.run_async()
+    plan.upload_scalars(gpu.gen_fr(43, n))  # drains, then uploads
+    want2 = plan.run()
+    assert want2 != want
+    plan.run_async()
+    plan.run_async()
+    assert plan.sync() == want2
+    plan.destroy()
+
+
 def test_gpu_plan_combine_matches_standalone(gpu, oracle_mod):
     """plan-attached combine (the per-step N>1 exchange path) matches the
     standalone combine and the oracle."""
