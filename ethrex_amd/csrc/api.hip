@@ -140,12 +140,15 @@ struct msm_plan_t {
         uint8_t *dest;
         int bytes;
         bool valid;
-    } pend[2] = {{nullptr, 0, false}, {nullptr, 0, false}};
+        bool convert;  // staged bytes are Jacobian; convert to affine on host
+    } pend[2] = {{nullptr, 0, false, false}, {nullptr, 0, false, false}};
     int apar = 0;
 };
 
 template <typename C>
 static int msm_sync_t(msm_plan_t<C> *p);
+template <typename C>
+static void host_jac_to_affine(const uint8_t *jac, uint8_t *out);
 
 template <typename C>
 static int msm_destroy_t(msm_plan_t<C> *p) {
@@ -562,11 +565,17 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
                        p->d_partials, p->d_windows);
     HIP_TRY(hipEventRecord(p->ev[3], 0));
     hipLaunchKernelGGL((k_final_combine<C, CFG>), dim3(1), dim3(64), 0, 0,
-                       p->d_windows, p->d_out, out_mode);
+                       p->d_windows, p->d_out, 1);
     HIP_TRY(hipEventRecord(p->ev[4], 0));
-    HIP_TRY(hipMemcpy(out, p->d_out,
-                      out_mode == 0 ? msm_plan_t<C>::AB : msm_plan_t<C>::JB,
-                      hipMemcpyDeviceToHost));
+    if (out_mode == 0) {
+        uint8_t jac[msm_plan_t<C>::JB];
+        HIP_TRY(hipMemcpy(jac, p->d_out, msm_plan_t<C>::JB,
+                          hipMemcpyDeviceToHost));
+        host_jac_to_affine<C>(jac, out);
+    } else {
+        HIP_TRY(hipMemcpy(out, p->d_out, msm_plan_t<C>::JB,
+                          hipMemcpyDeviceToHost));
+    }
     HIP_TRY(hipDeviceSynchronize());
     float ms;
     HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[1]));
@@ -582,12 +591,40 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     return EM_OK;
 }
 
+// Canonical Jacobian bytes -> affine bytes on the HOST (x=X/Z^2, y=Y/Z^3).
+// The GPU k_final_combine's to-affine path runs a 254-bit Fermat inversion
+// on ONE lane (~0.6-1.8 ms); on the host it is ~0.3 ms and — on the
+// pipelined path — overlaps the next step's GPU work entirely.
+template <typename C>
+static void host_jac_to_affine(const uint8_t *jac, uint8_t *out) {
+    using F = typename C::F;
+    constexpr int NB = F::W64 * 8;
+    bool zzero = true;
+    for (int i = 0; i < NB; i++)
+        if (jac[2 * NB + i]) zzero = false;
+    if (zzero) {
+        memset(out, 0, 2 * NB);
+        return;
+    }
+    feL<F::L> X = to_mont9<F>(feT_from_be<F>(jac));
+    feL<F::L> Y = to_mont9<F>(feT_from_be<F>(jac + NB));
+    feL<F::L> Z = to_mont9<F>(feT_from_be<F>(jac + 2 * NB));
+    feL<F::L> zi = mont_inv9<F>(Z);
+    feL<F::L> zi2 = mont_sqr9<F>(zi);
+    feL<F::L> zi3 = mont_mul9<F>(zi2, zi);
+    feT_to_be<F>(out, from_mont9<F>(fe9_csubp<F>(mont_mul9<F>(X, zi2))));
+    feT_to_be<F>(out + NB, from_mont9<F>(fe9_csubp<F>(mont_mul9<F>(Y, zi3))));
+}
+
 // deliver a completed pipelined result to its caller's buffer
 template <typename C>
 static int msm_deliver(msm_plan_t<C> *p, int par) {
     if (!p->pend[par].valid) return EM_OK;
     HIP_TRY(hipEventSynchronize(p->ev_comp_done[par]));
-    memcpy(p->pend[par].dest, p->h_out[par], p->pend[par].bytes);
+    if (p->pend[par].convert)
+        host_jac_to_affine<C>(p->h_out[par], p->pend[par].dest);
+    else
+        memcpy(p->pend[par].dest, p->h_out[par], p->pend[par].bytes);
     p->pend[par].valid = false;
     return EM_OK;
 }
@@ -679,13 +716,15 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
                        p->d_seg_wsum, p->d_partials);
     hipLaunchKernelGGL((k_window_sum<C, CFG>), dim3(CFG::NWIN), dim3(64), 0,
                        sc, p->d_partials, p->d_windows);
+    // always emit the Jacobian form; affine conversion happens on the host
+    // at delivery time (overlapped with the next steps' GPU work)
     hipLaunchKernelGGL((k_final_combine<C, CFG>), dim3(1), dim3(64), 0, sc,
-                       p->d_windows, DOUT, out_mode);
-    int bytes = out_mode == 0 ? msm_plan_t<C>::AB : msm_plan_t<C>::JB;
-    HIP_TRY(hipMemcpyAsync(p->h_out[par], DOUT, bytes,
+                       p->d_windows, DOUT, 1);
+    HIP_TRY(hipMemcpyAsync(p->h_out[par], DOUT, msm_plan_t<C>::JB,
                            hipMemcpyDeviceToHost, sc));
     HIP_TRY(hipEventRecord(p->ev_comp_done[par], sc));
-    p->pend[par] = {out, bytes, true};
+    p->pend[par] = {out, out_mode == 0 ? msm_plan_t<C>::AB : msm_plan_t<C>::JB,
+                    true, out_mode == 0};
     p->apar ^= 1;
     return EM_OK;
 }

@@ -117,7 +117,7 @@ __device__ __host__ __forceinline__ bool fe9_is_zero_raw(const feL<LN> &a) {
 
 // x ≡ 0 mod p for x norm2p (< 2p): x == 0 or x == p
 template <typename T = Fq9T>
-__device__ __forceinline__ bool fe9_is_zero_modp(const feL<T::L> &a) {
+__device__ __host__ __forceinline__ bool fe9_is_zero_modp(const feL<T::L> &a) {
     u32 z = 0, e = 0;
 #pragma unroll
     for (int i = 0; i < T::L; i++) {
@@ -311,7 +311,7 @@ __device__ __host__ __forceinline__ feL<T::L> from_mont9(const feL<T::L> &x) {
 
 // x^e (Montgomery in/out), e canonical u64[T::W64]
 template <typename T = Fq9T>
-__device__ __forceinline__ feL<T::L> mont_pow9(const feL<T::L> &x, const u64 *e) {
+__device__ __host__ __forceinline__ feL<T::L> mont_pow9(const feL<T::L> &x, const u64 *e) {
     feL<T::L> acc = fe9_load<T::L>(T::ONE);
     for (int i = 64 * T::W64 - 1; i >= 0; i--) {
         acc = mont_sqr9<T>(acc);
@@ -322,7 +322,7 @@ __device__ __forceinline__ feL<T::L> mont_pow9(const feL<T::L> &x, const u64 *e)
 
 // 1/x via Fermat (x norm2p, != 0 mod p); exponent p-2 (p odd => no borrow)
 template <typename T = Fq9T>
-__device__ __forceinline__ feL<T::L> mont_inv9(const feL<T::L> &x) {
+__device__ __host__ __forceinline__ feL<T::L> mont_inv9(const feL<T::L> &x) {
     u64 e[T::W64];
 #pragma unroll
     for (int i = 0; i < T::W64; i++) e[i] = T::MOD64[i];

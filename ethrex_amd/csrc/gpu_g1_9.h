@@ -171,7 +171,7 @@ __device__ __forceinline__ g1jT<C> g1_scalar_mul9(const g1aT<C> &p,
 
 // big-endian byte IO of canonical field elements (T::W64 * 8 bytes)
 template <typename T>
-__device__ __forceinline__ void feT_to_be(uint8_t *b, const feL<T::L> &canon) {
+__device__ __host__ __forceinline__ void feT_to_be(uint8_t *b, const feL<T::L> &canon) {
     u64 w[T::W64];
     fe9_to_u64<T>(w, canon);
     u64 *o = (u64 *)b;
@@ -181,7 +181,7 @@ __device__ __forceinline__ void feT_to_be(uint8_t *b, const feL<T::L> &canon) {
 }
 
 template <typename T>
-__device__ __forceinline__ feL<T::L> feT_from_be(const uint8_t *b) {
+__device__ __host__ __forceinline__ feL<T::L> feT_from_be(const uint8_t *b) {
     const u64 *w = (const u64 *)b;
     u64 v[T::W64];
 #pragma unroll
