@@ -121,7 +121,35 @@ k_ntt_row(fe9 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
         smem[j] = row[i];
     }
     __syncthreads();
-    for (int s = 1; s <= logM; s++) {
+    // radix-2^2: two DIT stages per LDS round trip (same mul count as
+    // radix-2 — 1 mul/element per 2 stages — but half the LDS traffic and
+    // half the __syncthreads).  W2b = W2a * w^(M/4).
+    int s = 1;
+    for (; s + 1 <= logM; s += 2) {
+        uint32_t q = 1u << (s - 1);
+        for (uint32_t t = threadIdx.x; t < (M >> 2); t += blockDim.x) {
+            uint32_t j = t & (q - 1);
+            uint32_t idx = ((t >> (s - 1)) << (s + 1)) + j;
+            fe9 w1 = tw_row[j << (logM - s)];
+            fe9 a = smem[idx];
+            fe9 b = mont_mul9<Fr9T>(smem[idx + q], w1);
+            fe9 c = smem[idx + 2 * q];
+            fe9 d = mont_mul9<Fr9T>(smem[idx + 3 * q], w1);
+            fe9 t0 = add9_n<Fr9T>(a, b);
+            fe9 t1 = subm9<Fr9T>(a, b);
+            fe9 t2 = add9_n<Fr9T>(c, d);
+            fe9 t3 = subm9<Fr9T>(c, d);
+            fe9 u2 = mont_mul9<Fr9T>(t2, tw_row[j << (logM - s - 1)]);
+            fe9 u3 = mont_mul9<Fr9T>(
+                t3, tw_row[(j + q) << (logM - s - 1)]);
+            smem[idx] = add9_n<Fr9T>(t0, u2);
+            smem[idx + 2 * q] = subm9<Fr9T>(t0, u2);
+            smem[idx + q] = add9_n<Fr9T>(t1, u3);
+            smem[idx + 3 * q] = subm9<Fr9T>(t1, u3);
+        }
+        __syncthreads();
+    }
+    for (; s <= logM; s++) {  // odd-logM tail: one radix-2 stage
         uint32_t half = 1u << (s - 1);
         for (uint32_t t = threadIdx.x; t < (M >> 1); t += blockDim.x) {
             uint32_t j = t & (half - 1);

@@ -140,7 +140,7 @@ def test_gpu_msm_shard_combine_parity(gpu, oracle_mod):
 
 
 def test_gpu_ntt_parity(gpu, oracle_mod):
-    for n in (1, 2, 256, 1 << 12, 1 << 16):
+    for n in (1, 2, 256, 1 << 12, 1 << 13, 1 << 16):
         elems = gpu.gen_fr(43, n)
         rc, fwd = gpu.fr_ntt(elems, n, False)
         rc2, want = oracle_mod.fr_ntt(elems, n, False)
