@@ -22,6 +22,8 @@ if not os.path.exists(_SO):
 _lib = ctypes.CDLL(_SO)
 
 _lib.oracle_g1_add.restype = ctypes.c_int
+for _f in ("bls_g2_add", "bls_g2_mul", "bls_g2_msm", "bls_g2_gen_points"):
+    getattr(_lib, f"oracle_{_f}").restype = ctypes.c_int
 _lib.oracle_g1_mul.restype = ctypes.c_int
 _lib.oracle_g1_msm.restype = ctypes.c_int
 _lib.oracle_g1_msm_naive.restype = ctypes.c_int
@@ -124,6 +126,33 @@ _lib.oracle_bls_g1_msm_naive.restype = ctypes.c_int
 _lib.oracle_bls_g1_msm_jacobian.restype = ctypes.c_int
 _lib.oracle_bls_g1_combine_jacobian.restype = ctypes.c_int
 _lib.oracle_bls_gen_points.restype = ctypes.c_int
+
+
+def bls_g2_add(p1: bytes, p2: bytes):
+    out = (ctypes.c_uint8 * 192)()
+    rc = _lib.oracle_bls_g2_add(_buf(p1), _buf(p2), out)
+    return rc, bytes(out)
+
+
+def bls_g2_mul(point: bytes, scalar: bytes):
+    out = (ctypes.c_uint8 * 192)()
+    rc = _lib.oracle_bls_g2_mul(_buf(point), _buf(scalar), out)
+    return rc, bytes(out)
+
+
+def bls_g2_msm(points: bytes, scalars: bytes, n: int):
+    out = (ctypes.c_uint8 * 192)()
+    rc = _lib.oracle_bls_g2_msm(_buf(points), _buf(scalars),
+                                ctypes.c_size_t(n), out)
+    return rc, bytes(out)
+
+
+def bls_g2_gen_points(start: int, n: int) -> bytes:
+    out = (ctypes.c_uint8 * (192 * n))()
+    rc = _lib.oracle_bls_g2_gen_points(ctypes.c_uint64(start),
+                                       ctypes.c_size_t(n), out)
+    assert rc == 0
+    return bytes(out)
 
 
 def bls_g1_add(p1: bytes, p2: bytes):

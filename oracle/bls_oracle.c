@@ -608,3 +608,352 @@ void oracle_bls_gen_fr(uint64_t seed, size_t n, uint8_t *out) {
                 out[32 * i + (3 - w) * 8 + j] = (uint8_t)(v[w] >> (56 - 8 * j));
     }
 }
+
+/* ================= BLS12-381 G2: Fp2 = Fp[u]/(u^2+1) =================
+ * EIP-2537 / blst semantics (crates/common/crypto/bls_blst.rs:226-255,
+ * 303-321, 338-345, 395-441): 192-byte points x.c0||x.c1||y.c0||y.c1 with
+ * canonical 48-B BE coords, (0,0,0,0) identity, on-curve for add,
+ * on-curve + r-subgroup for MSM inputs; curve y^2 = x^3 + 4(1+u). */
+
+typedef struct { fb c0, c1; } f2;
+typedef struct { f2 x, y; } bg2a;
+typedef struct { f2 x, y, z; } bg2j;
+
+static void f2_add(f2 *o, const f2 *a, const f2 *b) {
+    fb_mod_add(&o->c0, &a->c0, &b->c0);
+    fb_mod_add(&o->c1, &a->c1, &b->c1);
+}
+static void f2_sub(f2 *o, const f2 *a, const f2 *b) {
+    fb_mod_sub(&o->c0, &a->c0, &b->c0);
+    fb_mod_sub(&o->c1, &a->c1, &b->c1);
+}
+static void f2_mul(f2 *o, const f2 *a, const f2 *b) {
+    fb m0, m1, t0, t1, r0, r1;
+    fb_mont_mul(&m0, &a->c0, &b->c0);
+    fb_mont_mul(&m1, &a->c1, &b->c1);
+    fb_mont_mul(&t0, &a->c0, &b->c1);
+    fb_mont_mul(&t1, &a->c1, &b->c0);
+    fb_mod_sub(&r0, &m0, &m1);
+    fb_mod_add(&r1, &t0, &t1);
+    o->c0 = r0;
+    o->c1 = r1;
+}
+static void f2_sqr(f2 *o, const f2 *a) { f2_mul(o, a, a); }
+static int f2_is_zero(const f2 *a) {
+    return fb_is_zero(&a->c0) && fb_is_zero(&a->c1);
+}
+static int f2_eq(const f2 *a, const f2 *b) {
+    return fb_cmp(&a->c0, &b->c0) == 0 && fb_cmp(&a->c1, &b->c1) == 0;
+}
+static void f2_inv(f2 *o, const f2 *a) {
+    fb n0, n1, t, z;
+    fb_sqr(&n0, &a->c0);
+    fb_sqr(&n1, &a->c1);
+    fb_mod_add(&n0, &n0, &n1);
+    fb_inv(&t, &n0);
+    memset(&z, 0, sizeof(fb));
+    fb_mont_mul(&o->c0, &a->c0, &t);
+    fb_mont_mul(&n1, &a->c1, &t);
+    fb_mod_sub(&o->c1, &z, &n1);
+}
+
+static void bg2_set_inf(bg2j *p) {
+    fb_from_limbs(&p->x.c0, BLSP_R);
+    memset(&p->x.c1, 0, sizeof(fb));
+    p->y = p->x;
+    memset(&p->z, 0, sizeof(f2));
+}
+static int bg2_is_inf(const bg2j *p) { return f2_is_zero(&p->z); }
+
+static void bg2_dbl(bg2j *o, const bg2j *p) {
+    if (bg2_is_inf(p)) { *o = *p; return; }
+    f2 A, B, C, D, E, F, t, t2, y3, z3;
+    f2_sqr(&A, &p->x);
+    f2_sqr(&B, &p->y);
+    f2_sqr(&C, &B);
+    f2_add(&t, &p->x, &B);
+    f2_sqr(&t, &t);
+    f2_sub(&t, &t, &A);
+    f2_sub(&t, &t, &C);
+    f2_add(&D, &t, &t);
+    f2_add(&E, &A, &A);
+    f2_add(&E, &E, &A);
+    f2_sqr(&F, &E);
+    f2_sub(&t, &F, &D);
+    f2_sub(&o->x, &t, &D);
+    f2_sub(&t, &D, &o->x);
+    f2_mul(&t, &E, &t);
+    f2_add(&t2, &C, &C);
+    f2_add(&t2, &t2, &t2);
+    f2_add(&t2, &t2, &t2);
+    f2_sub(&y3, &t, &t2);
+    f2_mul(&z3, &p->y, &p->z);
+    f2_add(&z3, &z3, &z3);
+    o->y = y3;
+    o->z = z3;
+}
+
+static void bg2_add_affine(bg2j *o, const bg2j *p, const bg2a *q) {
+    if (bg2_is_inf(p)) {
+        o->x = q->x;
+        o->y = q->y;
+        fb_from_limbs(&o->z.c0, BLSP_R);
+        memset(&o->z.c1, 0, sizeof(fb));
+        return;
+    }
+    f2 z1z1, u2, s2, h, r, t;
+    f2_sqr(&z1z1, &p->z);
+    f2_mul(&u2, &q->x, &z1z1);
+    f2_mul(&t, &p->z, &z1z1);
+    f2_mul(&s2, &q->y, &t);
+    f2_sub(&h, &u2, &p->x);
+    f2_sub(&r, &s2, &p->y);
+    if (f2_is_zero(&h)) {
+        if (f2_is_zero(&r)) { bg2_dbl(o, p); return; }
+        bg2_set_inf(o);
+        return;
+    }
+    f2 hh, hhh, v, x3, y3, z3;
+    f2_sqr(&hh, &h);
+    f2_mul(&hhh, &h, &hh);
+    f2_mul(&v, &p->x, &hh);
+    f2_sqr(&x3, &r);
+    f2_sub(&x3, &x3, &hhh);
+    f2_sub(&x3, &x3, &v);
+    f2_sub(&x3, &x3, &v);
+    f2_sub(&t, &v, &x3);
+    f2_mul(&y3, &r, &t);
+    f2_mul(&t, &p->y, &hhh);
+    f2_sub(&y3, &y3, &t);
+    f2_mul(&z3, &p->z, &h);
+    o->x = x3;
+    o->y = y3;
+    o->z = z3;
+}
+
+static int bg2a_on_curve(const bg2a *p) {
+    f2 l, r, b2;
+    f2_sqr(&l, &p->y);
+    f2_sqr(&r, &p->x);
+    f2_mul(&r, &r, &p->x);
+    fb_from_limbs(&b2.c0, BLS_B4_MONT);
+    fb_from_limbs(&b2.c1, BLS_B4_MONT);
+    f2_add(&r, &r, &b2);
+    return f2_eq(&l, &r);
+}
+
+static void bg2_to_affine_be(uint8_t out[192], const bg2j *p) {
+    if (bg2_is_inf(p)) { memset(out, 0, 192); return; }
+    f2 zi, zi2, zi3, xa, ya;
+    fb c;
+    f2_inv(&zi, &p->z);
+    f2_sqr(&zi2, &zi);
+    f2_mul(&zi3, &zi2, &zi);
+    f2_mul(&xa, &p->x, &zi2);
+    f2_mul(&ya, &p->y, &zi3);
+    fb_from_mont(&c, &xa.c0); fb_to_be(out, &c);
+    fb_from_mont(&c, &xa.c1); fb_to_be(out + 48, &c);
+    fb_from_mont(&c, &ya.c0); fb_to_be(out + 96, &c);
+    fb_from_mont(&c, &ya.c1); fb_to_be(out + 144, &c);
+}
+
+static int bg2_parse_be(bg2a *o, int *is_inf, const uint8_t in[192]) {
+    fb v[4], m;
+    fb_from_limbs(&m, BLSP_MOD);
+    for (int k = 0; k < 4; k++) {
+        fb_from_be(&v[k], in + 48 * k);
+        if (fb_cmp(&v[k], &m) >= 0) return BORC_ERR_INPUT;
+    }
+    if (fb_is_zero(&v[0]) && fb_is_zero(&v[1]) && fb_is_zero(&v[2]) &&
+        fb_is_zero(&v[3])) {
+        *is_inf = 1;
+        return BORC_OK;
+    }
+    fb_to_mont(&o->x.c0, &v[0]);
+    fb_to_mont(&o->x.c1, &v[1]);
+    fb_to_mont(&o->y.c0, &v[2]);
+    fb_to_mont(&o->y.c1, &v[3]);
+    *is_inf = 0;
+    if (!bg2a_on_curve(o)) return BORC_ERR_POINT;
+    return BORC_OK;
+}
+
+static void bg2_scalar_mul_be(bg2j *o, const bg2a *p, const uint8_t k[32]) {
+    bg2j acc;
+    bg2_set_inf(&acc);
+    for (int i = 0; i < 256; i++) {
+        bg2_dbl(&acc, &acc);
+        if ((k[i / 8] >> (7 - (i % 8))) & 1) bg2_add_affine(&acc, &acc, p);
+    }
+    *o = acc;
+}
+
+static void bg2_full_add(bg2j *o, const bg2j *p, const bg2j *q) {
+    if (bg2_is_inf(p)) { *o = *q; return; }
+    if (bg2_is_inf(q)) { *o = *p; return; }
+    f2 z1z1, z2z2, u1, u2, s1, s2, h, r, t;
+    f2_sqr(&z1z1, &p->z);
+    f2_sqr(&z2z2, &q->z);
+    f2_mul(&u1, &p->x, &z2z2);
+    f2_mul(&u2, &q->x, &z1z1);
+    f2_mul(&t, &q->z, &z2z2);
+    f2_mul(&s1, &p->y, &t);
+    f2_mul(&t, &p->z, &z1z1);
+    f2_mul(&s2, &q->y, &t);
+    f2_sub(&h, &u2, &u1);
+    f2_sub(&r, &s2, &s1);
+    if (f2_is_zero(&h)) {
+        if (f2_is_zero(&r)) { bg2_dbl(o, p); return; }
+        bg2_set_inf(o);
+        return;
+    }
+    f2 hh, hhh, v, x3, y3, z3;
+    f2_sqr(&hh, &h);
+    f2_mul(&hhh, &h, &hh);
+    f2_mul(&v, &u1, &hh);
+    f2_sqr(&x3, &r);
+    f2_sub(&x3, &x3, &hhh);
+    f2_sub(&x3, &x3, &v);
+    f2_sub(&x3, &x3, &v);
+    f2_sub(&t, &v, &x3);
+    f2_mul(&y3, &r, &t);
+    f2_mul(&t, &s1, &hhh);
+    f2_sub(&y3, &y3, &t);
+    f2_mul(&z3, &p->z, &q->z);
+    f2_mul(&z3, &z3, &h);
+    o->x = x3;
+    o->y = y3;
+    o->z = z3;
+}
+
+static int bg2_in_subgroup(const bg2a *p) {
+    uint8_t rbe[32];
+    for (int i = 0; i < 4; i++)
+        for (int j = 0; j < 8; j++)
+            rbe[(3 - i) * 8 + j] = (uint8_t)(BLSR_MOD[i] >> (56 - 8 * j));
+    bg2j t;
+    bg2_scalar_mul_be(&t, p, rbe);
+    return bg2_is_inf(&t);
+}
+
+int oracle_bls_g2_add(const uint8_t p1[192], const uint8_t p2[192],
+                      uint8_t out[192]) {
+    bg2a a, b;
+    int ia, ib, rc;
+    if ((rc = bg2_parse_be(&a, &ia, p1))) return rc;
+    if ((rc = bg2_parse_be(&b, &ib, p2))) return rc;
+    bg2j acc;
+    bg2_set_inf(&acc);
+    if (!ia) bg2_add_affine(&acc, &acc, &a);
+    if (!ib) bg2_add_affine(&acc, &acc, &b);
+    bg2_to_affine_be(out, &acc);
+    return BORC_OK;
+}
+
+int oracle_bls_g2_mul(const uint8_t point[192], const uint8_t scalar[32],
+                      uint8_t out[192]) {
+    bg2a p;
+    int inf, rc;
+    if ((rc = bg2_parse_be(&p, &inf, point))) return rc;
+    if (inf) { memset(out, 0, 192); return BORC_OK; }
+    bg2j r;
+    bg2_scalar_mul_be(&r, &p, scalar);
+    bg2_to_affine_be(out, &r);
+    return BORC_OK;
+}
+
+/* Pippenger c=16, raw 256-bit scalars, per-point subgroup check
+ * (bls_blst.rs g2_msm -> read_g2_subgroup) */
+int oracle_bls_g2_msm(const uint8_t *points, const uint8_t *scalars, size_t n,
+                      uint8_t out[192]) {
+    bg2a *pts = malloc(n * sizeof(bg2a));
+    uint8_t *inf = malloc(n);
+    int err = 0;
+#ifdef _OPENMP
+#pragma omp parallel for schedule(dynamic, 16)
+#endif
+    for (size_t i = 0; i < n; i++) {
+        int ii, rc;
+        if ((rc = bg2_parse_be(&pts[i], &ii, points + 192 * i))) {
+#ifdef _OPENMP
+#pragma omp atomic write
+#endif
+            err = rc;
+            continue;
+        }
+        inf[i] = (uint8_t)ii;
+        if (!ii && !bg2_in_subgroup(&pts[i])) {
+#ifdef _OPENMP
+#pragma omp atomic write
+#endif
+            err = BORC_ERR_SUBGROUP;
+        }
+    }
+    if (err) { free(pts); free(inf); return err; }
+    bg2j wsum[BNWIN];
+#ifdef _OPENMP
+#pragma omp parallel
+#endif
+    {
+        bg2j *buckets = malloc(BNBUCKET * sizeof(bg2j));
+#ifdef _OPENMP
+#pragma omp for schedule(dynamic)
+#endif
+        for (int w = 0; w < BNWIN; w++) {
+            for (uint32_t b = 0; b < BNBUCKET; b++) bg2_set_inf(&buckets[b]);
+            for (size_t i = 0; i < n; i++) {
+                if (inf[i]) continue;
+                const uint8_t *k = scalars + 32 * i;
+                int bit = BC * w;
+                uint32_t d = 0;
+                for (int t = 0; t < BC; t++) {
+                    int bb = bit + t;
+                    d |= (uint32_t)((k[31 - bb / 8] >> (bb % 8)) & 1) << t;
+                }
+                if (d) bg2_add_affine(&buckets[d - 1], &buckets[d - 1], &pts[i]);
+            }
+            bg2j runj, sumj;
+            bg2_set_inf(&runj);
+            bg2_set_inf(&sumj);
+            for (uint32_t b = BNBUCKET; b-- > 0;) {
+                /* runj += buckets[b]; sumj += runj (full Jacobian add) */
+                bg2j tmp;
+                bg2_full_add(&tmp, &runj, &buckets[b]);
+                runj = tmp;
+                bg2_full_add(&tmp, &sumj, &runj);
+                sumj = tmp;
+            }
+            wsum[w] = sumj;
+        }
+        free(buckets);
+    }
+    free(pts);
+    free(inf);
+    bg2j acc = wsum[BNWIN - 1];
+    for (int w = BNWIN - 2; w >= 0; w--) {
+        for (int t = 0; t < BC; t++) bg2_dbl(&acc, &acc);
+        bg2j tmp;
+        bg2_full_add(&tmp, &acc, &wsum[w]);
+        acc = tmp;
+    }
+    bg2_to_affine_be(out, &acc);
+    return BORC_OK;
+}
+
+int oracle_bls_g2_gen_points(uint64_t start, size_t n, uint8_t *out) {
+    if (n == 0) return BORC_OK;
+    bg2a gen;
+    fb_from_limbs(&gen.x.c0, BLS_G2X0_MONT);
+    fb_from_limbs(&gen.x.c1, BLS_G2X1_MONT);
+    fb_from_limbs(&gen.y.c0, BLS_G2Y0_MONT);
+    fb_from_limbs(&gen.y.c1, BLS_G2Y1_MONT);
+    bg2j *acc = malloc(n * sizeof(bg2j));
+    uint8_t k[32] = {0};
+    uint64_t k0 = start + 1;
+    for (int j = 0; j < 8; j++) k[31 - j] = (uint8_t)(k0 >> (8 * j));
+    bg2_scalar_mul_be(&acc[0], &gen, k);
+    for (size_t i = 1; i < n; i++) bg2_add_affine(&acc[i], &acc[i - 1], &gen);
+    for (size_t i = 0; i < n; i++) bg2_to_affine_be(out + 192 * i, &acc[i]);
+    free(acc);
+    return BORC_OK;
+}
