@@ -83,10 +83,10 @@ static inline uint32_t blocks_for(size_t n, int bs) {
 template <typename C>
 struct msm_plan_t {
     using F = typename C::F;
-    static constexpr int PB = 2 * F::W64 * 8;  // affine point bytes
-    static constexpr int JB = 3 * F::W64 * 8;  // Jacobian partial bytes
-    static constexpr int AB = PB;              // affine out bytes
-    static constexpr int SB = std::is_same_v<C, BlsG1> ? 256 : 254;
+    static constexpr int PB = pt_bytes<C>::AFF;  // affine point bytes
+    static constexpr int JB = pt_bytes<C>::JAC;  // Jacobian partial bytes
+    static constexpr int AB = PB;                // affine out bytes
+    static constexpr int SB = std::is_same_v<C, Bn254G1> ? 254 : 256;
     size_t n;
     int cbits;                        // window config: 8 (small) or 16
     g1aT<C> *d_pts = nullptr;
@@ -347,7 +347,11 @@ static int msm_upload_points_t(msm_plan_t<C> *p, const uint8_t *points) {
     constexpr int PB = msm_plan_t<C>::PB;
     HIP_TRY(hipMemset(p->d_err, 0, 4));
     HIP_TRY(hipMemcpy(p->d_scratch, points, p->n * PB, hipMemcpyHostToDevice));
-    if constexpr (std::is_same_v<C, BlsG1>) {
+    if constexpr (std::is_same_v<C, BlsG2>) {
+        hipLaunchKernelGGL(k_bls_g2_parse_points, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_scratch, p->d_pts, p->d_inf,
+                           p->n, p->d_err);
+    } else if constexpr (std::is_same_v<C, BlsG1>) {
         hipLaunchKernelGGL(k_bls_parse_points, dim3(blocks_for(p->n, 256)),
                            dim3(256), 0, 0, p->d_scratch, p->d_pts, p->d_inf,
                            p->n, p->d_err);
@@ -375,7 +379,10 @@ static int msm_upload_points_t(msm_plan_t<C> *p, const uint8_t *points) {
 template <typename C>
 static int msm_gen_points_t(msm_plan_t<C> *p, uint64_t start) {
     if (!p) return EM_ERR_INPUT;
-    if constexpr (std::is_same_v<C, BlsG1>) {
+    if constexpr (std::is_same_v<C, BlsG2>) {
+        hipLaunchKernelGGL(k_bls_g2_gen_points, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_pts, p->d_inf, p->n, start);
+    } else if constexpr (std::is_same_v<C, BlsG1>) {
         hipLaunchKernelGGL(k_bls_gen_points, dim3(blocks_for(p->n, 256)),
                            dim3(256), 0, 0, p->d_pts, p->d_inf, p->n, start);
     } else {
@@ -392,7 +399,11 @@ template <typename C>
 static int msm_download_points_t(msm_plan_t<C> *p, uint8_t *out) {
     if (!p || !out || !p->have_points) return EM_ERR_INPUT;
     constexpr int PB = msm_plan_t<C>::PB;
-    if constexpr (std::is_same_v<C, BlsG1>) {
+    if constexpr (std::is_same_v<C, BlsG2>) {
+        hipLaunchKernelGGL(k_bls_g2_points_to_be, dim3(blocks_for(p->n, 256)),
+                           dim3(256), 0, 0, p->d_pts, p->d_inf, p->d_scratch,
+                           p->n);
+    } else if constexpr (std::is_same_v<C, BlsG1>) {
         hipLaunchKernelGGL(k_bls_points_to_be, dim3(blocks_for(p->n, 256)),
                            dim3(256), 0, 0, p->d_pts, p->d_inf, p->d_scratch,
                            p->n);
@@ -434,7 +445,7 @@ static int msm_upload_scalars_t(msm_plan_t<C> *p, const uint8_t *scalars32) {
     int rc0 = msm_sync_t(p);
     if (rc0) return rc0;
     HIP_TRY(hipMemcpy(p->d_scratch, scalars32, p->n * 32, hipMemcpyHostToDevice));
-    if constexpr (std::is_same_v<C, BlsG1>) {
+    if constexpr (!std::is_same_v<C, Bn254G1>) {
         // raw 256-bit scalars, no reduction (blst SCALAR_BITS = 256)
         hipLaunchKernelGGL(k_bls_parse_scalars, dim3(blocks_for(p->n, 256)),
                            dim3(256), 0, 0, p->d_scratch, p->d_scalars, p->n);
@@ -616,6 +627,25 @@ static void host_jac_to_affine(const uint8_t *jac, uint8_t *out) {
     feT_to_be<F>(out + NB, from_mont9<F>(fe9_csubp<F>(mont_mul9<F>(Y, zi3))));
 }
 
+template <>
+void host_jac_to_affine<BlsG2>(const uint8_t *jac, uint8_t *out) {
+    bool zzero = true;
+    for (int i = 0; i < 96; i++)
+        if (jac[192 + i]) zzero = false;
+    if (zzero) {
+        memset(out, 0, 192);
+        return;
+    }
+    fp2 X = fp2_from_be_mont(jac);
+    fp2 Y = fp2_from_be_mont(jac + 96);
+    fp2 Z = fp2_from_be_mont(jac + 192);
+    fp2 zi = fp2_inv(Z);
+    fp2 zi2 = fp2_sqr(zi);
+    fp2 zi3 = fp2_mul(zi2, zi);
+    fp2_to_be(out, fp2_mul(X, zi2));
+    fp2_to_be(out + 96, fp2_mul(Y, zi3));
+}
+
 // deliver a completed pipelined result to its caller's buffer
 template <typename C>
 static int msm_deliver(msm_plan_t<C> *p, int par) {
@@ -760,6 +790,7 @@ static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
 // opaque ABI types
 struct em_msm_plan : msm_plan_t<Bn254G1> {};
 struct em_bls_msm_plan : msm_plan_t<BlsG1> {};
+struct em_bls_g2_msm_plan : msm_plan_t<BlsG2> {};
 
 extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
     return msm_create_t(n, (msm_plan_t<Bn254G1> **)plan);
@@ -890,6 +921,65 @@ extern "C" int ethrex_mi355_bls12381_g1_msm(const uint8_t *points96,
     return rc;
 }
 
+// ---- BLS12-381 G2 plan + one-shot ABI (EIP-2537 192-byte points) ----
+extern "C" int ethrex_mi355_bls_g2_msm_plan_create(size_t n,
+                                                   em_bls_g2_msm_plan **plan) {
+    return msm_create_t(n, (msm_plan_t<BlsG2> **)plan);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_plan_destroy(em_bls_g2_msm_plan *p) {
+    return msm_destroy_t((msm_plan_t<BlsG2> *)p);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_upload_points(em_bls_g2_msm_plan *p,
+                                                     const uint8_t *pts192) {
+    return msm_upload_points_t((msm_plan_t<BlsG2> *)p, pts192);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_gen_points(em_bls_g2_msm_plan *p,
+                                                  uint64_t start) {
+    return msm_gen_points_t((msm_plan_t<BlsG2> *)p, start);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_download_points(em_bls_g2_msm_plan *p,
+                                                       uint8_t *out192) {
+    return msm_download_points_t((msm_plan_t<BlsG2> *)p, out192);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_upload_scalars(em_bls_g2_msm_plan *p,
+                                                      const uint8_t *s32) {
+    return msm_upload_scalars_t((msm_plan_t<BlsG2> *)p, s32);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_run(em_bls_g2_msm_plan *p,
+                                           uint8_t out[192]) {
+    return msm_run_inner_t((msm_plan_t<BlsG2> *)p, out, 0);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_run_async(em_bls_g2_msm_plan *p,
+                                                 uint8_t out[192]) {
+    return msm_run_async_t((msm_plan_t<BlsG2> *)p, out);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_sync(em_bls_g2_msm_plan *p) {
+    return msm_sync_t((msm_plan_t<BlsG2> *)p);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_run_partial(em_bls_g2_msm_plan *p,
+                                                   uint8_t out[288]) {
+    return msm_run_inner_t((msm_plan_t<BlsG2> *)p, out, 1);
+}
+extern "C" int ethrex_mi355_bls_g2_msm_last_times(em_bls_g2_msm_plan *p,
+                                                  double times_ms[5]) {
+    if (!p) return EM_ERR_INPUT;
+    for (int i = 0; i < 5; i++) times_ms[i] = ((msm_plan_t<BlsG2> *)p)->last_ms[i];
+    return EM_OK;
+}
+extern "C" int ethrex_mi355_bls12381_g2_msm(const uint8_t *points192,
+                                            const uint8_t *scalars32, size_t n,
+                                            uint8_t out[192]) {
+    if (!points192 || !scalars32 || !out || n == 0) return EM_ERR_INPUT;
+    em_bls_g2_msm_plan *p = nullptr;
+    int rc = ethrex_mi355_bls_g2_msm_plan_create(n, &p);
+    if (rc) return rc;
+    rc = ethrex_mi355_bls_g2_msm_upload_points(p, points192);
+    if (!rc) rc = ethrex_mi355_bls_g2_msm_upload_scalars(p, scalars32);
+    if (!rc) rc = ethrex_mi355_bls_g2_msm_run(p, out);
+    ethrex_mi355_bls_g2_msm_plan_destroy(p);
+    return rc;
+}
+
 // ============================ single ops ============================
 
 // generic 1-thread op runner: in_bytes staged, out_bytes copied back
@@ -942,6 +1032,18 @@ extern "C" int ethrex_mi355_bls12381_g1_mul(const uint8_t point[96],
                                             uint8_t out[96]) {
     if (!point || !scalar || !out) return EM_ERR_INPUT;
     return run_single(k_bls_g1_mul_single, point, 96, scalar, 32, out, 96);
+}
+
+extern "C" int ethrex_mi355_bls12381_g2_add(const uint8_t p1[192],
+                                            const uint8_t p2[192],
+                                            uint8_t out[192]) {
+    return run_single(k_bls_g2_add_single, p1, 192, p2, 192, out, 192);
+}
+
+extern "C" int ethrex_mi355_bls12381_g2_mul(const uint8_t point[192],
+                                            const uint8_t scalar[32],
+                                            uint8_t out[192]) {
+    return run_single(k_bls_g2_mul_single, point, 192, scalar, 32, out, 192);
 }
 
 extern "C" int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96,

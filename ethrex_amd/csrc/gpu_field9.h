@@ -214,7 +214,7 @@ __device__ __host__ __forceinline__ feL<T::L> subn9(const feL<T::L> &a,
 
 // -y mod p for y norm2p: 4p - y -> norm2p
 template <typename T = Fq9T>
-__device__ __forceinline__ feL<T::L> neg9(const feL<T::L> &y) {
+__device__ __host__ __forceinline__ feL<T::L> neg9(const feL<T::L> &y) {
     feL<T::L> t;
 #pragma unroll
     for (int i = 0; i < T::L; i++) t.v[i] = T::C4P[i] - y.v[i];

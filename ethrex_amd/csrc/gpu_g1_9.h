@@ -252,4 +252,47 @@ __device__ __forceinline__ g1jT<C> g1_jacobian_to_xyzz9(const feL<C::F::L> &X,
     return p;
 }
 
+// wire sizes per curve (bytes); G2 specializes (4 affine / 6 jac coords)
+template <typename C>
+struct pt_bytes {
+    static constexpr int NB = C::F::W64 * 8;  // one base-field coordinate
+    static constexpr int AFF = 2 * NB;        // affine point wire
+    static constexpr int JAC = 3 * NB;        // Jacobian exchange wire
+};
+
+// point negation in place (y -> -y)
+template <typename C>
+__device__ __forceinline__ void g1_neg_y9(g1jT<C> &p) {
+    p.y = neg9<typename C::F>(p.y);
+}
+
+// ---- Jacobian wire IO (the 3-coordinate exchange payload; Z=0 = inf) ----
+template <typename C>
+__device__ __forceinline__ void g1_jac_be9(uint8_t *out, const g1jT<C> &p) {
+    using F = typename C::F;
+    constexpr int NB = F::W64 * 8;
+    if (g1_is_inf9(p)) {
+        for (int j = 0; j < 3 * F::W64; j++) ((u64 *)out)[j] = 0;
+        return;
+    }
+    feL<F::L> X, Y, Z;
+    g1_xyzz_to_jacobian9(X, Y, Z, p);
+    feT_to_be<F>(out, from_mont9<F>(X));
+    feT_to_be<F>(out + NB, from_mont9<F>(Y));
+    feT_to_be<F>(out + 2 * NB, from_mont9<F>(Z));
+}
+
+// parse one Jacobian wire payload; returns false for infinity
+template <typename C>
+__device__ __forceinline__ bool g1_jac_from_be9(g1jT<C> &o, const uint8_t *in) {
+    using F = typename C::F;
+    constexpr int NB = F::W64 * 8;
+    feL<F::L> X = to_mont9<F>(feT_from_be<F>(in));
+    feL<F::L> Y = to_mont9<F>(feT_from_be<F>(in + NB));
+    feL<F::L> Z = to_mont9<F>(feT_from_be<F>(in + 2 * NB));
+    if (fe9_is_zero_modp<F>(Z)) return false;
+    o = g1_jacobian_to_xyzz9<C>(X, Y, Z);
+    return true;
+}
+
 }  // namespace em

@@ -23,6 +23,7 @@
 #include <hip/hip_runtime.h>
 #include "gpu_field.h"   // fe4: Fr scalar handling
 #include "gpu_g1_9.h"    // fe9: Fq / G1 compute core
+#include "gpu_g2.h"      // fp2: BLS12-381 G2 specializations
 
 namespace em {
 
@@ -455,7 +456,7 @@ k_weighted_reduce(const g1jT<C> *__restrict__ seg_sum,
         g1jT<C> ss = seg_sum[t];
         if (j == 0) {
             // weight -1: subtract sum_0
-            if (!g1_is_inf9(ss)) ss.y = neg9<typename C::F>(ss.y);
+            if (!g1_is_inf9(ss)) g1_neg_y9<C>(ss);
             val = g1_add9(ws, ss);
         } else {
             uint32_t weight = j * CFG::SEG - 1;  // < 2^C
@@ -546,15 +547,7 @@ k_final_combine(const g1jT<C> *__restrict__ windows,
         g1_to_affine_be9(out, acc);
     } else {
         // Jacobian exchange payload (ABI): convert XYZZ -> Jacobian
-        if (g1_is_inf9(acc)) {
-            for (int j = 0; j < 3 * F::W64; j++) ((u64 *)out)[j] = 0;
-        } else {
-            feL<F::L> X, Y, Z;
-            g1_xyzz_to_jacobian9(X, Y, Z, acc);
-            feT_to_be<F>(out, from_mont9<F>(X));
-            feT_to_be<F>(out + NB, from_mont9<F>(Y));
-            feT_to_be<F>(out + 2 * NB, from_mont9<F>(Z));
-        }
+        g1_jac_be9(out, acc);
     }
 }
 
@@ -610,11 +603,10 @@ __global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     g1jT<C> acc = g1_inf9<C>();
     for (size_t i = 0; i < count; i++) {
-        feL<F::L> X = to_mont9<F>(feT_from_be<F>(in + 3 * NB * i));
-        feL<F::L> Y = to_mont9<F>(feT_from_be<F>(in + 3 * NB * i + NB));
-        feL<F::L> Z = to_mont9<F>(feT_from_be<F>(in + 3 * NB * i + 2 * NB));
-        if (fe9_is_zero_modp<F>(Z)) continue;  // Z=0 encodes infinity
-        acc = g1_add9(acc, g1_jacobian_to_xyzz9<C>(X, Y, Z));
+        g1jT<C> t;
+        if (!g1_jac_from_be9<C>(t, in + (size_t)pt_bytes<C>::JAC * i))
+            continue;  // infinity
+        acc = g1_add9(acc, t);
     }
     g1_to_affine_be9(out, acc);
 }
@@ -797,6 +789,142 @@ __global__ void k_bls_g1_mul_single(const uint8_t *in /* 96+32 B */,
     u64 k[4] = {__builtin_bswap64(w[3]), __builtin_bswap64(w[2]),
                 __builtin_bswap64(w[1]), __builtin_bswap64(w[0])};
     g1_to_affine_be9(out, g1_scalar_mul9(p, k, 4));
+}
+
+
+// ---- BLS12-381 G2 input kernels (EIP-2537 192-byte points) ----
+using g1aG2 = g1aT<BlsG2>;
+using g1jG2 = g1jT<BlsG2>;
+
+// parse + validate: canonical coords, (0,0,0,0) identity, on-curve, and
+// r-subgroup (bls_blst.rs read_g2_subgroup) when check_subgroup is set
+__global__ void k_bls_g2_parse_points(const uint8_t *__restrict__ in,
+                                      g1aG2 *__restrict__ pts,
+                                      uint8_t *__restrict__ inf, size_t n,
+                                      uint32_t *__restrict__ err) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    const uint8_t *b = in + 192 * i;
+    fe14 c[4];
+    bool allz = true;
+#pragma unroll
+    for (int k = 0; k < 4; k++) {
+        c[k] = feT_from_be<FpB14T>(b + 48 * k);
+        if (fe9_geq_raw<14>(c[k], bn254::FPB_P)) {
+            atomicOr(err, 2u);
+            return;
+        }
+        if (!fe9_is_zero_raw<14>(c[k])) allz = false;
+    }
+    if (allz) {
+        inf[i] = 1;
+        pts[i].x = fp2_zero();
+        pts[i].y = fp2_zero();
+        return;
+    }
+    g1aG2 p;
+    p.x = {to_mont9<FpB14T>(c[0]), to_mont9<FpB14T>(c[1])};
+    p.y = {to_mont9<FpB14T>(c[2]), to_mont9<FpB14T>(c[3])};
+    inf[i] = 0;
+    if (!g1a9_on_curve<BlsG2>(p)) {
+        atomicOr(err, 1u);
+        return;
+    }
+    g1jG2 t = g1_scalar_mul9(p, bn254::FRB_ORDER, 4);
+    if (!g1_is_inf9<BlsG2>(t)) atomicOr(err, 4u);
+    pts[i] = p;
+}
+
+// P_i = (start+i+1) * G2gen
+__global__ void k_bls_g2_gen_points(g1aG2 *__restrict__ pts,
+                                    uint8_t *__restrict__ inf, size_t n,
+                                    uint64_t start) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    u64 k[4] = {start + i + 1, 0, 0, 0};
+    g1aG2 g = g1_generator9<BlsG2>();
+    g1jG2 acc = g1_inf9<BlsG2>();
+    for (int b = 63; b >= 0; b--) {
+        acc = g1_dbl9<BlsG2>(acc);
+        if ((k[0] >> b) & 1) acc = g1_add_affine9<BlsG2>(acc, g);
+    }
+    pts[i] = g1_to_affine9<BlsG2>(acc);
+    inf[i] = 0;
+}
+
+__global__ void k_bls_g2_points_to_be(const g1aG2 *__restrict__ pts,
+                                      const uint8_t *__restrict__ inf,
+                                      uint8_t *__restrict__ out, size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    if (inf[i]) {
+        for (int j = 0; j < 24; j++) ((u64 *)(out + 192 * i))[j] = 0;
+        return;
+    }
+    fp2_to_be(out + 192 * i, pts[i].x);
+    fp2_to_be(out + 192 * i + 96, pts[i].y);
+}
+
+// single ops (bls_blst.rs g2_add / p2_mult semantics; add: no subgroup check)
+__device__ __forceinline__ int bls_g2_parse_one(const uint8_t *b, g1aG2 &p,
+                                                bool &is_inf) {
+    fe14 c[4];
+    bool allz = true;
+#pragma unroll
+    for (int k = 0; k < 4; k++) {
+        c[k] = feT_from_be<FpB14T>(b + 48 * k);
+        if (fe9_geq_raw<14>(c[k], bn254::FPB_P)) return 2;
+        if (!fe9_is_zero_raw<14>(c[k])) allz = false;
+    }
+    if (allz) {
+        is_inf = true;
+        return 0;
+    }
+    p.x = {to_mont9<FpB14T>(c[0]), to_mont9<FpB14T>(c[1])};
+    p.y = {to_mont9<FpB14T>(c[2]), to_mont9<FpB14T>(c[3])};
+    is_inf = false;
+    if (!g1a9_on_curve<BlsG2>(p)) return 1;
+    return 0;
+}
+
+__global__ void k_bls_g2_add_single(const uint8_t *in /* 384 B */,
+                                    uint8_t *out, uint32_t *err) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    g1jG2 acc = g1_inf9<BlsG2>();
+#pragma unroll
+    for (int k = 0; k < 2; k++) {
+        g1aG2 p;
+        bool inf;
+        int rc = bls_g2_parse_one(in + 192 * k, p, inf);
+        if (rc) {
+            atomicOr(err, rc == 2 ? 2u : 1u);
+            return;
+        }
+        if (!inf) acc = g1_add_affine9<BlsG2>(acc, p);
+    }
+    g1_to_affine_be9<BlsG2>(out, acc);
+}
+
+__global__ void k_bls_g2_mul_single(const uint8_t *in /* 192 + 32 B */,
+                                    uint8_t *out, uint32_t *err) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    g1aG2 p;
+    bool inf;
+    int rc = bls_g2_parse_one(in, p, inf);
+    if (rc) {
+        atomicOr(err, rc == 2 ? 2u : 1u);
+        return;
+    }
+    if (inf) {
+        for (int j = 0; j < 24; j++) ((u64 *)out)[j] = 0;
+        return;
+    }
+    u64 k[4];
+    const u64 *w = (const u64 *)(in + 192);
+#pragma unroll
+    for (int j = 0; j < 4; j++) k[j] = __builtin_bswap64(w[3 - j]);
+    g1jG2 r = g1_scalar_mul9(p, k, 4);
+    g1_to_affine_be9<BlsG2>(out, r);
 }
 
 }  // namespace em

@@ -231,7 +231,13 @@ for _f in ("bls12381_g1_add", "bls12381_g1_mul", "bls12381_g1_msm",
            "bls12381_g1_combine", "bls_msm_plan_create", "bls_msm_plan_destroy",
            "bls_msm_upload_points", "bls_msm_gen_points",
            "bls_msm_download_points", "bls_msm_upload_scalars", "bls_msm_run",
-           "bls_msm_run_async", "bls_msm_sync",
+           "bls_msm_run_async", "bls_msm_sync", "bls12381_g2_add",
+           "bls12381_g2_mul", "bls12381_g2_msm", "bls_g2_msm_plan_create",
+           "bls_g2_msm_plan_destroy", "bls_g2_msm_upload_points",
+           "bls_g2_msm_gen_points", "bls_g2_msm_download_points",
+           "bls_g2_msm_upload_scalars", "bls_g2_msm_run",
+           "bls_g2_msm_run_async", "bls_g2_msm_sync",
+           "bls_g2_msm_run_partial", "bls_g2_msm_last_times",
            "bls_msm_run_partial", "bls_msm_last_times", "bls_msm_precompute"):
     getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
 
@@ -336,6 +342,99 @@ class BlsMsmPlan:
     def destroy(self):
         if self._p:
             _lib.ethrex_mi355_bls_msm_plan_destroy(self._p)
+            self._p = ctypes.c_void_p()
+
+    def __del__(self):
+        try:
+            self.destroy()
+        except Exception:
+            pass
+
+
+def bls_g2_add(p1: bytes, p2: bytes):
+    out = (ctypes.c_uint8 * 192)()
+    rc = _lib.ethrex_mi355_bls12381_g2_add(_buf(p1), _buf(p2), out)
+    return rc, bytes(out)
+
+
+def bls_g2_mul(point: bytes, scalar: bytes):
+    out = (ctypes.c_uint8 * 192)()
+    rc = _lib.ethrex_mi355_bls12381_g2_mul(_buf(point), _buf(scalar), out)
+    return rc, bytes(out)
+
+
+def bls_g2_msm(points: bytes, scalars: bytes, n: int):
+    out = (ctypes.c_uint8 * 192)()
+    rc = _lib.ethrex_mi355_bls12381_g2_msm(_buf(points), _buf(scalars),
+                                           ctypes.c_size_t(n), out)
+    return rc, bytes(out)
+
+
+class BlsG2MsmPlan:
+    """Device-resident BLS12-381 G2 MSM plan (EIP-2537 G2MSM shape)."""
+
+    def __init__(self, n: int):
+        self.n = n
+        self._p = ctypes.c_void_p()
+        self._pending = []
+        _check(_lib.ethrex_mi355_bls_g2_msm_plan_create(ctypes.c_size_t(n),
+                                                        ctypes.byref(self._p)),
+               "bls_g2_msm_plan_create")
+
+    def upload_points(self, points: bytes):
+        _check(_lib.ethrex_mi355_bls_g2_msm_upload_points(self._p,
+                                                          _buf(points)),
+               "bls_g2_msm_upload_points")
+
+    def gen_points(self, start: int = 0):
+        _check(_lib.ethrex_mi355_bls_g2_msm_gen_points(
+            self._p, ctypes.c_uint64(start)), "bls_g2_msm_gen_points")
+
+    def download_points(self) -> bytes:
+        out = (ctypes.c_uint8 * (192 * self.n))()
+        _check(_lib.ethrex_mi355_bls_g2_msm_download_points(self._p, out),
+               "bls_g2_msm_download_points")
+        return bytes(out)
+
+    def upload_scalars(self, scalars: bytes):
+        _check(_lib.ethrex_mi355_bls_g2_msm_upload_scalars(self._p,
+                                                           _buf(scalars)),
+               "bls_g2_msm_upload_scalars")
+
+    def run(self) -> bytes:
+        out = (ctypes.c_uint8 * 192)()
+        _check(_lib.ethrex_mi355_bls_g2_msm_run(self._p, out),
+               "bls_g2_msm_run")
+        return bytes(out)
+
+    def run_partial(self) -> bytes:
+        out = (ctypes.c_uint8 * 288)()
+        _check(_lib.ethrex_mi355_bls_g2_msm_run_partial(self._p, out),
+               "bls_g2_msm_run_partial")
+        return bytes(out)
+
+    def run_async(self):
+        buf = (ctypes.c_uint8 * 192)()
+        _check(_lib.ethrex_mi355_bls_g2_msm_run_async(self._p, buf),
+               "bls_g2_msm_run_async")
+        self._pending.append(buf)
+
+    def sync(self) -> bytes:
+        _check(_lib.ethrex_mi355_bls_g2_msm_sync(self._p), "bls_g2_msm_sync")
+        outs = [bytes(b) for b in self._pending]
+        self._pending = []
+        return outs[-1] if outs else b""
+
+    def last_times(self):
+        t = (ctypes.c_double * 5)()
+        _check(_lib.ethrex_mi355_bls_g2_msm_last_times(self._p, t),
+               "bls_g2_msm_last_times")
+        return {"digits_sort_ms": t[0], "bucket_acc_ms": t[1],
+                "reduce_ms": t[2], "combine_ms": t[3], "total_ms": t[4]}
+
+    def destroy(self):
+        if self._p:
+            _lib.ethrex_mi355_bls_g2_msm_plan_destroy(self._p)
             self._p = ctypes.c_void_p()
 
     def __del__(self):

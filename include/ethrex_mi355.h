@@ -148,6 +148,38 @@ int ethrex_mi355_bls_msm_run(em_bls_msm_plan *plan, uint8_t out[96]);
 int ethrex_mi355_bls_msm_run_partial(em_bls_msm_plan *plan, uint8_t out[144]);
 int ethrex_mi355_bls_msm_run_async(em_bls_msm_plan *plan, uint8_t out[96]);
 int ethrex_mi355_bls_msm_sync(em_bls_msm_plan *plan);
+
+/* ---- BLS12-381 G2 (EIP-2537 / blst semantics, bls_blst.rs:338-441):
+ * 192-byte points x.c0||x.c1||y.c0||y.c1 (canonical 48-B BE coords over
+ * Fp2 = Fp[u]/(u^2+1)), (0,0,0,0) = identity; add checks on-curve only,
+ * MSM additionally enforces the r-subgroup check per point.  Jacobian
+ * exchange payload: 288 B (X||Y||Z, each c0||c1). ---- */
+typedef struct em_bls_g2_msm_plan em_bls_g2_msm_plan;
+int ethrex_mi355_bls12381_g2_add(const uint8_t p1[192], const uint8_t p2[192],
+                                 uint8_t out[192]);
+int ethrex_mi355_bls12381_g2_mul(const uint8_t point[192],
+                                 const uint8_t scalar[32], uint8_t out[192]);
+int ethrex_mi355_bls12381_g2_msm(const uint8_t *points192,
+                                 const uint8_t *scalars32, size_t n,
+                                 uint8_t out[192]);
+int ethrex_mi355_bls_g2_msm_plan_create(size_t n, em_bls_g2_msm_plan **plan);
+int ethrex_mi355_bls_g2_msm_plan_destroy(em_bls_g2_msm_plan *plan);
+int ethrex_mi355_bls_g2_msm_upload_points(em_bls_g2_msm_plan *plan,
+                                          const uint8_t *points192);
+int ethrex_mi355_bls_g2_msm_gen_points(em_bls_g2_msm_plan *plan,
+                                       uint64_t start);
+int ethrex_mi355_bls_g2_msm_download_points(em_bls_g2_msm_plan *plan,
+                                            uint8_t *out192);
+int ethrex_mi355_bls_g2_msm_upload_scalars(em_bls_g2_msm_plan *plan,
+                                           const uint8_t *scalars32);
+int ethrex_mi355_bls_g2_msm_run(em_bls_g2_msm_plan *plan, uint8_t out[192]);
+int ethrex_mi355_bls_g2_msm_run_async(em_bls_g2_msm_plan *plan,
+                                      uint8_t out[192]);
+int ethrex_mi355_bls_g2_msm_sync(em_bls_g2_msm_plan *plan);
+int ethrex_mi355_bls_g2_msm_run_partial(em_bls_g2_msm_plan *plan,
+                                        uint8_t out[288]);
+int ethrex_mi355_bls_g2_msm_last_times(em_bls_g2_msm_plan *plan,
+                                       double times_ms[5]);
 int ethrex_mi355_bls_msm_last_times(em_bls_msm_plan *plan, double times_ms[5]);
 
 /* n elements uniform in [0, r_bls) (canonical blob field elements) */
