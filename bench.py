@@ -258,6 +258,28 @@ def main():
         }
         bp.destroy()
 
+        # G2 half of the EIP-2537 precompile surface:
+        # pipelined 4096-point G2 MSM over Fp2
+        g2p = ethrex_amd.BlsG2MsmPlan(4096)
+        g2p.gen_points(0)
+        g2p.upload_scalars(ethrex_amd.bls_gen_fr(47, 4096))
+        g2_first = None
+        for _ in range(max(args.warmup, 1)):
+            g2_first = g2p.run()
+        t3 = time.perf_counter()
+        for _ in range(args.steps):
+            g2p.run_async()
+        g2_last = g2p.sync()
+        g2_dt = (time.perf_counter() - t3) / args.steps
+        assert g2_last == g2_first, "pipelined G2 != sync G2"
+        bls["g2_msm_4096"] = {
+            "metric": "bls12381_g2_msm_4096_per_s",
+            "value": 1.0 / g2_dt,
+            "ms_per_msm": g2_dt * 1000.0,
+            "phase_ms": {k: round(v, 3) for k, v in g2p.last_times().items()},
+        }
+        g2p.destroy()
+
     cpu_baseline = None
     if rank == 0 and not args.no_cpu_baseline:
         cpu_baseline = cpu_baseline_leg()
