@@ -66,6 +66,21 @@ def test_gpu_g2_msm_plan_parity_4096(gpu, oracle_mod):
     assert rc == 0 and got == want
 
 
+def test_gpu_g2_msm_plan_parity_2_17_c16(gpu, oracle_mod):
+    """n > 2^16 switches the G2 plan to the c=16 window config — the only
+    test that exercises the large-window G2 instantiation."""
+    n = 1 << 17
+    plan = gpu.BlsG2MsmPlan(n)
+    plan.gen_points(0)
+    pts = plan.download_points()
+    scs = gpu.bls_gen_fr(55, n)
+    plan.upload_scalars(scs)
+    got = plan.run()
+    plan.destroy()
+    rc, want = oracle_mod.bls_g2_msm(pts, scs, n)
+    assert rc == 0 and got == want
+
+
 def test_gpu_g2_msm_identity_and_zero_scalars(gpu, oracle_mod):
     n = 256
     pts = bytearray(oracle_mod.bls_g2_gen_points(0, n))

@@ -195,6 +195,22 @@ def test_gpu_msm_parity_2_20_direct(gpu, oracle_mod):
     assert rc == 0 and got == want
 
 
+def test_gpu_msm_parity_2_22_direct(gpu, oracle_mod):
+    """Direct bit-exact parity vs the oracle at 2^22 (oracle ~10-20 s on
+    the GPU box's 256 host threads) — the largest direct-comparison size;
+    2^24 is covered by the sum identity + sharded-combine tests."""
+    n = 1 << 22
+    plan = gpu.MsmPlan(n)
+    plan.gen_points(0)
+    pts = plan.download_points()
+    scs = gpu.gen_fr(42, n)
+    plan.upload_scalars(scs)
+    got = plan.run()
+    plan.destroy()
+    rc, want = oracle_mod.g1_msm(pts, scs, n)
+    assert rc == 0 and got == want
+
+
 def test_gpu_msm_tree_edge_cases(gpu, oracle_mod, monkeypatch):
     """Exercises the EXPERIMENTAL batch-affine pairing tree (EM_MSM_TREE=1;
     active for n > 2^16 with avg bucket runs >= 8): odd n, heavy duplicate
