@@ -760,16 +760,18 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
 }
 
 template <typename C>
-static int msm_run_async_t(msm_plan_t<C> *p, uint8_t *out) {
+static int msm_run_async_t(msm_plan_t<C> *p, uint8_t *out, int out_mode = 0) {
     if (!p || !out) return EM_ERR_INPUT;
     if (!p->have_points || !p->have_scalars) {
         g_last_err = "msm_run_async: points/scalars not uploaded";
         return EM_ERR_INPUT;
     }
-    if (p->fixed_base) return msm_run_async_cfg<C, CfgFB>(p, out, 0);
+    if (p->fixed_base) return msm_run_async_cfg<C, CfgFB>(p, out, out_mode);
     if (p->cbits == 8)
-        return msm_run_async_cfg<C, msm_cfg<8, msm_plan_t<C>::SB>>(p, out, 0);
-    return msm_run_async_cfg<C, msm_cfg<16, msm_plan_t<C>::SB>>(p, out, 0);
+        return msm_run_async_cfg<C, msm_cfg<8, msm_plan_t<C>::SB>>(p, out,
+                                                                   out_mode);
+    return msm_run_async_cfg<C, msm_cfg<16, msm_plan_t<C>::SB>>(p, out,
+                                                                out_mode);
 }
 
 template <typename C>
@@ -820,6 +822,13 @@ extern "C" int ethrex_mi355_msm_run_async(em_msm_plan *p, uint8_t out[64]) {
 }
 extern "C" int ethrex_mi355_msm_sync(em_msm_plan *p) {
     return msm_sync_t((msm_plan_t<Bn254G1> *)p);
+}
+/* pipelined shard step: like run_async but delivers the 96-B Jacobian
+ * partial (multi-GPU: the NEXT step's sort chain overlaps this step's
+ * compute even across the AllGather + combine exchange). */
+extern "C" int ethrex_mi355_msm_run_partial_async(em_msm_plan *p,
+                                                  uint8_t out[96]) {
+    return msm_run_async_t((msm_plan_t<Bn254G1> *)p, out, 1);
 }
 extern "C" int ethrex_mi355_msm_run_partial(em_msm_plan *p, uint8_t out[96]) {
     return msm_run_inner_t((msm_plan_t<Bn254G1> *)p, out, 1);

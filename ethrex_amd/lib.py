@@ -29,6 +29,7 @@ for _f in ("device_count", "set_device", "bn254_g1_add", "bn254_g1_mul",
            "msm_plan_create", "msm_plan_destroy", "msm_upload_points",
            "msm_gen_points", "msm_download_points", "msm_upload_scalars",
            "msm_run", "msm_run_partial", "msm_run_async", "msm_sync",
+           "msm_run_partial_async",
            "msm_last_times", "msm_combine",
            "ntt_plan_create", "ntt_plan_destroy", "ntt_upload", "ntt_run",
            "ntt_download", "ntt_last_times"):
@@ -158,6 +159,14 @@ class MsmPlan:
         overlaps this step's compute chain).  Result delivered at sync()."""
         buf = (ctypes.c_uint8 * 64)()
         _check(_lib.ethrex_mi355_msm_run_async(self._p, buf), "msm_run_async")
+        self._pending.append(buf)
+
+    def run_partial_async(self):
+        """Pipelined shard step: delivers the 96-B Jacobian partial at
+        sync() (the N>1 exchange payload)."""
+        buf = (ctypes.c_uint8 * 96)()
+        _check(_lib.ethrex_mi355_msm_run_partial_async(self._p, buf),
+               "msm_run_partial_async")
         self._pending.append(buf)
 
     def sync(self) -> bytes:

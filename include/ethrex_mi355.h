@@ -88,6 +88,9 @@ int ethrex_mi355_msm_run_partial(em_msm_plan *plan, uint8_t out[96]);
 int ethrex_mi355_msm_run_async(em_msm_plan *plan, uint8_t out[64]);
 /* drain all pipelined steps and deliver their results */
 int ethrex_mi355_msm_sync(em_msm_plan *plan);
+/* pipelined shard step: run_async delivering the 96-B Jacobian partial
+ * (the multi-GPU exchange payload) instead of the affine result */
+int ethrex_mi355_msm_run_partial_async(em_msm_plan *plan, uint8_t out[96]);
 /* combine Jacobian partials on the GPU -> affine result */
 int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96, size_t count,
                                   uint8_t out[64]);

@@ -275,6 +275,21 @@ def test_gpu_msm_async_pipeline_matches_sync(gpu, oracle_mod):
     plan.destroy()
 
 
+def test_gpu_run_partial_async_matches_sync(gpu, oracle_mod):
+    """run_partial_async (the pipelined N>1 exchange payload) delivers the
+    same 96-B Jacobian partial as the sync run_partial."""
+    n = 1 << 16
+    plan = gpu.MsmPlan(n)
+    plan.gen_points(0)
+    plan.upload_scalars(gpu.gen_fr(42, n))
+    want = plan.run_partial()
+    plan.run_partial_async()
+    plan.run_partial_async()
+    got = plan.sync()
+    plan.destroy()
+    assert got == want
+
+
 def test_gpu_plan_combine_matches_standalone(gpu, oracle_mod):
     """plan-attached combine (the per-step N>1 exchange path) matches the
     standalone combine and the oracle."""
