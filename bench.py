@@ -294,6 +294,28 @@ def main():
         }
         g2p.destroy()
 
+        # batched keccak256 (witness/trie hashing),
+        # 2^20 x 136-byte messages resident in HBM, kernel-time only
+        kn = 1 << 20
+        kmsgs = bytes(136) * kn
+        kp = ethrex_amd.KeccakPlan(len(kmsgs), kn)
+        kp.upload(kmsgs, list(range(0, 136 * (kn + 1), 136)))
+        kp.run()
+        kms = []
+        for _ in range(max(args.steps // 2, 3)):
+            kp.run()
+            kms.append(kp.last_ms())
+        kavg = sum(kms) / len(kms)
+        bls["keccak256_batch"] = {
+            "metric": "keccak256_hashes_per_s",
+            "value": kn / kavg * 1000.0,
+            "n_msgs": kn,
+            "msg_bytes": 136,
+            "ms_per_batch": round(kavg, 3),
+            "GBps": round(kn * 136 / kavg * 1000.0 / 1e9, 1),
+        }
+        kp.destroy()
+
     cpu_baseline = None
     if rank == 0 and not args.no_cpu_baseline:
         cpu_baseline = cpu_baseline_leg()

@@ -15,7 +15,7 @@ from .lib import (EM_OK, EM_ERR_POINT, EM_ERR_INPUT, EM_ERR_HIP,
                   version, g1_add, g1_mul, g1_msm, fr_ntt, g1_combine, gen_fr,
                   bls_g1_add, bls_g1_mul, bls_g1_msm, bls_g1_combine,
                   bls_gen_fr, bls_g2_add, bls_g2_mul, bls_g2_msm,
-                  BlsG2MsmPlan, last_error)
+                  BlsG2MsmPlan, keccak256_batch, KeccakPlan, last_error)
 
 __all__ = [
     "EM_OK", "EM_ERR_POINT", "EM_ERR_INPUT", "EM_ERR_HIP",
@@ -23,5 +23,6 @@ __all__ = [
     "version", "g1_add", "g1_mul", "g1_msm", "fr_ntt", "g1_combine", "gen_fr",
     "bls_g1_add", "bls_g1_mul", "bls_g1_msm", "bls_g1_combine", "bls_gen_fr",
     "bls_g2_add", "bls_g2_mul", "bls_g2_msm", "BlsG2MsmPlan",
+    "keccak256_batch", "KeccakPlan",
     "last_error",
 ]

@@ -247,6 +247,9 @@ for _f in ("bls12381_g1_add", "bls12381_g1_mul", "bls12381_g1_msm",
            "bls_g2_msm_upload_scalars", "bls_g2_msm_run",
            "bls_g2_msm_run_async", "bls_g2_msm_sync",
            "bls_g2_msm_run_partial", "bls_g2_msm_last_times",
+           "keccak256_batch", "keccak_plan_create", "keccak_plan_destroy",
+           "keccak_upload", "keccak_run", "keccak_download",
+           "keccak_last_ms",
            "bls_msm_run_partial", "bls_msm_last_times", "bls_msm_precompute"):
     getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
 
@@ -444,6 +447,59 @@ class BlsG2MsmPlan:
     def destroy(self):
         if self._p:
             _lib.ethrex_mi355_bls_g2_msm_plan_destroy(self._p)
+            self._p = ctypes.c_void_p()
+
+    def __del__(self):
+        try:
+            self.destroy()
+        except Exception:
+            pass
+
+
+def keccak256_batch(msgs: bytes, offsets, n: int):
+    """Batched Ethereum keccak256 (one-shot, PCIe-inclusive); offsets is a
+    list of n+1 byte offsets into msgs."""
+    offs = (ctypes.c_uint64 * (n + 1))(*offsets)
+    out = (ctypes.c_uint8 * (32 * n))()
+    rc = _lib.ethrex_mi355_keccak256_batch(
+        _buf(msgs) if msgs else None, offs, ctypes.c_size_t(n), out)
+    return rc, bytes(out)
+
+
+class KeccakPlan:
+    """Device-resident batched keccak256 (witness/trie hashing shape)."""
+
+    def __init__(self, max_bytes: int, max_n: int):
+        self._p = ctypes.c_void_p()
+        _check(_lib.ethrex_mi355_keccak_plan_create(
+            ctypes.c_size_t(max_bytes), ctypes.c_size_t(max_n),
+            ctypes.byref(self._p)), "keccak_plan_create")
+
+    def upload(self, msgs: bytes, offsets):
+        self.n = len(offsets) - 1
+        offs = (ctypes.c_uint64 * len(offsets))(*offsets)
+        _check(_lib.ethrex_mi355_keccak_upload(
+            self._p, _buf(msgs) if msgs else None, offs,
+            ctypes.c_size_t(self.n)), "keccak_upload")
+
+    def run(self):
+        _check(_lib.ethrex_mi355_keccak_run(self._p), "keccak_run")
+
+    def download(self) -> bytes:
+        out = (ctypes.c_uint8 * (32 * self.n))()
+        _check(_lib.ethrex_mi355_keccak_download(self._p, out),
+               "keccak_download")
+        return bytes(out)
+
+    def last_ms(self) -> float:
+        ms = ctypes.c_double(0)
+        _check(_lib.ethrex_mi355_keccak_last_ms(self._p, ctypes.byref(ms)),
+               "keccak_last_ms")
+        return ms.value
+
+    def destroy(self):
+        if self._p:
+            _lib.ethrex_mi355_keccak_plan_destroy(self._p)
             self._p = ctypes.c_void_p()
 
     def __del__(self):

@@ -102,6 +102,21 @@ int ethrex_mi355_msm_last_times(em_msm_plan *plan, double times_ms[5]);
 int ethrex_mi355_msm_combine(em_msm_plan *plan, const uint8_t *jacobians96,
                              size_t count, uint8_t out[64]);
 
+/* ---- batched Keccak-256 (witness/statement hashing; original Keccak
+ * padding, Ethereum keccak256).  offsets[n+1] delimits message i as
+ * [offsets[i], offsets[i+1]); out32 receives 32 bytes per message. ---- */
+typedef struct em_keccak_plan em_keccak_plan;
+int ethrex_mi355_keccak256_batch(const uint8_t *msgs, const uint64_t *offsets,
+                                 size_t n, uint8_t *out32);
+int ethrex_mi355_keccak_plan_create(size_t max_bytes, size_t max_n,
+                                    em_keccak_plan **plan);
+int ethrex_mi355_keccak_plan_destroy(em_keccak_plan *plan);
+int ethrex_mi355_keccak_upload(em_keccak_plan *plan, const uint8_t *msgs,
+                               const uint64_t *offsets, size_t n);
+int ethrex_mi355_keccak_run(em_keccak_plan *plan);
+int ethrex_mi355_keccak_download(em_keccak_plan *plan, uint8_t *out32);
+int ethrex_mi355_keccak_last_ms(em_keccak_plan *plan, double *ms);
+
 int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan);
 int ethrex_mi355_ntt_plan_destroy(em_ntt_plan *plan);
 int ethrex_mi355_ntt_upload(em_ntt_plan *plan, const uint8_t *elems32);
