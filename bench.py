@@ -355,6 +355,7 @@ def main():
         print(json.dumps(result), flush=True)
 
     if dist is not None:
+        dist.barrier()  # rank 0 runs the single-GPU legs; others wait here
         dist.destroy_process_group()
 
 

@@ -158,44 +158,44 @@ static int msm_destroy_t(msm_plan_t<C> *p) {
         (void)hipStreamSynchronize(p->s_sort);
         (void)hipStreamSynchronize(p->s_comp);
     }
-    hipFree(p->d_pts);
-    hipFree(p->d_pts_ext);
-    hipFree(p->d_inf);
-    hipFree(p->d_scalars);
-    hipFree(p->d_scratch);
-    hipFree(p->d_keys);
-    hipFree(p->d_vals);
-    hipFree(p->d_keys_out);
-    hipFree(p->d_vals_out);
-    hipFree(p->d_sort_tmp);
-    hipFree(p->d_offsets);
-    hipFree(p->d_blen);
-    hipFree(p->d_blen_out);
-    hipFree(p->d_bids);
-    hipFree(p->d_sched);
-    hipFree(p->d_buckets);
-    hipFree(p->d_seg_sum);
-    hipFree(p->d_seg_wsum);
-    hipFree(p->d_partials);
-    hipFree(p->d_windows);
-    hipFree(p->d_out);
-    hipFree(p->d_err);
-    hipFree(p->d_lvl[0]);
-    hipFree(p->d_lvl[1]);
-    hipFree(p->d_aux);
-    hipFree(p->d_loff[0]);
-    hipFree(p->d_loff[1]);
-    hipFree(p->d_cnt);
-    hipFree(p->d_scan_tmp);
-    hipFree(p->d_keys_out2);
-    hipFree(p->d_vals_out2);
-    hipFree(p->d_offsets2);
-    hipFree(p->d_sched2);
-    hipFree(p->d_out2);
-    if (p->h_out[0]) hipHostFree(p->h_out[0]);
-    if (p->h_out[1]) hipHostFree(p->h_out[1]);
-    if (p->s_sort) hipStreamDestroy(p->s_sort);
-    if (p->s_comp) hipStreamDestroy(p->s_comp);
+    (void)hipFree(p->d_pts);
+    (void)hipFree(p->d_pts_ext);
+    (void)hipFree(p->d_inf);
+    (void)hipFree(p->d_scalars);
+    (void)hipFree(p->d_scratch);
+    (void)hipFree(p->d_keys);
+    (void)hipFree(p->d_vals);
+    (void)hipFree(p->d_keys_out);
+    (void)hipFree(p->d_vals_out);
+    (void)hipFree(p->d_sort_tmp);
+    (void)hipFree(p->d_offsets);
+    (void)hipFree(p->d_blen);
+    (void)hipFree(p->d_blen_out);
+    (void)hipFree(p->d_bids);
+    (void)hipFree(p->d_sched);
+    (void)hipFree(p->d_buckets);
+    (void)hipFree(p->d_seg_sum);
+    (void)hipFree(p->d_seg_wsum);
+    (void)hipFree(p->d_partials);
+    (void)hipFree(p->d_windows);
+    (void)hipFree(p->d_out);
+    (void)hipFree(p->d_err);
+    (void)hipFree(p->d_lvl[0]);
+    (void)hipFree(p->d_lvl[1]);
+    (void)hipFree(p->d_aux);
+    (void)hipFree(p->d_loff[0]);
+    (void)hipFree(p->d_loff[1]);
+    (void)hipFree(p->d_cnt);
+    (void)hipFree(p->d_scan_tmp);
+    (void)hipFree(p->d_keys_out2);
+    (void)hipFree(p->d_vals_out2);
+    (void)hipFree(p->d_offsets2);
+    (void)hipFree(p->d_sched2);
+    (void)hipFree(p->d_out2);
+    if (p->h_out[0]) (void)hipHostFree(p->h_out[0]);
+    if (p->h_out[1]) (void)hipHostFree(p->h_out[1]);
+    if (p->s_sort) (void)hipStreamDestroy(p->s_sort);
+    if (p->s_comp) (void)hipStreamDestroy(p->s_comp);
     delete p;
     return EM_OK;
 }
@@ -291,13 +291,13 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
             if (e == hipErrorOutOfMemory) {
                 // tree buffers don't fit (very large n): fall back to the
                 // XYZZ bucket walk rather than failing plan creation
-                hipFree(p->d_lvl[0]);
-                hipFree(p->d_lvl[1]);
-                hipFree(p->d_aux);
-                hipFree(p->d_loff[0]);
-                hipFree(p->d_loff[1]);
-                hipFree(p->d_cnt);
-                hipFree(p->d_scan_tmp);
+                (void)hipFree(p->d_lvl[0]);
+                (void)hipFree(p->d_lvl[1]);
+                (void)hipFree(p->d_aux);
+                (void)hipFree(p->d_loff[0]);
+                (void)hipFree(p->d_loff[1]);
+                (void)hipFree(p->d_cnt);
+                (void)hipFree(p->d_scan_tmp);
                 p->d_lvl[0] = p->d_lvl[1] = nullptr;
                 p->d_aux = nullptr;
                 p->d_loff[0] = p->d_loff[1] = nullptr;
@@ -1017,9 +1017,9 @@ extern "C" int ethrex_mi355_keccak_plan_create(size_t max_bytes, size_t max_n,
     for (int i = 0; i < 2 && e == hipSuccess; i++)
         e = hipEventCreate(&p->ev[i]);
     if (e != hipSuccess) {
-        hipFree(p->d_msgs);
-        hipFree(p->d_offs);
-        hipFree(p->d_out);
+        (void)hipFree(p->d_msgs);
+        (void)hipFree(p->d_offs);
+        (void)hipFree(p->d_out);
         delete p;
         return hip_fail(e, "keccak_plan_create");
     }
@@ -1029,9 +1029,9 @@ extern "C" int ethrex_mi355_keccak_plan_create(size_t max_bytes, size_t max_n,
 
 extern "C" int ethrex_mi355_keccak_plan_destroy(em_keccak_plan *p) {
     if (!p) return EM_ERR_INPUT;
-    hipFree(p->d_msgs);
-    hipFree(p->d_offs);
-    hipFree(p->d_out);
+    (void)hipFree(p->d_msgs);
+    (void)hipFree(p->d_offs);
+    (void)hipFree(p->d_out);
     delete p;
     return EM_OK;
 }
@@ -1111,9 +1111,9 @@ static int run_single(K kern, const uint8_t *a, size_t la, const uint8_t *b,
     uint32_t err;
     HIP_TRY(hipMemcpy(&err, d_err, 4, hipMemcpyDeviceToHost));
     if (!err) HIP_TRY(hipMemcpy(out, d_out, lo, hipMemcpyDeviceToHost));
-    hipFree(d_in);
-    hipFree(d_out);
-    hipFree(d_err);
+    (void)hipFree(d_in);
+    (void)hipFree(d_out);
+    (void)hipFree(d_err);
     if (err & 2u) return EM_ERR_INPUT;
     return err ? EM_ERR_POINT : EM_OK;
 }
@@ -1169,8 +1169,8 @@ extern "C" int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96,
     hipLaunchKernelGGL(k_g1_combine<Bn254G1>, dim3(1), dim3(64), 0, 0, d_in,
                        count, d_out);
     HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
-    hipFree(d_in);
-    hipFree(d_out);
+    (void)hipFree(d_in);
+    (void)hipFree(d_out);
     return EM_OK;
 }
 
@@ -1186,8 +1186,8 @@ extern "C" int ethrex_mi355_bls12381_g1_combine(const uint8_t *jacobians144,
     hipLaunchKernelGGL(k_g1_combine<BlsG1>, dim3(1), dim3(64), 0, 0, d_in,
                        count, d_out);
     HIP_TRY(hipMemcpy(out, d_out, 96, hipMemcpyDeviceToHost));
-    hipFree(d_in);
-    hipFree(d_out);
+    (void)hipFree(d_in);
+    (void)hipFree(d_out);
     return EM_OK;
 }
 
@@ -1304,19 +1304,19 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
 
 extern "C" int ethrex_mi355_ntt_plan_destroy(em_ntt_plan *p) {
     if (!p) return EM_ERR_INPUT;
-    hipFree(p->d_data);
-    hipFree(p->d_work);
-    hipFree(p->d_bytes);
-    hipFree(p->d_tw);
-    hipFree(p->d_tw_inv);
-    hipFree(p->d_twfull);
-    hipFree(p->d_twfull_inv);
-    hipFree(p->d_twrow1);
-    hipFree(p->d_twrow1_inv);
-    hipFree(p->d_twrow2);
-    hipFree(p->d_twrow2_inv);
-    hipFree(p->d_ninv);
-    hipFree(p->d_err);
+    (void)hipFree(p->d_data);
+    (void)hipFree(p->d_work);
+    (void)hipFree(p->d_bytes);
+    (void)hipFree(p->d_tw);
+    (void)hipFree(p->d_tw_inv);
+    (void)hipFree(p->d_twfull);
+    (void)hipFree(p->d_twfull_inv);
+    (void)hipFree(p->d_twrow1);
+    (void)hipFree(p->d_twrow1_inv);
+    (void)hipFree(p->d_twrow2);
+    (void)hipFree(p->d_twrow2_inv);
+    (void)hipFree(p->d_ninv);
+    (void)hipFree(p->d_err);
     delete p;
     return EM_OK;
 }
