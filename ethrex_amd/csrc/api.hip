@@ -1197,7 +1197,9 @@ struct em_ntt_plan {
     size_t n;
     int logn;
     bool fused = false;       // four-step path (13 <= logn <= 24)
+    bool fused2 = false;      // two-level four-step (25 <= logn <= 26)
     int logN1 = 0, logN2 = 0;
+    int logM1 = 0, logM2 = 0; // inner split of N2 (fused2)
     int cur = 0;              // which buffer holds the data: 0=d_data 1=d_work
     fe9 *d_data = nullptr;
     fe9 *d_work = nullptr;    // fused: transpose ping-pong buffer
@@ -1210,6 +1212,12 @@ struct em_ntt_plan {
     fe9 *d_twrow1_inv = nullptr;
     fe9 *d_twrow2 = nullptr;      // fused: N2/2 (+inv)
     fe9 *d_twrow2_inv = nullptr;
+    fe9 *d_twfull2 = nullptr;     // fused2: inner w_{N2}^j, j<N2 (+inv)
+    fe9 *d_twfull2_inv = nullptr;
+    fe9 *d_twrowA = nullptr;      // fused2: inner M1/2 row twiddles (+inv)
+    fe9 *d_twrowA_inv = nullptr;
+    fe9 *d_twrowB = nullptr;      // fused2: inner M2/2 (+inv)
+    fe9 *d_twrowB_inv = nullptr;
     fe9 *d_ninv = nullptr;        // 1/n (fe9 Montgomery)
     uint32_t *d_err = nullptr;
     hipEvent_t ev[4];
@@ -1247,6 +1255,7 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
     p->n = n;
     p->logn = logn;
     p->fused = (logn > 12 && logn <= 24);
+    p->fused2 = (logn >= 25 && logn <= 26);
     size_t half = n > 1 ? n / 2 : 1;
     hipError_t e = hipSuccess;
     auto mal = [&](void **ptr, size_t bytes) {
@@ -1255,9 +1264,18 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
     mal((void **)&p->d_data, n * sizeof(fe9));
     mal((void **)&p->d_bytes, n * 32);
     mal((void **)&p->d_err, 4);
-    if (p->fused) {
-        p->logN1 = (logn + 1) / 2;
-        p->logN2 = logn / 2;
+    if (p->fused || p->fused2) {
+        if (p->fused) {
+            p->logN1 = (logn + 1) / 2;
+            p->logN2 = logn / 2;
+        } else {
+            // two-level: outer P1 rows of 2^12; inner four-step over the
+            // 2^13/2^14-long rows (M1 x M2, both <= 2^7)
+            p->logN1 = 12;
+            p->logN2 = logn - 12;
+            p->logM1 = (p->logN2 + 1) / 2;
+            p->logM2 = p->logN2 / 2;
+        }
         mal((void **)&p->d_work, n * sizeof(fe9));
         mal((void **)&p->d_twfull, n * sizeof(fe9));
         mal((void **)&p->d_twfull_inv, n * sizeof(fe9));
@@ -1268,6 +1286,19 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
         mal((void **)&p->d_twrow2_inv,
             ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe9));
         mal((void **)&p->d_ninv, sizeof(fe9));
+        if (p->fused2) {
+            mal((void **)&p->d_twfull2, ((size_t)1 << p->logN2) * sizeof(fe9));
+            mal((void **)&p->d_twfull2_inv,
+                ((size_t)1 << p->logN2) * sizeof(fe9));
+            mal((void **)&p->d_twrowA,
+                ((size_t)1 << (p->logM1 - 1)) * sizeof(fe9));
+            mal((void **)&p->d_twrowA_inv,
+                ((size_t)1 << (p->logM1 - 1)) * sizeof(fe9));
+            mal((void **)&p->d_twrowB,
+                ((size_t)1 << (p->logM2 - 1)) * sizeof(fe9));
+            mal((void **)&p->d_twrowB_inv,
+                ((size_t)1 << (p->logM2 - 1)) * sizeof(fe9));
+        }
     } else {
         mal((void **)&p->d_tw, half * sizeof(fe9));
         mal((void **)&p->d_tw_inv, half * sizeof(fe9));
@@ -1277,7 +1308,7 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
         ethrex_mi355_ntt_plan_destroy(p);
         return hip_fail(e, "ntt_plan_create");
     }
-    if (p->fused) {
+    if (p->fused || p->fused2) {
         int rc2;
         if ((rc2 = gen_tw_table(p->d_twfull, n, logn, false, logn))) return rc2;
         if ((rc2 = gen_tw_table(p->d_twfull_inv, n, logn, true, logn))) return rc2;
@@ -1289,6 +1320,23 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
                                 p->logN2, false, p->logN2))) return rc2;
         if ((rc2 = gen_tw_table(p->d_twrow2_inv, (size_t)1 << (p->logN2 - 1),
                                 p->logN2, true, p->logN2))) return rc2;
+        if (p->fused2) {
+            size_t n2 = (size_t)1 << p->logN2;
+            if ((rc2 = gen_tw_table(p->d_twfull2, n2, p->logN2, false,
+                                    p->logN2))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twfull2_inv, n2, p->logN2, true,
+                                    p->logN2))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twrowA, (size_t)1 << (p->logM1 - 1),
+                                    p->logM1, false, p->logM1))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twrowA_inv,
+                                    (size_t)1 << (p->logM1 - 1), p->logM1,
+                                    true, p->logM1))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twrowB, (size_t)1 << (p->logM2 - 1),
+                                    p->logM2, false, p->logM2))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twrowB_inv,
+                                    (size_t)1 << (p->logM2 - 1), p->logM2,
+                                    true, p->logM2))) return rc2;
+        }
         fe9 ninv = fe9_load(bn254::FR9_INV_POW2[logn]);
         HIP_TRY(hipMemcpy(p->d_ninv, &ninv, sizeof(fe9), hipMemcpyHostToDevice));
     } else {
@@ -1310,6 +1358,12 @@ extern "C" int ethrex_mi355_ntt_plan_destroy(em_ntt_plan *p) {
     (void)hipFree(p->d_tw);
     (void)hipFree(p->d_tw_inv);
     (void)hipFree(p->d_twfull);
+    (void)hipFree(p->d_twfull2);
+    (void)hipFree(p->d_twfull2_inv);
+    (void)hipFree(p->d_twrowA);
+    (void)hipFree(p->d_twrowA_inv);
+    (void)hipFree(p->d_twrowB);
+    (void)hipFree(p->d_twrowB_inv);
     (void)hipFree(p->d_twfull_inv);
     (void)hipFree(p->d_twrow1);
     (void)hipFree(p->d_twrow1_inv);
@@ -1362,6 +1416,47 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
         hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256), 0,
                            0, cur, oth, N1, N2);
         p->cur ^= 1;
+    } else if (p->fused2) {
+        // two-level four-step (25 <= logn <= 26): the outer P2 row length
+        // (2^13/2^14) exceeds the 4096-element LDS row kernel, so each of
+        // the N1 outer rows is itself transformed by a BATCHED inner
+        // four-step (M1 x M2, both <= 2^7).  The inner transform is
+        // natural-in / natural-out, so it composes exactly where the
+        // one-level P2 sat.  8 full-array passes instead of logn.
+        uint32_t N1 = 1u << p->logN1, N2 = 1u << p->logN2;
+        uint32_t M1 = 1u << p->logM1, M2 = 1u << p->logM2;
+        const fe9 *tr1 = inverse ? p->d_twrow1_inv : p->d_twrow1;
+        const fe9 *tf = inverse ? p->d_twfull_inv : p->d_twfull;
+        const fe9 *tf2 = inverse ? p->d_twfull2_inv : p->d_twfull2;
+        const fe9 *trA = inverse ? p->d_twrowA_inv : p->d_twrowA;
+        const fe9 *trB = inverse ? p->d_twrowB_inv : p->d_twrowB;
+        // T0 + P1 + T1: outer column NTTs (length N1) + w_n^(k1 c)
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256),
+                           0, 0, cur, oth, N1, N2);
+        HIP_TRY(hipEventRecord(p->ev[1], 0));
+        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(1024), 0, 0, oth,
+                           p->logN1, tr1, tf, (const fe9 *)nullptr);
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(N1 / 32, N2 / 32), dim3(256),
+                           0, 0, oth, cur, N2, N1);
+        // inner batched four-step over the N1 rows of length N2 = M1*M2
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(M2 / 32, M1 / 32, N1),
+                           dim3(256), 0, 0, cur, oth, M1, M2);
+        hipLaunchKernelGGL(k_ntt_row_small, dim3(N1 * M2 * M1 / 1024),
+                           dim3(512), 0, 0, oth, p->logM1, trA, tf2,
+                           (const fe9 *)nullptr, M2 - 1);
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(M1 / 32, M2 / 32, N1),
+                           dim3(256), 0, 0, oth, cur, M2, M1);
+        hipLaunchKernelGGL(k_ntt_row_small, dim3(N1 * M1 * M2 / 1024),
+                           dim3(512), 0, 0, cur, p->logM2, trB,
+                           (const fe9 *)nullptr,
+                           inverse ? p->d_ninv : (const fe9 *)nullptr,
+                           0xffffffffu);
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(M2 / 32, M1 / 32, N1),
+                           dim3(256), 0, 0, cur, oth, M1, M2);
+        // outer T2 -> natural order
+        hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256),
+                           0, 0, oth, cur, N1, N2);
+        // data ends in `cur` (8 passes): no buffer flip
     } else {
         if (n > 1) {
             hipLaunchKernelGGL(k_bit_reverse, dim3(blocks_for(n, 256)), dim3(256),

@@ -95,24 +95,27 @@ __global__ void k_ntt_stage(fe9 *__restrict__ a, const fe9 *__restrict__ tw,
 //   T0: A1[c][r] = A[r][c]; P1: row NTT_N1 + w^(k1*c) twiddle;
 //   T1; P2: row NTT_N2 (+ 1/n for iNTT); T2 -> natural order.
 
-// tiled fe9 transpose, 32x32 tiles (+1 pad column for LDS banking)
+// tiled fe9 transpose, 32x32 tiles (+1 pad column for LDS banking);
+// grid.z = batch of independent R x C sub-matrices at stride R*C
 __global__ void __launch_bounds__(256)
 k_transpose_fe9(const fe9 *__restrict__ src, fe9 *__restrict__ dst,
                 uint32_t R, uint32_t C) {
     __shared__ fe9 tile[32][33];
+    size_t base = (size_t)blockIdx.z * R * C;
     uint32_t c0 = blockIdx.x * 32, r0 = blockIdx.y * 32;
     uint32_t tx = threadIdx.x & 31, ty = threadIdx.x >> 5;  // 8 rows/pass
     for (uint32_t dy = ty; dy < 32; dy += 8)
-        tile[dy][tx] = src[(size_t)(r0 + dy) * C + c0 + tx];
+        tile[dy][tx] = src[base + (size_t)(r0 + dy) * C + c0 + tx];
     __syncthreads();
     for (uint32_t dy = ty; dy < 32; dy += 8)
-        dst[(size_t)(c0 + dy) * R + r0 + tx] = tile[tx][dy];
+        dst[base + (size_t)(c0 + dy) * R + r0 + tx] = tile[tx][dy];
 }
 
 // one row NTT of length M = 2^logM fully in LDS (fe9: 4096*36 B = 144 KiB).
 __global__ void __launch_bounds__(1024)
 k_ntt_row(fe9 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
-          const fe9 *__restrict__ tw_full, const fe9 *__restrict__ scale) {
+          const fe9 *__restrict__ tw_full, const fe9 *__restrict__ scale,
+          uint32_t cmask = 0xffffffffu) {
     __shared__ fe9 smem[4096];
     const uint32_t M = 1u << logM;
     fe9 *row = data + (size_t)blockIdx.x * M;
@@ -161,12 +164,53 @@ k_ntt_row(fe9 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
         }
         __syncthreads();
     }
-    uint64_t c = blockIdx.x;
+    uint64_t c = blockIdx.x & cmask;
     for (uint32_t k = threadIdx.x; k < M; k += blockDim.x) {
         fe9 x = smem[k];
         if (tw_full) x = mont_mul9<Fr9T>(x, tw_full[(size_t)k * c]);
         if (scale) x = mont_mul9<Fr9T>(x, *scale);
         row[k] = x;
+    }
+}
+
+// packed small-row NTT: 1024/M rows of length M <= 512 per 36-KB block
+// (the 144-KB k_ntt_row runs one block/CU and idles 1-(M/4096) of its
+// threads on the two-level path's 64/128-point inner rows).  One butterfly
+// per thread per stage; rows packed smem[rr*M + j].
+__global__ void __launch_bounds__(512)
+k_ntt_row_small(fe9 *__restrict__ data, int logM,
+                const fe9 *__restrict__ tw_row,
+                const fe9 *__restrict__ tw_full,
+                const fe9 *__restrict__ scale, uint32_t cmask) {
+    __shared__ fe9 smem[1024];
+    const uint32_t M = 1u << logM;
+    const uint32_t RPB = 1024u >> logM;  // rows per block
+    size_t row0 = (size_t)blockIdx.x * RPB;
+    fe9 *base = data + row0 * M;
+    for (uint32_t i = threadIdx.x; i < 1024; i += blockDim.x) {
+        uint32_t rr = i >> logM, j = i & (M - 1);
+        smem[rr * M + (__brev(j) >> (32 - logM))] = base[i];
+    }
+    __syncthreads();
+    for (int s = 1; s <= logM; s++) {
+        uint32_t half = 1u << (s - 1);
+        uint32_t t = threadIdx.x;  // 512 butterflies = 1024 elements
+        uint32_t rr = t >> (logM - 1), ti = t & ((M >> 1) - 1);
+        uint32_t j = ti & (half - 1);
+        uint32_t idx = rr * M + ((ti >> (s - 1)) << s) + j;
+        fe9 u = smem[idx];
+        fe9 v = mont_mul9<Fr9T>(smem[idx + half], tw_row[j << (logM - s)]);
+        smem[idx] = add9_n<Fr9T>(u, v);
+        smem[idx + half] = subm9<Fr9T>(u, v);
+        __syncthreads();
+    }
+    for (uint32_t i = threadIdx.x; i < 1024; i += blockDim.x) {
+        uint32_t rr = i >> logM, kk = i & (M - 1);
+        uint64_t c = (uint64_t)((row0 + rr) & cmask);
+        fe9 x = smem[i];
+        if (tw_full) x = mont_mul9<Fr9T>(x, tw_full[(size_t)kk * c]);
+        if (scale) x = mont_mul9<Fr9T>(x, *scale);
+        base[i] = x;
     }
 }
 

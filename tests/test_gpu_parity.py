@@ -195,6 +195,36 @@ def test_gpu_msm_parity_2_20_direct(gpu, oracle_mod):
     assert rc == 0 and got == want
 
 
+def test_gpu_ntt_parity_2_25_two_level(gpu, oracle_mod):
+    """The two-level four-step path (logn 25-26: outer 2^12 columns, inner
+    batched M1 x M2 four-step) vs the oracle at 2^25 (~10 s host time)."""
+    n = 1 << 25
+    elems = gpu.gen_fr(46, n)
+    plan = gpu.NttPlan(n)
+    plan.upload(elems)
+    plan.run(False)
+    fwd = plan.download()
+    rc, want = oracle_mod.fr_ntt(elems, n, False)
+    assert rc == 0 and fwd == want
+    plan.run(True)  # inverse of the forward: back to inputs
+    plan.run(False)
+    assert plan.download() == fwd
+    plan.destroy()
+
+
+def test_gpu_ntt_roundtrip_2_26(gpu):
+    n = 1 << 26
+    elems = gpu.gen_fr(47, n)
+    plan = gpu.NttPlan(n)
+    plan.upload(elems)
+    plan.run(False)
+    fwd = plan.download()
+    assert fwd != elems
+    plan.run(True)
+    assert plan.download() == elems
+    plan.destroy()
+
+
 def test_gpu_msm_parity_2_22_direct(gpu, oracle_mod):
     """Direct bit-exact parity vs the oracle at 2^22 (oracle ~10-20 s on
     the GPU box's 256 host threads) — the largest direct-comparison size;
