@@ -380,6 +380,8 @@ static int msm_upload_points_t(msm_plan_t<C> *p, const uint8_t *points) {
 template <typename C>
 static int msm_gen_points_t(msm_plan_t<C> *p, uint64_t start) {
     if (!p) return EM_ERR_INPUT;
+    int rc0 = msm_sync_t(p);  // drain async steps: d_pts is live on s_comp
+    if (rc0) return rc0;
     if constexpr (std::is_same_v<C, BlsG2>) {
         hipLaunchKernelGGL(k_bls_g2_gen_points, dim3(blocks_for(p->n, 256)),
                            dim3(256), 0, 0, p->d_pts, p->d_inf, p->n, start);
@@ -399,6 +401,8 @@ static int msm_gen_points_t(msm_plan_t<C> *p, uint64_t start) {
 template <typename C>
 static int msm_download_points_t(msm_plan_t<C> *p, uint8_t *out) {
     if (!p || !out || !p->have_points) return EM_ERR_INPUT;
+    int rc0 = msm_sync_t(p);  // drain async steps: d_scratch races otherwise
+    if (rc0) return rc0;
     constexpr int PB = msm_plan_t<C>::PB;
     if constexpr (std::is_same_v<C, BlsG2>) {
         hipLaunchKernelGGL(k_bls_g2_points_to_be, dim3(blocks_for(p->n, 256)),
@@ -422,6 +426,8 @@ static int msm_download_points_t(msm_plan_t<C> *p, uint8_t *out) {
 template <typename C>
 static int msm_precompute_t(msm_plan_t<C> *p) {
     if (!p || !p->have_points) return EM_ERR_INPUT;
+    int rc0 = msm_sync_t(p);  // drain async steps before rebuilding tables
+    if (rc0) return rc0;
     if (p->cbits != 8) {
         g_last_err = "fixed-base precompute requires n <= 65536";
         return EM_ERR_INPUT;

@@ -33,11 +33,6 @@ __device__ __host__ __forceinline__ bool fe_is_zero(const fe4 &a) {
     return (a.v[0] | a.v[1] | a.v[2] | a.v[3]) == 0;
 }
 
-__device__ __host__ __forceinline__ bool fe_eq(const fe4 &a, const fe4 &b) {
-    return ((a.v[0] ^ b.v[0]) | (a.v[1] ^ b.v[1]) | (a.v[2] ^ b.v[2]) |
-            (a.v[3] ^ b.v[3])) == 0;
-}
-
 // a >= b ?
 __device__ __host__ __forceinline__ bool fe_geq(const fe4 &a, const fe4 &b) {
 #pragma unroll
@@ -45,67 +40,6 @@ __device__ __host__ __forceinline__ bool fe_geq(const fe4 &a, const fe4 &b) {
         if (a.v[i] != b.v[i]) return a.v[i] > b.v[i];
     }
     return true;
-}
-
-template <typename F>
-__device__ __host__ __forceinline__ fe4 fe_load_const(const u64 (&c)[4]) {
-    return fe4{{c[0], c[1], c[2], c[3]}};
-}
-
-// ---- modular add/sub (F = bn254::Fq or bn254::Fr) ----
-
-template <typename F>
-__device__ __host__ __forceinline__ fe4 mod_add(const fe4 &a, const fe4 &b) {
-    u64 r[4];
-    u128 c = 0;
-#pragma unroll
-    for (int i = 0; i < 4; i++) {
-        c += (u128)a.v[i] + b.v[i];
-        r[i] = (u64)c;
-        c >>= 64;
-    }
-    // subtract MOD if carry or r >= MOD
-    fe4 out{{r[0], r[1], r[2], r[3]}};
-    bool ge = (bool)c ||
-              fe_geq(out, fe4{{F::MOD[0], F::MOD[1], F::MOD[2], F::MOD[3]}});
-    if (ge) {
-        u128 bor = 0;
-#pragma unroll
-        for (int i = 0; i < 4; i++) {
-            u128 t = (u128)out.v[i] - F::MOD[i] - bor;
-            out.v[i] = (u64)t;
-            bor = (t >> 64) & 1;
-        }
-    }
-    return out;
-}
-
-template <typename F>
-__device__ __host__ __forceinline__ fe4 mod_sub(const fe4 &a, const fe4 &b) {
-    u64 r[4];
-    u128 bor = 0;
-#pragma unroll
-    for (int i = 0; i < 4; i++) {
-        u128 t = (u128)a.v[i] - b.v[i] - bor;
-        r[i] = (u64)t;
-        bor = (t >> 64) & 1;
-    }
-    fe4 out{{r[0], r[1], r[2], r[3]}};
-    if (bor) {
-        u128 c = 0;
-#pragma unroll
-        for (int i = 0; i < 4; i++) {
-            c += (u128)out.v[i] + F::MOD[i];
-            out.v[i] = (u64)c;
-            c >>= 64;
-        }
-    }
-    return out;
-}
-
-template <typename F>
-__device__ __host__ __forceinline__ fe4 mod_dbl(const fe4 &a) {
-    return mod_add<F>(a, a);
 }
 
 // ---- CIOS Montgomery multiplication, 32-bit-limb form ----
@@ -174,11 +108,6 @@ __device__ __host__ __forceinline__ fe4 mont_mul(const fe4 &A, const fe4 &B) {
     return out;
 }
 
-template <typename F>
-__device__ __host__ __forceinline__ fe4 mont_sqr(const fe4 &a) {
-    return mont_mul<F>(a, a);
-}
-
 // to Montgomery form; input may be ANY 256-bit value — CIOS with b = R2 < MOD
 // and a < 2^256 keeps the accumulator within the guard limbs and fully
 // reduces, mirroring ark's from_be_bytes_mod_order for 32-byte inputs.
@@ -190,30 +119,6 @@ __device__ __host__ __forceinline__ fe4 to_mont(const fe4 &a) {
 template <typename F>
 __device__ __host__ __forceinline__ fe4 from_mont(const fe4 &a) {
     return mont_mul<F>(a, fe4{{1, 0, 0, 0}});
-}
-
-template <typename F>
-__device__ __host__ __forceinline__ fe4 fe_one_mont() {
-    return fe4{{F::R[0], F::R[1], F::R[2], F::R[3]}};
-}
-
-// x^e (Montgomery in/out), e canonical 4x64
-template <typename F>
-__device__ __forceinline__ fe4 mont_pow(const fe4 &x, const fe4 &e) {
-    fe4 acc = fe_one_mont<F>();
-    fe4 base = x;
-    for (int i = 255; i >= 0; i--) {
-        acc = mont_sqr<F>(acc);
-        if ((e.v[i >> 6] >> (i & 63)) & 1) acc = mont_mul<F>(acc, base);
-    }
-    return acc;
-}
-
-// inverse via Fermat: x^(MOD-2)
-template <typename F>
-__device__ __forceinline__ fe4 mont_inv(const fe4 &x) {
-    fe4 e{{F::MOD[0] - 2, F::MOD[1], F::MOD[2], F::MOD[3]}};  // MOD odd => no borrow
-    return mont_pow<F>(x, e);
 }
 
 // ---- big-endian byte conversion ----

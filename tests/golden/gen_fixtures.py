@@ -78,7 +78,7 @@ fx["g1_mul"] = []
 for k in [0, 1, 2, 7, 255]:
     fx["g1_mul"].append({
         "name": f"G*{k}", "point": enc(G), "scalar": k.to_bytes(32, "big").hex(),
-        "out": enc(ec_mul(k, G)) if k else "00" * 128 and "00" * 64,
+        "out": enc(ec_mul(k, G)) if k else "00" * 64,
     })
 big = (1 << 256) - 1
 fx["g1_mul"].append({
