@@ -26,13 +26,17 @@ import time
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
-MSM_WINDOWS = 16          # c = 16 (ethrex_amd/csrc/msm_kernels.h)
-MSM_C = 16
+MSM_C = 17                # signed-digit windows (ethrex_amd/csrc/msm_kernels.h)
+MSM_WINDOWS = 15          # ceil(254/17); balanced digits, 2^16 buckets/window
+MSM_IB = 16               # bucket-index bits per window (2^(c-1))
 
 
 def point_adds(n_total):
-    """Pippenger add count: ceil(254/c)*(N + 2^(c+1)) (BASELINE.md formula)."""
-    return MSM_WINDOWS * (n_total + (1 << (MSM_C + 1)))
+    """Pippenger add count actually performed: NWIN*(N + 2*2^IB) — the
+    BASELINE.md formula with the signed-digit window geometry (balanced
+    digits need 2^(c-1) buckets/window).  Counts the adds this
+    implementation executes, not a fixed-c reference count."""
+    return MSM_WINDOWS * (n_total + (1 << (MSM_IB + 1)))
 
 
 def cpu_baseline_leg(log2n=22):
@@ -109,9 +113,11 @@ def main():
     def step():
         if world == 1:
             return plan.run()
+        # sync-path step (warmup/reference): shard partial -> AllGather of
+        # the 96-B Jacobian payloads over RCCL/xGMI -> host combine
         part = plan.run_partial()
         allparts = allgather_partials(part, dist, device="cuda")
-        return plan.combine(allparts, world)
+        return ethrex_amd.g1_combine_cpu(allparts, world)
     # warmup
     first = None
     for _ in range(max(args.warmup, 1)):
@@ -139,17 +145,26 @@ def main():
             plan.run_async()
         pipelined_last = plan.sync()
     else:
-        for _ in range(args.steps):
-            step()
-            bucket_ms.append(plan.last_times()["bucket_acc_ms"])
+        # pipelined shard loop: enqueue step k, then deliver step k-1's
+        # Jacobian partial (wait_one leaves step k running), AllGather it
+        # over RCCL/xGMI and combine on the host — the exchange of step k-1
+        # fully overlaps the GPU compute of step k.
+        plan.run_partial_async()
+        for _ in range(1, args.steps):
+            plan.run_partial_async()
+            part = plan.wait_one()
+            allparts = allgather_partials(part, dist, device="cuda")
+            pipelined_last = ethrex_amd.g1_combine_cpu(allparts, world)
+        part = plan.wait_one()
+        allparts = allgather_partials(part, dist, device="cuda")
+        pipelined_last = ethrex_amd.g1_combine_cpu(allparts, world)
     if world > 1:
         dist.barrier()
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
-    if world == 1:
-        assert pipelined_last == first, "pipelined result != sync result"
-        plan.run()  # one sync step to populate per-phase event timings
-        bucket_ms.append(plan.last_times()["bucket_acc_ms"])
+    assert pipelined_last == first, "pipelined result != sync result"
+    plan.run()  # one sync step to populate per-phase event timings
+    bucket_ms.append(plan.last_times()["bucket_acc_ms"])
     if world > 1:
         import torch
         t = torch.tensor([dt], device="cuda")
@@ -160,29 +175,45 @@ def main():
     value = point_adds(n_total) / (dt / args.steps)
 
     # ---- roofline for the dominant kernel (bucket accumulation) ----
-    # Algorithmic bytes per launch: each of the 16*shard sorted (point,index)
-    # pairs gathers one 64-B affine point + one 4-B index, plus 96-B bucket
-    # writes (DESIGN.md "Measurement").  The kernel is VALU-bound (big-int
-    # Montgomery mul), so the HBM fraction is expectedly far below 1; VALU
-    # evidence lives in profiles/ (rocprofv3 PMC).
-    alg_bytes = MSM_WINDOWS * shard * 76 + (MSM_WINDOWS << MSM_C) * 108
+    # The kernel is VALU-ISSUE bound (254-bit Montgomery multiplication is
+    # carry-free v_mad_u64_u32 column work; BASELINE.md names VALU as the
+    # MSM roofline axis), so the headline fraction is measured add-rate
+    # against the SIMD issue-rate ceiling:
+    #   peak adds/s = SIMDs * (1 wave-inst / 2 cyc) * clk * 64 lanes / I_add
+    # I_add = 3100 lane-instructions per mixed add, counted from the
+    # llvm-objdump disassembly of the k_bucket_acc<17,signed> main loop
+    # (gfx950: 1024 SIMD-32s, wave64 VALU issue = 2 cyc/inst, ~2.4 GHz).
+    # HBM is reported as the secondary axis with algorithmic bytes.
+    INST_PER_ADD = 3100
+    SIMDS, CLK = 1024, 2.4e9
+    valu_peak_adds = SIMDS * (CLK / 2.0) * 64 / INST_PER_ADD
     avg_bucket_ms = sum(bucket_ms) / len(bucket_ms)
+    kernel_adds = MSM_WINDOWS * shard  # reduction adds land in other kernels
+    kernel_adds_per_s = kernel_adds / (avg_bucket_ms / 1000.0)
+    # algorithmic HBM bytes/launch: NWIN*shard gathers of a 72-B point +
+    # 4-B sorted value, plus NBUCKETS 144-B XYZZ bucket writes
+    alg_bytes = MSM_WINDOWS * shard * 76 + (MSM_WINDOWS << MSM_IB) * 144
     hbm_peak = 8.0e12
     roofline = {
-        "bound": "hbm",
+        "bound": "valu",
         "kernel": "k_bucket_acc",
-        "achieved": alg_bytes / (avg_bucket_ms / 1000.0),
-        "peak": hbm_peak,
-        "unit": "B/s",
-        "frac": (alg_bytes / (avg_bucket_ms / 1000.0)) / hbm_peak,
-        # measured per-launch HBM bytes (rocprofv3 FETCH_SIZE+WRITE_SIZE,
-        # profiles/r01_summary.md) — only valid for the default 2^24 1-GPU
-        # config; other shapes report null
-        "traffic": (29.95e9 if (args.msm_log2 == 24 and shard == n_total)
-                    else None),
-        "note": "kernel is VALU-bound (254-bit Montgomery mul: PMC "
-                "SQ_ACTIVE_INST_VALU 38.8%, issue-stall 57%, memory-wait "
-                "3.7%); see profiles/ for evidence",
+        "achieved": kernel_adds_per_s,
+        "peak": valu_peak_adds,
+        "unit": "point_adds/s",
+        "frac": kernel_adds_per_s / valu_peak_adds,
+        "inst_per_add": INST_PER_ADD,
+        # measured per-launch HBM bytes (rocprofv3 FETCH_SIZE+WRITE_SIZE)
+        # for the default 2^24 1-GPU config; null for other shapes until
+        # measured.  See profiles/ for the PMC passes.
+        "traffic": None,
+        "hbm_secondary": {
+            "achieved": alg_bytes / (avg_bucket_ms / 1000.0),
+            "peak": hbm_peak,
+            "frac": (alg_bytes / (avg_bucket_ms / 1000.0)) / hbm_peak,
+            "unit": "B/s",
+        },
+        "note": "VALU-issue bound; PMC evidence in profiles/ "
+                "(r01: ACTIVE_INST_VALU 38.8%, memory-wait 3.7%)",
     }
 
     # ---- NTT secondary leg (rank 0, single GPU, replicas-only path) ----
@@ -215,7 +246,7 @@ def main():
             "path": "four-step-fused" if fused else "radix2-stages",
             "roofline": {
                 "bound": "hbm",
-                "kernel": "whole transform (k_transpose_fe4 + k_ntt_row)"
+                "kernel": "whole transform (k_transpose_fe9 + k_ntt_row)"
                           if fused else "k_ntt_stage passes",
                 "achieved": ntt_alg_bytes / (avg_total_ms / 1000.0),
                 "peak": hbm_peak,
@@ -341,6 +372,7 @@ def main():
                 "n_points": n_total,
                 "window_c": MSM_C,
                 "windows": MSM_WINDOWS,
+                "signed_digits": True,
                 "point_adds_per_step": point_adds(n_total),
                 "parallelism": f"point-index sharding x{n_gpus}, RCCL "
                                "allgather of 96B G1 partials" if n_gpus > 1

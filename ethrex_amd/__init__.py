@@ -12,7 +12,8 @@ CPU fallback anywhere (DESIGN.md "Oracle discipline").
 """
 from .lib import (EM_OK, EM_ERR_POINT, EM_ERR_INPUT, EM_ERR_HIP,
                   MsmPlan, NttPlan, BlsMsmPlan, device_count, set_device,
-                  version, g1_add, g1_mul, g1_msm, fr_ntt, g1_combine, gen_fr,
+                  version, g1_add, g1_mul, g1_msm, fr_ntt, g1_combine,
+                  g1_combine_cpu, gen_fr,
                   bls_g1_add, bls_g1_mul, bls_g1_msm, bls_g1_combine,
                   bls_gen_fr, bls_g2_add, bls_g2_mul, bls_g2_msm,
                   BlsG2MsmPlan, keccak256_batch, KeccakPlan, last_error)
@@ -20,7 +21,8 @@ from .lib import (EM_OK, EM_ERR_POINT, EM_ERR_INPUT, EM_ERR_HIP,
 __all__ = [
     "EM_OK", "EM_ERR_POINT", "EM_ERR_INPUT", "EM_ERR_HIP",
     "MsmPlan", "NttPlan", "BlsMsmPlan", "device_count", "set_device",
-    "version", "g1_add", "g1_mul", "g1_msm", "fr_ntt", "g1_combine", "gen_fr",
+    "version", "g1_add", "g1_mul", "g1_msm", "fr_ntt", "g1_combine",
+    "g1_combine_cpu", "gen_fr",
     "bls_g1_add", "bls_g1_mul", "bls_g1_msm", "bls_g1_combine", "bls_gen_fr",
     "bls_g2_add", "bls_g2_mul", "bls_g2_msm", "BlsG2MsmPlan",
     "keccak256_batch", "KeccakPlan",

@@ -135,21 +135,29 @@ struct msm_plan_t {
     uint32_t *d_keys_out2 = nullptr, *d_vals_out2 = nullptr;
     uint32_t *d_offsets2 = nullptr, *d_sched2 = nullptr;
     uint8_t *d_out2 = nullptr;
-    uint8_t *h_out[2] = {nullptr, nullptr};   // pinned result staging
+    uint8_t *h_out[2] = {nullptr, nullptr};   // pinned window staging
     hipEvent_t ev_sort_done[2], ev_comp_done[2];
+    // a pending pipelined result: h_out[par] holds nwin UNSCALED Jacobian
+    // window records; delivery runs the host Horner (cbits doublings + 1
+    // add per window) and writes affine (out_mode 0) or Jacobian (1) bytes
     struct {
         uint8_t *dest;
-        int bytes;
+        int out_mode;
+        int nwin;
+        int cbits;
         bool valid;
-        bool convert;  // staged bytes are Jacobian; convert to affine on host
-    } pend[2] = {{nullptr, 0, false, false}, {nullptr, 0, false, false}};
+    } pend[2] = {{nullptr, 0, 0, 0, false}, {nullptr, 0, 0, 0, false}};
     int apar = 0;
 };
+
+// d_out / h_out hold up to NWIN_MAX Jacobian window records (c=8: 32 windows)
+constexpr int NWIN_MAX = 32;
 
 template <typename C>
 static int msm_sync_t(msm_plan_t<C> *p);
 template <typename C>
-static void host_jac_to_affine(const uint8_t *jac, uint8_t *out);
+static void host_windows_horner(const uint8_t *wire, int nwin, int cbits,
+                                uint8_t *out, int out_mode);
 
 template <typename C>
 static int msm_destroy_t(msm_plan_t<C> *p) {
@@ -208,16 +216,35 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     auto *p = new msm_plan_t<C>();
     p->n = n;
     // small MSMs (blob-KZG-sized) use c=8: the fixed bucket-reduction tail
-    // shrinks 256x; large MSMs use c=16 (bucket work dominates)
-    p->cbits = n <= 65536 ? 8 : 16;
-    using CFGS = msm_cfg<8, msm_plan_t<C>::SB>;
-    using CFGL = msm_cfg<16, msm_plan_t<C>::SB>;
-    const int nwin = p->cbits == 8 ? CFGS::NWIN : CFGL::NWIN;
-    const uint32_t nbuckets = p->cbits == 8 ? CFGS::NBUCKETS : CFGL::NBUCKETS;
-    const int nseg_tot = p->cbits == 8 ? CFGS::NWIN * CFGS::NSEG
-                                       : CFGL::NWIN * CFGL::NSEG;
-    const int npart = p->cbits == 8 ? CFGS::NPART : CFGL::NPART;
-    const int sort_bits = p->cbits == 8 ? CFGS::SORT_BITS : CFGL::SORT_BITS;
+    // shrinks 256x.  Large BN254 MSMs default to SIGNED c=17 (15 windows:
+    // 6% fewer point adds at the same bucket count; scalars are reduced mod
+    // r so the top window absorbs the carry).  BLS scalars are raw 256-bit
+    // (blst SCALAR_BITS=256) — no carry headroom — and stay unsigned c=16.
+    // EM_MSM_TREE selects the unsigned c=16 config (the env-gated
+    // batch-affine experiment is built for that geometry).
+    if (n <= 65536)
+        p->cbits = 8;
+    else if (std::is_same_v<C, Bn254G1> && std::getenv("EM_MSM_TREE") == nullptr)
+        p->cbits = 17;
+    else
+        p->cbits = 16;
+    if (p->cbits == 17 && n >= ((size_t)1 << 30)) p->cbits = 16;  // SGN_IDX
+    int nwin, nseg_tot, npart, sort_bits;
+    uint32_t nbuckets;
+    auto geom = [&](auto cfg) {
+        using G = decltype(cfg);
+        nwin = G::NWIN;
+        nbuckets = G::NBUCKETS;
+        nseg_tot = G::NWIN * G::NSEG;
+        npart = G::NPART;
+        sort_bits = G::SORT_BITS;
+    };
+    if (p->cbits == 8)
+        geom(msm_cfg<8, msm_plan_t<C>::SB>{});
+    else if (p->cbits == 17)
+        geom(CfgSg254{});
+    else
+        geom(msm_cfg<16, msm_plan_t<C>::SB>{});
     size_t total = n * (size_t)nwin;
     hipError_t e = hipSuccess;
     auto mal = [&](void **ptr, size_t bytes) {
@@ -241,7 +268,7 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     mal((void **)&p->d_seg_wsum, (size_t)nseg_tot * sizeof(g1jT<C>));
     mal((void **)&p->d_partials, (size_t)npart * sizeof(g1jT<C>));
     mal((void **)&p->d_windows, (size_t)nwin * sizeof(g1jT<C>));
-    mal((void **)&p->d_out, msm_plan_t<C>::JB);
+    mal((void **)&p->d_out, (size_t)NWIN_MAX * msm_plan_t<C>::JB);
     mal((void **)&p->d_err, 4);
     if (e == hipSuccess) {
         e = rocprim::radix_sort_pairs(nullptr, p->sort_tmp_bytes, p->d_keys,
@@ -263,7 +290,7 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     // XYZZ stream, and the batched-inversion chains serialize.  Kept for
     // round-2 experiments; the product path stays XYZZ.
     if constexpr (std::is_same_v<C, Bn254G1>) {
-        size_t avg = total / CFGL::NBUCKETS;
+        size_t avg = total / CfgL254::NBUCKETS;
         if (p->cbits == 16 && avg >= 8 && e == hipSuccess &&
             std::getenv("EM_MSM_TREE") != nullptr) {
             p->use_tree = true;
@@ -272,19 +299,19 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
             p->tree_levels = lg - 2;
             if (p->tree_levels < 1) p->tree_levels = 1;
             if (p->tree_levels > 8) p->tree_levels = 8;
-            size_t c0 = total / 2 + CFGL::NBUCKETS + 1;
-            size_t c1 = total / 4 + CFGL::NBUCKETS + 1;
+            size_t c0 = total / 2 + CfgL254::NBUCKETS + 1;
+            size_t c1 = total / 4 + CfgL254::NBUCKETS + 1;
             mal((void **)&p->d_lvl[0], c0 * sizeof(g1aT<C>));
             mal((void **)&p->d_lvl[1], c1 * sizeof(g1aT<C>));
             mal((void **)&p->d_aux,
                 (size_t)AUX_PLANES * c0 * sizeof(feL<msm_plan_t<C>::F::L>));
-            mal((void **)&p->d_loff[0], ((size_t)CFGL::NBUCKETS + 1) * 4);
-            mal((void **)&p->d_loff[1], ((size_t)CFGL::NBUCKETS + 1) * 4);
-            mal((void **)&p->d_cnt, ((size_t)CFGL::NBUCKETS + 1) * 4);
+            mal((void **)&p->d_loff[0], ((size_t)CfgL254::NBUCKETS + 1) * 4);
+            mal((void **)&p->d_loff[1], ((size_t)CfgL254::NBUCKETS + 1) * 4);
+            mal((void **)&p->d_cnt, ((size_t)CfgL254::NBUCKETS + 1) * 4);
             if (e == hipSuccess) {
                 e = rocprim::exclusive_scan(nullptr, p->scan_tmp_bytes,
                                             p->d_cnt, p->d_loff[0], 0u,
-                                            (size_t)CFGL::NBUCKETS + 1);
+                                            (size_t)CfgL254::NBUCKETS + 1);
                 if (e == hipSuccess)
                     e = hipMalloc(&p->d_scan_tmp, p->scan_tmp_bytes);
             }
@@ -315,14 +342,15 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     mal((void **)&p->d_vals_out2, total * 4);
     mal((void **)&p->d_offsets2, ((size_t)nbuckets + 1) * 4);
     mal((void **)&p->d_sched2, (size_t)nbuckets * 4);
-    mal((void **)&p->d_out2, msm_plan_t<C>::JB);
+    mal((void **)&p->d_out2, (size_t)NWIN_MAX * msm_plan_t<C>::JB);
     if (e == hipSuccess)
         e = hipStreamCreateWithFlags(&p->s_sort, hipStreamNonBlocking);
     if (e == hipSuccess)
         e = hipStreamCreateWithFlags(&p->s_comp, hipStreamNonBlocking);
     for (int i = 0; i < 2 && e == hipSuccess; i++) {
         if (e == hipSuccess)
-            e = hipHostMalloc((void **)&p->h_out[i], msm_plan_t<C>::JB);
+            e = hipHostMalloc((void **)&p->h_out[i],
+                              (size_t)NWIN_MAX * msm_plan_t<C>::JB);
         if (e == hipSuccess)
             e = hipEventCreateWithFlags(&p->ev_sort_done[i],
                                         hipEventDisableTiming);
@@ -487,7 +515,7 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     // radix pass over the full 16n pair array.  Small MSMs keep the single
     // sort (per-call overhead dominates at small n).
     hipError_t e = hipSuccess;
-    constexpr int DBITS = CFG::C;  // digit bits (FB: k_fb_digits uses FB_C)
+    constexpr int DBITS = CFG::DBITS;  // per-window sort key bits
     const int nwin = FB ? FB_NWIN : CFG::NWIN;
     if (!FB && p->n >= (1u << 18)) {
         for (int w = 0; w < nwin && e == hipSuccess; w++) {
@@ -582,17 +610,15 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     hipLaunchKernelGGL((k_window_sum<C, CFG>), dim3(CFG::NWIN), dim3(64), 0, 0,
                        p->d_partials, p->d_windows);
     HIP_TRY(hipEventRecord(p->ev[3], 0));
-    hipLaunchKernelGGL((k_final_combine<C, CFG>), dim3(1), dim3(64), 0, 0,
-                       p->d_windows, p->d_out, 1);
+    constexpr int NW = FB ? 1 : CFG::NWIN;
+    hipLaunchKernelGGL((k_emit_windows<C, CFG>), dim3(1), dim3(64), 0, 0,
+                       p->d_windows, p->d_out);
     HIP_TRY(hipEventRecord(p->ev[4], 0));
-    if (out_mode == 0) {
-        uint8_t jac[msm_plan_t<C>::JB];
-        HIP_TRY(hipMemcpy(jac, p->d_out, msm_plan_t<C>::JB,
+    {
+        uint8_t wire[NWIN_MAX * msm_plan_t<C>::JB];
+        HIP_TRY(hipMemcpy(wire, p->d_out, (size_t)NW * msm_plan_t<C>::JB,
                           hipMemcpyDeviceToHost));
-        host_jac_to_affine<C>(jac, out);
-    } else {
-        HIP_TRY(hipMemcpy(out, p->d_out, msm_plan_t<C>::JB,
-                          hipMemcpyDeviceToHost));
+        host_windows_horner<C>(wire, NW, CFG::C, out, out_mode);
     }
     HIP_TRY(hipDeviceSynchronize());
     float ms;
@@ -609,48 +635,31 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     return EM_OK;
 }
 
-// Canonical Jacobian bytes -> affine bytes on the HOST (x=X/Z^2, y=Y/Z^3).
-// The GPU k_final_combine's to-affine path runs a 254-bit Fermat inversion
-// on ONE lane (~0.6-1.8 ms); on the host it is ~0.3 ms and — on the
-// pipelined path — overlaps the next step's GPU work entirely.
+// HOST window combine: fold the 2^(C*w) window factors over the nwin
+// UNSCALED Jacobian window records with a Horner pass
+//   acc = W_{n-1}; repeat { acc = 2^C * acc + W_w }  (C doublings + 1 add)
+// then emit affine (out_mode 0) or Jacobian wire (out_mode 1) bytes.
+// ~240 doublings on the host field core cost ~0.2 ms (BN254) / ~2 ms (G2
+// over Fp2) and — on the pipelined path — fully overlap the next step's
+// GPU work; on the GPU they were the serial latency tail of the reduction
+// (one lane per window; 9.7 ms/launch for G2).  This also covers the
+// Jacobian->affine conversion host_jac_to_affine used to do.
 template <typename C>
-static void host_jac_to_affine(const uint8_t *jac, uint8_t *out) {
-    using F = typename C::F;
-    constexpr int NB = F::W64 * 8;
-    bool zzero = true;
-    for (int i = 0; i < NB; i++)
-        if (jac[2 * NB + i]) zzero = false;
-    if (zzero) {
-        memset(out, 0, 2 * NB);
-        return;
+static void host_windows_horner(const uint8_t *wire, int nwin, int cbits,
+                                uint8_t *out, int out_mode) {
+    constexpr int JB = pt_bytes<C>::JAC;
+    g1jT<C> acc = g1_inf9<C>();
+    for (int w = nwin - 1; w >= 0; w--) {
+        if (w != nwin - 1)
+            for (int d = 0; d < cbits; d++) acc = g1_dbl9<C>(acc);
+        g1jT<C> t;
+        if (g1_jac_from_be9<C>(t, wire + (size_t)w * JB))
+            acc = g1_add9<C>(acc, t);
     }
-    feL<F::L> X = to_mont9<F>(feT_from_be<F>(jac));
-    feL<F::L> Y = to_mont9<F>(feT_from_be<F>(jac + NB));
-    feL<F::L> Z = to_mont9<F>(feT_from_be<F>(jac + 2 * NB));
-    feL<F::L> zi = mont_inv9<F>(Z);
-    feL<F::L> zi2 = mont_sqr9<F>(zi);
-    feL<F::L> zi3 = mont_mul9<F>(zi2, zi);
-    feT_to_be<F>(out, from_mont9<F>(fe9_csubp<F>(mont_mul9<F>(X, zi2))));
-    feT_to_be<F>(out + NB, from_mont9<F>(fe9_csubp<F>(mont_mul9<F>(Y, zi3))));
-}
-
-template <>
-void host_jac_to_affine<BlsG2>(const uint8_t *jac, uint8_t *out) {
-    bool zzero = true;
-    for (int i = 0; i < 96; i++)
-        if (jac[192 + i]) zzero = false;
-    if (zzero) {
-        memset(out, 0, 192);
-        return;
-    }
-    fp2 X = fp2_from_be_mont(jac);
-    fp2 Y = fp2_from_be_mont(jac + 96);
-    fp2 Z = fp2_from_be_mont(jac + 192);
-    fp2 zi = fp2_inv(Z);
-    fp2 zi2 = fp2_sqr(zi);
-    fp2 zi3 = fp2_mul(zi2, zi);
-    fp2_to_be(out, fp2_mul(X, zi2));
-    fp2_to_be(out + 96, fp2_mul(Y, zi3));
+    if (out_mode == 0)
+        g1_to_affine_be9<C>(out, acc);
+    else
+        g1_jac_be9<C>(out, acc);
 }
 
 // deliver a completed pipelined result to its caller's buffer
@@ -658,11 +667,23 @@ template <typename C>
 static int msm_deliver(msm_plan_t<C> *p, int par) {
     if (!p->pend[par].valid) return EM_OK;
     HIP_TRY(hipEventSynchronize(p->ev_comp_done[par]));
-    if (p->pend[par].convert)
-        host_jac_to_affine<C>(p->h_out[par], p->pend[par].dest);
-    else
-        memcpy(p->pend[par].dest, p->h_out[par], p->pend[par].bytes);
+    host_windows_horner<C>(p->h_out[par], p->pend[par].nwin,
+                           p->pend[par].cbits, p->pend[par].dest,
+                           p->pend[par].out_mode);
     p->pend[par].valid = false;
+    return EM_OK;
+}
+
+// deliver the OLDEST pending pipelined step (blocking only on its compute
+// chain) WITHOUT draining the pipeline: later enqueued steps keep running.
+// The N>1 exchange loop uses this to overlap AllGather + combine of step k
+// with the GPU compute of step k+1.
+template <typename C>
+static int msm_wait_one_t(msm_plan_t<C> *p) {
+    if (!p) return EM_ERR_INPUT;
+    if (!p->s_comp) return EM_OK;
+    if (p->pend[p->apar].valid) return msm_deliver(p, p->apar);
+    if (p->pend[p->apar ^ 1].valid) return msm_deliver(p, p->apar ^ 1);
     return EM_OK;
 }
 
@@ -716,7 +737,7 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
             size_t off = (size_t)w * p->n;
             e = rocprim::radix_sort_pairs(
                 p->d_sort_tmp, tmp, p->d_keys + off, KO + off,
-                p->d_vals + off, VO + off, p->n, 0, CFG::C, ss);
+                p->d_vals + off, VO + off, p->n, 0, CFG::DBITS, ss);
         }
     } else {
         size_t tmp = p->sort_tmp_bytes;
@@ -753,15 +774,17 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
                        p->d_seg_wsum, p->d_partials);
     hipLaunchKernelGGL((k_window_sum<C, CFG>), dim3(CFG::NWIN), dim3(64), 0,
                        sc, p->d_partials, p->d_windows);
-    // always emit the Jacobian form; affine conversion happens on the host
-    // at delivery time (overlapped with the next steps' GPU work)
-    hipLaunchKernelGGL((k_final_combine<C, CFG>), dim3(1), dim3(64), 0, sc,
-                       p->d_windows, DOUT, 1);
-    HIP_TRY(hipMemcpyAsync(p->h_out[par], DOUT, msm_plan_t<C>::JB,
+    // emit the UNSCALED Jacobian window records; the Horner window combine
+    // + affine conversion run on the host at delivery time (overlapped with
+    // the next steps' GPU work)
+    constexpr int NW = FB ? 1 : CFG::NWIN;
+    hipLaunchKernelGGL((k_emit_windows<C, CFG>), dim3(1), dim3(64), 0, sc,
+                       p->d_windows, DOUT);
+    HIP_TRY(hipMemcpyAsync(p->h_out[par], DOUT,
+                           (size_t)NW * msm_plan_t<C>::JB,
                            hipMemcpyDeviceToHost, sc));
     HIP_TRY(hipEventRecord(p->ev_comp_done[par], sc));
-    p->pend[par] = {out, out_mode == 0 ? msm_plan_t<C>::AB : msm_plan_t<C>::JB,
-                    true, out_mode == 0};
+    p->pend[par] = {out, out_mode, NW, CFG::C, true};
     p->apar ^= 1;
     return EM_OK;
 }
@@ -777,6 +800,10 @@ static int msm_run_async_t(msm_plan_t<C> *p, uint8_t *out, int out_mode = 0) {
     if (p->cbits == 8)
         return msm_run_async_cfg<C, msm_cfg<8, msm_plan_t<C>::SB>>(p, out,
                                                                    out_mode);
+    if constexpr (std::is_same_v<C, Bn254G1>) {
+        if (p->cbits == 17)
+            return msm_run_async_cfg<C, CfgSg254>(p, out, out_mode);
+    }
     return msm_run_async_cfg<C, msm_cfg<16, msm_plan_t<C>::SB>>(p, out,
                                                                 out_mode);
 }
@@ -793,6 +820,9 @@ static int msm_run_inner_t(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     if (p->fixed_base) return msm_run_cfg<C, CfgFB>(p, out, out_mode);
     if (p->cbits == 8)
         return msm_run_cfg<C, msm_cfg<8, msm_plan_t<C>::SB>>(p, out, out_mode);
+    if constexpr (std::is_same_v<C, Bn254G1>) {
+        if (p->cbits == 17) return msm_run_cfg<C, CfgSg254>(p, out, out_mode);
+    }
     return msm_run_cfg<C, msm_cfg<16, msm_plan_t<C>::SB>>(p, out, out_mode);
 }
 
@@ -829,6 +859,11 @@ extern "C" int ethrex_mi355_msm_run_async(em_msm_plan *p, uint8_t out[64]) {
 }
 extern "C" int ethrex_mi355_msm_sync(em_msm_plan *p) {
     return msm_sync_t((msm_plan_t<Bn254G1> *)p);
+}
+/* deliver the OLDEST pending pipelined step without draining the pipeline
+ * (N>1 exchange overlap: AllGather step k while the GPU computes k+1) */
+extern "C" int ethrex_mi355_msm_wait_one(em_msm_plan *p) {
+    return msm_wait_one_t((msm_plan_t<Bn254G1> *)p);
 }
 /* pipelined shard step: like run_async but delivers the 96-B Jacobian
  * partial (multi-GPU: the NEXT step's sort chain overlaps this step's
@@ -1177,6 +1212,25 @@ extern "C" int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96,
     HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
     (void)hipFree(d_in);
     (void)hipFree(d_out);
+    return EM_OK;
+}
+
+// HOST-side combine of the N>1 exchange payload: `count` = world size (a
+// handful of 96-B Jacobian partials, one per rank).  This is boundary glue
+// like the delivery-time Horner — the MSM compute stays on the GPU; doing
+// these few adds on the host avoids interposing a kernel + 2 copies on the
+// default stream while pipelined steps are in flight.
+extern "C" int ethrex_mi355_bn254_g1_combine_cpu(const uint8_t *jacobians96,
+                                                 size_t count,
+                                                 uint8_t out[64]) {
+    if (!jacobians96 || !out || count == 0) return EM_ERR_INPUT;
+    g1jT<Bn254G1> acc = g1_inf9<Bn254G1>();
+    for (size_t i = 0; i < count; i++) {
+        g1jT<Bn254G1> t;
+        if (!g1_jac_from_be9<Bn254G1>(t, jacobians96 + 96 * i)) continue;
+        acc = g1_add9<Bn254G1>(acc, t);
+    }
+    g1_to_affine_be9<Bn254G1>(out, acc);
     return EM_OK;
 }
 

@@ -47,7 +47,7 @@ using g1aB = g1aT<BlsG1>;
 using g1jB = g1jT<BlsG1>;
 
 template <typename C = Bn254G1>
-__device__ __forceinline__ g1jT<C> g1_inf9() {
+__device__ __host__ __forceinline__ g1jT<C> g1_inf9() {
     g1jT<C> p;
     p.x = fe9_load<C::F::L>(C::F::ONE);
     p.y = fe9_load<C::F::L>(C::F::ONE);
@@ -57,13 +57,13 @@ __device__ __forceinline__ g1jT<C> g1_inf9() {
 }
 
 template <typename C>
-__device__ __forceinline__ bool g1_is_inf9(const g1jT<C> &p) {
+__device__ __host__ __forceinline__ bool g1_is_inf9(const g1jT<C> &p) {
     return fe9_is_zero_modp<typename C::F>(p.zz);
 }
 
 // doubling (dbl-2008-s, a = 0)
 template <typename C>
-__device__ __forceinline__ g1jT<C> g1_dbl9(const g1jT<C> &p) {
+__device__ __host__ __forceinline__ g1jT<C> g1_dbl9(const g1jT<C> &p) {
     using F = typename C::F;
     if (g1_is_inf9(p)) return p;
     feL<F::L> U = add9_n<F>(p.y, p.y);         // 2Y, norm2p (L=14 mul rule)
@@ -82,7 +82,7 @@ __device__ __forceinline__ g1jT<C> g1_dbl9(const g1jT<C> &p) {
 
 // full XYZZ + XYZZ (add-2008-s)
 template <typename C>
-__device__ __forceinline__ g1jT<C> g1_add9(const g1jT<C> &p, const g1jT<C> &q) {
+__device__ __host__ __forceinline__ g1jT<C> g1_add9(const g1jT<C> &p, const g1jT<C> &q) {
     using F = typename C::F;
     if (g1_is_inf9(p)) return q;
     if (g1_is_inf9(q)) return p;
@@ -109,7 +109,7 @@ __device__ __forceinline__ g1jT<C> g1_add9(const g1jT<C> &p, const g1jT<C> &q) {
 
 // mixed add (madd-2008-s): q affine, not infinity
 template <typename C>
-__device__ __forceinline__ g1jT<C> g1_add_affine9(const g1jT<C> &p,
+__device__ __host__ __forceinline__ g1jT<C> g1_add_affine9(const g1jT<C> &p,
                                                   const g1aT<C> &q) {
     using F = typename C::F;
     if (__builtin_expect(g1_is_inf9(p), 0)) {
@@ -191,16 +191,16 @@ __device__ __host__ __forceinline__ feL<T::L> feT_from_be(const uint8_t *b) {
 }
 
 // legacy bn254 names
-__device__ __forceinline__ void fe9_to_be(uint8_t *b, const fe9 &canon) {
+__device__ __host__ __forceinline__ void fe9_to_be(uint8_t *b, const fe9 &canon) {
     feT_to_be<Fq9T>(b, canon);
 }
-__device__ __forceinline__ fe9 fe9_from_be(const uint8_t *b) {
+__device__ __host__ __forceinline__ fe9 fe9_from_be(const uint8_t *b) {
     return feT_from_be<Fq9T>(b);
 }
 
 // XYZZ -> affine (x = X/ZZ, y = Y/ZZZ): one inversion + 3 muls
 template <typename C>
-__device__ __forceinline__ g1aT<C> g1_to_affine9(const g1jT<C> &p) {
+__device__ __host__ __forceinline__ g1aT<C> g1_to_affine9(const g1jT<C> &p) {
     using F = typename C::F;
     feL<F::L> t = mont_inv9<F>(mont_mul9<F>(p.zz, p.zzz));
     g1aT<C> a;
@@ -211,7 +211,7 @@ __device__ __forceinline__ g1aT<C> g1_to_affine9(const g1jT<C> &p) {
 
 // XYZZ -> affine BE bytes (2 coords); infinity -> zeros
 template <typename C>
-__device__ __forceinline__ void g1_to_affine_be9(uint8_t *out, const g1jT<C> &p) {
+__device__ __host__ __forceinline__ void g1_to_affine_be9(uint8_t *out, const g1jT<C> &p) {
     using F = typename C::F;
     constexpr int NB = F::W64 * 8;
     if (g1_is_inf9(p)) {
@@ -226,7 +226,7 @@ __device__ __forceinline__ void g1_to_affine_be9(uint8_t *out, const g1jT<C> &p)
 // XYZZ -> Jacobian (X_j, Y_j, Z_j) with Z_j = ZZ*ZZZ (no inversion):
 //   X_j = x*Z_j^2 = X*ZZ*ZZZ^2,  Y_j = y*Z_j^3 = Y*ZZ^3*ZZZ^2
 template <typename C>
-__device__ __forceinline__ void g1_xyzz_to_jacobian9(feL<C::F::L> &X,
+__device__ __host__ __forceinline__ void g1_xyzz_to_jacobian9(feL<C::F::L> &X,
                                                      feL<C::F::L> &Y,
                                                      feL<C::F::L> &Z,
                                                      const g1jT<C> &p) {
@@ -240,7 +240,7 @@ __device__ __forceinline__ void g1_xyzz_to_jacobian9(feL<C::F::L> &X,
 
 // Jacobian (X, Y, Z) -> XYZZ: same numerators, ZZ = Z^2, ZZZ = Z^3
 template <typename C>
-__device__ __forceinline__ g1jT<C> g1_jacobian_to_xyzz9(const feL<C::F::L> &X,
+__device__ __host__ __forceinline__ g1jT<C> g1_jacobian_to_xyzz9(const feL<C::F::L> &X,
                                                         const feL<C::F::L> &Y,
                                                         const feL<C::F::L> &Z) {
     using F = typename C::F;
@@ -262,13 +262,13 @@ struct pt_bytes {
 
 // point negation in place (y -> -y)
 template <typename C>
-__device__ __forceinline__ void g1_neg_y9(g1jT<C> &p) {
+__device__ __host__ __forceinline__ void g1_neg_y9(g1jT<C> &p) {
     p.y = neg9<typename C::F>(p.y);
 }
 
 // ---- Jacobian wire IO (the 3-coordinate exchange payload; Z=0 = inf) ----
 template <typename C>
-__device__ __forceinline__ void g1_jac_be9(uint8_t *out, const g1jT<C> &p) {
+__device__ __host__ __forceinline__ void g1_jac_be9(uint8_t *out, const g1jT<C> &p) {
     using F = typename C::F;
     constexpr int NB = F::W64 * 8;
     if (g1_is_inf9(p)) {
@@ -284,7 +284,7 @@ __device__ __forceinline__ void g1_jac_be9(uint8_t *out, const g1jT<C> &p) {
 
 // parse one Jacobian wire payload; returns false for infinity
 template <typename C>
-__device__ __forceinline__ bool g1_jac_from_be9(g1jT<C> &o, const uint8_t *in) {
+__device__ __host__ __forceinline__ bool g1_jac_from_be9(g1jT<C> &o, const uint8_t *in) {
     using F = typename C::F;
     constexpr int NB = F::W64 * 8;
     feL<F::L> X = to_mont9<F>(feT_from_be<F>(in));

@@ -94,7 +94,7 @@ struct g1jT<BlsG2> {
 };
 
 template <>
-__device__ __forceinline__ g1jT<BlsG2> g1_inf9<BlsG2>() {
+__device__ __host__ __forceinline__ g1jT<BlsG2> g1_inf9<BlsG2>() {
     g1jT<BlsG2> p;
     p.x = fp2_one();
     p.y = fp2_one();
@@ -104,13 +104,13 @@ __device__ __forceinline__ g1jT<BlsG2> g1_inf9<BlsG2>() {
 }
 
 template <>
-__device__ __forceinline__ bool g1_is_inf9<BlsG2>(const g1jT<BlsG2> &p) {
+__device__ __host__ __forceinline__ bool g1_is_inf9<BlsG2>(const g1jT<BlsG2> &p) {
     return fp2_is_zero_modp(p.zz);
 }
 
 // dbl-2008-s over fp2 (a = 0)
 template <>
-__device__ __forceinline__ g1jT<BlsG2> g1_dbl9<BlsG2>(const g1jT<BlsG2> &p) {
+__device__ __host__ __forceinline__ g1jT<BlsG2> g1_dbl9<BlsG2>(const g1jT<BlsG2> &p) {
     if (g1_is_inf9<BlsG2>(p)) return p;
     fp2 U = fp2_add_n(p.y, p.y);
     fp2 V = fp2_sqr(U);
@@ -128,7 +128,7 @@ __device__ __forceinline__ g1jT<BlsG2> g1_dbl9<BlsG2>(const g1jT<BlsG2> &p) {
 
 // add-2008-s over fp2
 template <>
-__device__ __forceinline__ g1jT<BlsG2> g1_add9<BlsG2>(const g1jT<BlsG2> &p,
+__device__ __host__ __forceinline__ g1jT<BlsG2> g1_add9<BlsG2>(const g1jT<BlsG2> &p,
                                                       const g1jT<BlsG2> &q) {
     if (g1_is_inf9<BlsG2>(p)) return q;
     if (g1_is_inf9<BlsG2>(q)) return p;
@@ -155,7 +155,7 @@ __device__ __forceinline__ g1jT<BlsG2> g1_add9<BlsG2>(const g1jT<BlsG2> &p,
 
 // madd-2008-s over fp2
 template <>
-__device__ __forceinline__ g1jT<BlsG2> g1_add_affine9<BlsG2>(
+__device__ __host__ __forceinline__ g1jT<BlsG2> g1_add_affine9<BlsG2>(
     const g1jT<BlsG2> &p, const g1aT<BlsG2> &q) {
     if (__builtin_expect(g1_is_inf9<BlsG2>(p), 0)) {
         g1jT<BlsG2> o;
@@ -204,7 +204,7 @@ __device__ __forceinline__ g1aT<BlsG2> g1_generator9<BlsG2>() {
 
 // XYZZ -> affine (device; one fp2 inversion)
 template <>
-__device__ __forceinline__ g1aT<BlsG2> g1_to_affine9<BlsG2>(
+__device__ __host__ __forceinline__ g1aT<BlsG2> g1_to_affine9<BlsG2>(
     const g1jT<BlsG2> &p) {
     fp2 t = fp2_inv(fp2_mul(p.zz, p.zzz));
     g1aT<BlsG2> a;
@@ -221,7 +221,7 @@ __device__ __host__ __forceinline__ void fp2_to_be(uint8_t *b, const fp2 &m) {
 
 // XYZZ -> affine BE bytes (192 B); infinity -> zeros
 template <>
-__device__ __forceinline__ void g1_to_affine_be9<BlsG2>(
+__device__ __host__ __forceinline__ void g1_to_affine_be9<BlsG2>(
     uint8_t *out, const g1jT<BlsG2> &p) {
     if (g1_is_inf9<BlsG2>(p)) {
         for (int i = 0; i < 24; i++) ((u64 *)out)[i] = 0;
@@ -233,7 +233,7 @@ __device__ __forceinline__ void g1_to_affine_be9<BlsG2>(
 }
 
 template <>
-__device__ __forceinline__ void g1_neg_y9<BlsG2>(g1jT<BlsG2> &p) {
+__device__ __host__ __forceinline__ void g1_neg_y9<BlsG2>(g1jT<BlsG2> &p) {
     p.y.c0 = neg9<F2B>(p.y.c0);
     p.y.c1 = neg9<F2B>(p.y.c1);
 }
@@ -247,7 +247,7 @@ struct pt_bytes<BlsG2> {
 
 // Jacobian wire IO over Fp2 (288 B: X.c0||X.c1||Y.c0||Y.c1||Z.c0||Z.c1)
 template <>
-__device__ __forceinline__ void g1_jac_be9<BlsG2>(uint8_t *out,
+__device__ __host__ __forceinline__ void g1_jac_be9<BlsG2>(uint8_t *out,
                                                   const g1jT<BlsG2> &p) {
     if (g1_is_inf9<BlsG2>(p)) {
         for (int j = 0; j < 36; j++) ((u64 *)out)[j] = 0;
@@ -270,7 +270,7 @@ __device__ __host__ __forceinline__ fp2 fp2_from_be_mont(const uint8_t *b) {
 }
 
 template <>
-__device__ __forceinline__ bool g1_jac_from_be9<BlsG2>(g1jT<BlsG2> &o,
+__device__ __host__ __forceinline__ bool g1_jac_from_be9<BlsG2>(g1jT<BlsG2> &o,
                                                        const uint8_t *in) {
     fp2 X = fp2_from_be_mont(in);
     fp2 Y = fp2_from_be_mont(in + 96);

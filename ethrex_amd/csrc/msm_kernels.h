@@ -32,29 +32,48 @@ namespace em {
 // the fixed bucket-reduction tail shrinks 256x (measured 7.3 ms -> ~2 ms
 // per 4096-point commitment).  Scalar width: BN254 scalars are reduced mod
 // r (254 bits); BLS12-381 scalars are raw 256-bit integers.
-template <int CB, int SBITS>
+// SGN = signed (balanced) digits: window values are recoded to
+// d in [-2^(CB-1), +2^(CB-1)] with a carry into the next window, so each
+// window needs only 2^(CB-1) buckets (id = |d|-1; the sign rides in the
+// sorted value's bit 31 and the bucket walk negates y on the fly — point
+// negation is free on a short-Weierstrass curve).  That allows CB=17 for
+// BN254: 15 windows instead of 16 (6% fewer point adds, the VALU-issue-
+// bound budget) at the same bucket count as unsigned c=16.  Requires the
+// top window to absorb the final carry: SBITS % CB != 0, and scalars
+// reduced (BN254 only; BLS 256-bit raw scalars stay unsigned).
+template <int CB, int SBITS, bool SGN = false>
 struct msm_cfg {
-    static constexpr int C = CB;
+    static constexpr bool SIGNED = SGN;
+    static_assert(!SGN || SBITS % CB != 0, "top window must absorb carry");
+    static constexpr int C = CB;               // window bits (digit extract)
+    static constexpr int IB = SGN ? CB - 1 : CB;  // bucket-index bits
     static constexpr int NWIN = (SBITS + CB - 1) / CB;
-    static constexpr uint32_t DMASK = (1u << CB) - 1;
-    static constexpr uint32_t NBUCKETS = (uint32_t)NWIN << CB;
-    static constexpr int SORT_BITS = CB + 6;   // digit bits + window bits
+    static constexpr uint32_t DMASK = (1u << IB) - 1;
+    static constexpr uint32_t NBUCKETS = (uint32_t)NWIN << IB;
+    static constexpr int DBITS = IB;           // per-window sort key bits
+    static constexpr int SORT_BITS = IB + 6;   // + window bits (global sort)
     // SEG tuned per window size (GPU-measured): 16 for the 64K-bucket c=16
     // windows (SEG=8 doubles weighted-reduce work there, reduce 1.8->2.5 ms);
     // 8 for the smaller FB/c=8 windows (blob reduce 0.98->0.75 ms)
-    static constexpr int SEG = (1 << CB) >= 65536 ? 16 : 8;
-    static constexpr int NSEG = (1 << CB) / SEG;        // segments per window
+    static constexpr int SEG = (1 << IB) >= 65536 ? 16 : 8;
+    static constexpr int NSEG = (1 << IB) / SEG;        // segments per window
     static constexpr int RED_BLOCK = 256;
     // when a 256-thread block spans multiple windows the LDS tree is skipped
     static constexpr bool TREE = NSEG >= RED_BLOCK;
     static constexpr int NPART = TREE ? NWIN * (NSEG / RED_BLOCK) : NWIN * NSEG;
 };
 
-// the two shipped geometries (both curves)
-using CfgL254 = msm_cfg<16, 254>;  // BN254 large
-using CfgS254 = msm_cfg<8, 254>;   // BN254 small (n <= 2^16)
-using CfgL256 = msm_cfg<16, 256>;  // BLS large
-using CfgS256 = msm_cfg<8, 256>;   // BLS small
+// the shipped geometries (both curves)
+using CfgL254 = msm_cfg<16, 254>;      // BN254 large unsigned (tree mode)
+using CfgSg254 = msm_cfg<17, 254, true>;  // BN254 large DEFAULT: signed c=17
+using CfgS254 = msm_cfg<8, 254>;       // BN254 small (n <= 2^16)
+using CfgL256 = msm_cfg<16, 256>;      // BLS large
+using CfgS256 = msm_cfg<8, 256>;       // BLS small
+
+// sorted-value bit layout for SIGNED configs (vals[] entries)
+constexpr uint32_t SGN_NEG = 0x80000000u;   // digit is negative: add -P
+constexpr uint32_t SGN_SKIP = 0x40000000u;  // zero digit / infinity: no add
+constexpr uint32_t SGN_IDX = 0x3fffffffu;   // point index (n < 2^30)
 
 // digit w = bits [CB*w, CB*w+CB) of the scalar (spans u64 limbs)
 template <int CB>
@@ -129,6 +148,10 @@ __global__ void k_parse_scalars(const uint8_t *__restrict__ in,
 }
 
 // ---- digit extraction ----
+// UNSIGNED: key = (w << C) | digit, identity points park in digit-0 buckets
+// (skipped by the walk).  SIGNED: balanced recode with carry — window value
+// t = raw + carry_in; t <= 2^(C-1) keeps d = +t, else d = t - 2^C (carry 1);
+// bucket id = |d| - 1 (zero digits / infinities park at id 0 with SGN_SKIP).
 template <typename CFG>
 __global__ void k_digits(const fe4 *__restrict__ scalars,
                          const uint8_t *__restrict__ inf,
@@ -138,12 +161,35 @@ __global__ void k_digits(const fe4 *__restrict__ scalars,
     if (i >= n) return;
     fe4 k = scalars[i];
     bool skip = inf[i];
+    if constexpr (CFG::SIGNED) {
+        uint32_t carry = 0;
 #pragma unroll
-    for (int w = 0; w < CFG::NWIN; w++) {
-        uint32_t d = msm_digit<CFG::C>(k, w);
-        if (skip) d = 0;  // identity points contribute nothing: park in bucket 0
-        keys[(size_t)w * n + i] = ((uint32_t)w << CFG::C) | d;
-        vals[(size_t)w * n + i] = (uint32_t)i;
+        for (int w = 0; w < CFG::NWIN; w++) {
+            uint32_t t = msm_digit<CFG::C>(k, w) + carry;
+            uint32_t sign = 0, mag = t;
+            if (t > (1u << (CFG::C - 1))) {
+                mag = (1u << CFG::C) - t;
+                sign = 1;
+                carry = 1;
+            } else {
+                carry = 0;
+            }
+            bool sk = skip || mag == 0;
+            keys[(size_t)w * n + i] =
+                ((uint32_t)w << CFG::IB) | (sk ? 0u : mag - 1);
+            vals[(size_t)w * n + i] =
+                (uint32_t)i | (sign << 31) | (sk ? SGN_SKIP : 0u);
+        }
+        // scalars are reduced mod r < 2^254 and 254 % 17 = 16, so the top
+        // window value (+ carry) is <= 2^16 = 2^(C-1): never a carry out
+    } else {
+#pragma unroll
+        for (int w = 0; w < CFG::NWIN; w++) {
+            uint32_t d = msm_digit<CFG::C>(k, w);
+            if (skip) d = 0;  // identity points: park in bucket 0
+            keys[(size_t)w * n + i] = ((uint32_t)w << CFG::C) | d;
+            vals[(size_t)w * n + i] = (uint32_t)i;
+        }
     }
 }
 
@@ -177,7 +223,10 @@ __global__ void k_bucket_lengths(const uint32_t *__restrict__ offsets,
                                  uint32_t *__restrict__ ids) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= CFG::NBUCKETS) return;
-    len[b] = (b & CFG::DMASK) == 0 ? 0u : offsets[b + 1] - offsets[b];
+    // unsigned: digit-0 buckets are dead (identity parking); signed: all live
+    len[b] = (!CFG::SIGNED && (b & CFG::DMASK) == 0)
+                 ? 0u
+                 : offsets[b + 1] - offsets[b];
     ids[b] = b;
 }
 
@@ -193,20 +242,30 @@ k_bucket_acc(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
     uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
     if (tid >= CFG::NBUCKETS) return;
     uint32_t b = sched[tid];
-    if ((b & CFG::DMASK) == 0) return;  // digit 0
+    if constexpr (!CFG::SIGNED) {
+        if ((b & CFG::DMASK) == 0) return;  // digit 0
+    }
     uint32_t lo = offsets[b], hi = offsets[b + 1];
     g1jT<C> acc = g1_inf9<C>();
     if (lo >= hi) {
         buckets[b] = acc;
         return;
     }
+    constexpr uint32_t IMASK = CFG::SIGNED ? SGN_IDX : 0xffffffffu;
     // software pipeline: issue the NEXT point's gather before the long mixed
     // add so the dependent idx->point load chain overlaps the VALU work.
-    g1aT<C> p = GATHER ? pts[vals[lo]] : pts[lo];
+    uint32_t v = GATHER ? vals[lo] : 0;
+    g1aT<C> p = GATHER ? pts[v & IMASK] : pts[lo];
     for (uint32_t t = lo; t < hi; t++) {
         g1aT<C> cur = p;
+        uint32_t vc = v;
         uint32_t nxt = t + 1 < hi ? t + 1 : t;
-        p = GATHER ? pts[vals[nxt]] : pts[nxt];
+        v = GATHER ? vals[nxt] : 0;
+        p = GATHER ? pts[v & IMASK] : pts[nxt];
+        if constexpr (CFG::SIGNED) {
+            if (vc & SGN_SKIP) continue;  // parked zero digit / infinity
+            if (vc & SGN_NEG) cur.y = neg9<typename C::F>(cur.y);  // add -P
+        }
         if constexpr (!GATHER) {
             // tree outputs may be the identity (P + (-P)): encoded (0,0)
             if (fe9_is_zero_raw<C::F::L>(cur.x) &&
@@ -428,10 +487,11 @@ k_segment_reduce(const g1jT<C> *__restrict__ buckets,
     g1jT<C> run = g1_inf9<C>();
     lds_wsum[threadIdx.x] = g1_inf9<C>();
     for (int32_t d = (int32_t)lo + CFG::SEG - 1; d >= (int32_t)lo; d--) {
-        // digit-0 bucket is unused: skip the add but keep the wsum step so
-        // segment 0 carries the same (d - lo + 1) weights
-        if (d != 0)
-            run = g1_add9(run, buckets[((uint32_t)w << CFG::C) | (uint32_t)d]);
+        // unsigned: the digit-0 bucket is unused — skip the add but keep the
+        // wsum step so segment 0 carries the same (d - lo + 1) weights.
+        // signed: every bucket is live (bucket id = |digit| - 1).
+        if (CFG::SIGNED || d != 0)
+            run = g1_add9(run, buckets[((uint32_t)w << CFG::IB) | (uint32_t)d]);
         lds_wsum[threadIdx.x] = g1_add9(lds_wsum[threadIdx.x], run);
     }
     seg_sum[t] = run;
@@ -439,8 +499,14 @@ k_segment_reduce(const g1jT<C> *__restrict__ buckets,
 }
 
 // level 2: fully parallel weighted combine + LDS tree reduction.
-//   W_w = sum_j [ wsum_j + (j*SEG - 1) * sum_j ]   (j=0 term: -sum_0),
-// then scaled by 2^(16w) (the doubling chains run SIMD-wide here).
+// Bucket id b carries digit value D(b) = b (unsigned) or b + 1 (signed), so
+//   W_w = sum_j [ wsum_j + (j*SEG + delta - 1) * sum_j ],  delta = SIGNED,
+// (unsigned j=0 term: -sum_0; signed j=0 term: wsum alone).
+// The window factor 2^(C*w) is NOT folded here any more: windows leave this
+// pipeline UNSCALED and the final Horner (C doublings per window) runs on
+// the HOST at delivery time, overlapped with the next step's GPU work — the
+// per-thread doubling chains used to be the latency tail of this kernel
+// (catastrophically so for Fp2/G2: 248 serial doublings on one lane).
 template <typename C, typename CFG>
 __global__ void __launch_bounds__(CFG::RED_BLOCK)
 k_weighted_reduce(const g1jT<C> *__restrict__ seg_sum,
@@ -454,12 +520,14 @@ k_weighted_reduce(const g1jT<C> *__restrict__ seg_sum,
     if (live) {
         g1jT<C> ws = seg_wsum[t];
         g1jT<C> ss = seg_sum[t];
-        if (j == 0) {
+        if (j == 0 && !CFG::SIGNED) {
             // weight -1: subtract sum_0
             if (!g1_is_inf9(ss)) g1_neg_y9<C>(ss);
             val = g1_add9(ws, ss);
+        } else if (j == 0) {
+            val = ws;  // signed: weight 0
         } else {
-            uint32_t weight = j * CFG::SEG - 1;  // < 2^C
+            uint32_t weight = j * CFG::SEG + (CFG::SIGNED ? 0 : -1);  // < 2^C
             g1jT<C> acc = g1_inf9<C>();
             for (int b = CFG::C; b >= 0; b--) {
                 acc = g1_dbl9(acc);
@@ -467,10 +535,6 @@ k_weighted_reduce(const g1jT<C> *__restrict__ seg_sum,
             }
             val = g1_add9(ws, acc);
         }
-        // fold the window factor 2^(C*w) HERE: the doubling chains run
-        // SIMD-wide across all threads
-        uint32_t w = t / CFG::NSEG;
-        for (uint32_t d = 0; d < (uint32_t)CFG::C * w; d++) val = g1_dbl9(val);
     }
     if constexpr (CFG::TREE) {
         // one window per block: LDS tree -> one partial per block
@@ -518,37 +582,19 @@ k_window_sum(const g1jT<C> *__restrict__ partials,
     if (t == 0) windows[w] = lds[0];
 }
 
-// ---- final combine + output ----
-// out_mode 0: 64-byte BE affine (infinity -> zeros)
-// out_mode 1: 96-byte BE Jacobian canonical X||Y||Z (Z=0 -> infinity)
+// ---- window emission ----
+// Windows leave the GPU UNSCALED as NWIN Jacobian wire records; the host
+// folds the 2^(C*w) factors with a Horner pass (C doublings + 1 add per
+// window, ~0.2 ms for BN254) at delivery time, fully overlapped with the
+// next pipelined step's GPU work.  This removes the per-thread doubling
+// chains that were the fixed latency tail of the reduction (≥6x scaling
+// target needs the 2^21-shard tail small; G2's chains cost 9.7 ms alone).
 template <typename C, typename CFG>
 __global__ void __launch_bounds__(64)
-k_final_combine(const g1jT<C> *__restrict__ windows,
-                uint8_t *__restrict__ out, int out_mode) {
-    using F = typename C::F;
-    constexpr int NB = F::W64 * 8;
-    // windows[] arrive pre-scaled by 2^(C*w); 64-lane LDS tree for the
-    // window sum (a serial 32-add chain costs ~1 ms on the 14-limb field)
-    __shared__ g1jT<C> lds[64];
-    uint32_t t = threadIdx.x;
-    lds[t] = t < (uint32_t)CFG::NWIN ? windows[t] : g1_inf9<C>();
-    __syncthreads();
-    for (int sh = 32; sh > 0; sh >>= 1) {
-        if (t < (uint32_t)sh) {
-            g1jT<C> o = lds[t + sh];
-            g1jT<C> m = g1_add9(lds[t], o);
-            lds[t] = m;
-        }
-        __syncthreads();
-    }
-    if (t != 0) return;
-    g1jT<C> acc = lds[0];
-    if (out_mode == 0) {
-        g1_to_affine_be9(out, acc);
-    } else {
-        // Jacobian exchange payload (ABI): convert XYZZ -> Jacobian
-        g1_jac_be9(out, acc);
-    }
+k_emit_windows(const g1jT<C> *__restrict__ windows, uint8_t *__restrict__ out) {
+    uint32_t w = threadIdx.x;
+    if (w >= (uint32_t)CFG::NWIN) return;
+    g1_jac_be9<C>(out + (size_t)w * pt_bytes<C>::JAC, windows[w]);
 }
 
 // ---- single-op kernels (zisk-mirror ABI + Jacobian combine) ----

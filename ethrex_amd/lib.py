@@ -29,7 +29,7 @@ for _f in ("device_count", "set_device", "bn254_g1_add", "bn254_g1_mul",
            "msm_plan_create", "msm_plan_destroy", "msm_upload_points",
            "msm_gen_points", "msm_download_points", "msm_upload_scalars",
            "msm_run", "msm_run_partial", "msm_run_async", "msm_sync",
-           "msm_run_partial_async",
+           "msm_run_partial_async", "msm_wait_one", "bn254_g1_combine_cpu",
            "msm_last_times", "msm_combine",
            "ntt_plan_create", "ntt_plan_destroy", "ntt_upload", "ntt_run",
            "ntt_download", "ntt_last_times"):
@@ -100,6 +100,18 @@ def g1_combine(jacobians: bytes, count: int):
     rc = _lib.ethrex_mi355_bn254_g1_combine(_buf(jacobians),
                                             ctypes.c_size_t(count), out)
     return rc, bytes(out)
+
+
+def g1_combine_cpu(jacobians: bytes, count: int) -> bytes:
+    """Host-side combine of the N>1 exchange payload (world-size 96-B
+    Jacobian partials) — boundary glue; keeps GPU streams untouched while
+    pipelined steps are in flight."""
+    out = (ctypes.c_uint8 * 64)()
+    rc = _lib.ethrex_mi355_bn254_g1_combine_cpu(_buf(jacobians),
+                                                ctypes.c_size_t(count), out)
+    if rc != EM_OK:
+        raise HipCoreError(rc, "bn254_g1_combine_cpu")
+    return bytes(out)
 
 
 def gen_fr(seed: int, n: int) -> bytes:
@@ -175,6 +187,15 @@ class MsmPlan:
         outs = [bytes(b) for b in self._pending]
         self._pending = []
         return outs[-1] if outs else b""
+
+    def wait_one(self) -> bytes:
+        """Deliver ONLY the oldest pending pipelined step (later steps keep
+        running on the GPU).  The N>1 loop exchanges step k's partial while
+        the GPU computes step k+1."""
+        if not self._pending:
+            return b""
+        _check(_lib.ethrex_mi355_msm_wait_one(self._p), "msm_wait_one")
+        return bytes(self._pending.pop(0))
 
     def combine(self, jacobians: bytes, count: int) -> bytes:
         """combine Jacobian partials reusing this plan's device buffers"""

@@ -91,9 +91,17 @@ int ethrex_mi355_msm_sync(em_msm_plan *plan);
 /* pipelined shard step: run_async delivering the 96-B Jacobian partial
  * (the multi-GPU exchange payload) instead of the affine result */
 int ethrex_mi355_msm_run_partial_async(em_msm_plan *plan, uint8_t out[96]);
+/* deliver the OLDEST pending pipelined step only (blocks on its compute
+ * chain; later enqueued steps keep running).  The N>1 exchange loop calls
+ * this to AllGather+combine step k while the GPU computes step k+1. */
+int ethrex_mi355_msm_wait_one(em_msm_plan *plan);
 /* combine Jacobian partials on the GPU -> affine result */
 int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96, size_t count,
                                   uint8_t out[64]);
+/* combine the N>1 exchange payload (world-size 96-B Jacobian partials) on
+ * the HOST — boundary glue; avoids touching GPU streams mid-pipeline */
+int ethrex_mi355_bn254_g1_combine_cpu(const uint8_t *jacobians96,
+                                      size_t count, uint8_t out[64]);
 /* per-phase HIP-event timings of the last run, milliseconds:
  * [0]=digits+sort, [1]=bucket accumulation, [2]=bucket reduction,
  * [3]=window combine + affine, [4]=total */

@@ -27,6 +27,10 @@ def _worker(rank, world, port, results_q):
         allparts = allgather_partials(partial, tdist, device="cpu")
         rc, combined = orc.g1_combine_jacobian(allparts, world)
         assert rc == 0
+        # the product path's HOST combine (the pipelined N>1 loop's
+        # per-step combine, bench.py) must agree with the oracle's
+        import ethrex_amd
+        assert ethrex_amd.g1_combine_cpu(allparts, world) == combined
         if rank == 0:
             # unsharded reference: same points, concatenated per-rank scalars
             all_pts, all_scs = b"", b""
