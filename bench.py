@@ -70,6 +70,9 @@ def main():
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--no-ntt", action="store_true")
     ap.add_argument("--no-bls", action="store_true")
+    ap.add_argument("--no-pipeline", action="store_true")
+    ap.add_argument("--pipeline-log2", type=int, default=26,
+                    help="composed wrap-step leg size (BASELINE config 5)")
     ap.add_argument("--check", action="store_true",
                     help="verify the first step's result against the oracle "
                          "via the shard-combine identity (adds oracle time)")
@@ -259,6 +262,65 @@ def main():
 
     phase_ms = {k: round(v, 3) for k, v in plan.last_times().items()}
 
+    # ---- composed wrap-step leg (BASELINE config 5: "full synthetic-
+    # witness block proof (MSM+NTT pipeline) at 2^26 constraints") ----
+    # One wrap step = forward NTT of the 2^26-element witness vector +
+    # the 2^26-point proving MSM over its output, composed ON DEVICE
+    # (msm_scalars_from_ntt; the sp1.rs:122-134 Groth16-wrap flow).  The
+    # MSM shards across ranks (partials AllGathered + host-combined); the
+    # NTT runs replicated per rank (its output feeds each rank's scalar
+    # shard).  The NTT transforms its own output every step, so scalars
+    # differ step to step — nothing inside the timed region is cached.
+    pipeline = None
+    if not args.no_pipeline:
+        pm = 1 << args.pipeline_log2
+        plo, phi = shard_range(pm, n_gpus, rank)
+        nplan2 = ethrex_amd.NttPlan(pm)
+        nplan2.upload(ethrex_amd.gen_fr(49, pm))
+        mplan2 = ethrex_amd.MsmPlan(phi - plo)
+        mplan2.gen_points(plo)
+
+        def wrap_step():
+            nplan2.run(False)
+            mplan2.scalars_from_ntt(nplan2, plo)
+            if world == 1:
+                return mplan2.run()
+            part = mplan2.run_partial()
+            allp = allgather_partials(part, dist, device="cuda")
+            return ethrex_amd.g1_combine_cpu(allp, world)
+
+        for _ in range(max(args.warmup, 1)):
+            wrap_step()
+        if world > 1:
+            dist.barrier()
+            torch.cuda.synchronize()
+        tp = time.perf_counter()
+        for _ in range(args.steps):
+            wrap_step()
+        if world > 1:
+            dist.barrier()
+            torch.cuda.synchronize()
+        dtp = (time.perf_counter() - tp) / args.steps
+        if world > 1:
+            t = torch.tensor([dtp], device="cuda")
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            dtp = float(t.item())
+        pipeline = {
+            "metric": "wrap_pipeline_constraints_per_s",
+            "value": pm / dtp,
+            "n_constraints": pm,
+            "ms_per_step": dtp * 1000.0,
+            "n_gpus": n_gpus,
+            "phase_ms": {
+                "ntt": {k: round(v, 3)
+                        for k, v in nplan2.last_times().items()},
+                "msm": {k: round(v, 3)
+                        for k, v in mplan2.last_times().items()},
+            },
+        }
+        mplan2.destroy()
+        nplan2.destroy()
+
     # ---- BLS12-381 blob-KZG commitment leg (the
     # 4096-point G1 MSM the sequencer computes per blob,
     # crates/common/crypto/kzg.rs:208-230; rank 0, single GPU) ----
@@ -347,7 +409,35 @@ def main():
         }
         kp.destroy()
 
-    cpu_baseline = None
+        # trie root with level-synchronized GPU node hashing (§8f row 4
+        # second half): 2^15 account leaves; the host builds the radix
+        # structure (Python host mirror), every level hashes as one
+        # batched keccak launch
+        import random as _random
+
+        from ethrex_amd import trie as _trie
+        from ethrex_amd import witness as _witness
+        from ethrex_amd.prover import Mi355Backend as _Be
+        _be = _Be()
+        rng = _random.Random(13)
+        tp_pairs = {}
+        for i in range(1 << 15):
+            k = bytes(rng.randrange(256) for _ in range(32))
+            tp_pairs[k] = _trie.account_leaf(
+                i, i * 31, _witness.EMPTY_TRIE_HASH, bytes(32))
+        th0 = time.perf_counter()
+        troot = _trie.trie_root(tp_pairs, _be._gpu_hash_batch)
+        t_total = time.perf_counter() - th0
+        bls["trie_root"] = {
+            "metric": "mpt_root_leaves_per_s",
+            "value": len(tp_pairs) / t_total,
+            "n_leaves": len(tp_pairs),
+            "total_ms": round(t_total * 1000.0, 1),
+            "root": troot.hex(),
+            "note": "host radix build (Python host mirror) + per-level "
+                    "batched GPU keccak; the reference hashes per node "
+                    "on CPU (crates/common/trie)",
+        }
     if rank == 0 and not args.no_cpu_baseline:
         cpu_baseline = cpu_baseline_leg()
 
@@ -379,6 +469,7 @@ def main():
                                else "single GPU",
             },
             "roofline": roofline,
+            "pipeline_2_26": pipeline,
             "ntt": ntt,
             "cpu_baseline": cpu_baseline,
             "bls_blob": bls,

@@ -1,0 +1,379 @@
+// ============================================================================
+// ethrex_mi355 C-ABI implementation — MI355X-native BN254 MSM/NTT core.
+//
+// See include/ethrex_mi355.h for the boundary contract (mirrors the in-repo
+// ZisK accelerator FFI convention, crates/guest-program/src/crypto/zisk.rs:71-137)
+// and DESIGN.md for the kernel design.  Threading: the backend is called
+// from ONE actor on a blocking thread (crates/prover/src/prover.rs:241-251),
+// so plans are not internally locked.
+//
+// NO CPU FALLBACK: every compute entry point requires a visible GPU and
+// returns EM_ERR_HIP otherwise.
+// ============================================================================
+// api_ntt.hip — radix-2 NTT over BN254 Fr (plan + one-shot).
+#include "em_api_common.h"
+#include "../../include/ethrex_mi355.h"
+#include "ntt_kernels.h"
+
+using namespace em;
+
+// ============================ NTT plan ============================
+
+struct em_ntt_plan {
+    size_t n;
+    int logn;
+    bool fused = false;       // four-step path (13 <= logn <= 24)
+    bool fused2 = false;      // two-level four-step (25 <= logn <= 26)
+    int logN1 = 0, logN2 = 0;
+    int logM1 = 0, logM2 = 0; // inner split of N2 (fused2)
+    int cur = 0;              // which buffer holds the data: 0=d_data 1=d_work
+    fe4 *d_data = nullptr;    // packed 4x64 Montgomery (fe4m), 32 B/elem
+    fe4 *d_work = nullptr;    // fused: transpose ping-pong buffer
+    uint8_t *d_bytes = nullptr;
+    fe9 *d_tw = nullptr;      // fallback: forward twiddles, n/2
+    fe9 *d_tw_inv = nullptr;
+    fe4 *d_twfull = nullptr;      // fused: TW2[c][k] = w^(k*c), streamed
+    fe4 *d_twfull_inv = nullptr;
+    fe9 *d_twrow1 = nullptr;      // fused: N1/2 row twiddles (+inv)
+    fe9 *d_twrow1_inv = nullptr;
+    fe9 *d_twrow2 = nullptr;      // fused: N2/2 (+inv)
+    fe9 *d_twrow2_inv = nullptr;
+    fe4 *d_twfull2 = nullptr;     // fused2: inner TW2B[c][k], size N2 (+inv)
+    fe4 *d_twfull2_inv = nullptr;
+    fe9 *d_twrowA = nullptr;      // fused2: inner M1/2 row twiddles (+inv)
+    fe9 *d_twrowA_inv = nullptr;
+    fe9 *d_twrowB = nullptr;      // fused2: inner M2/2 (+inv)
+    fe9 *d_twrowB_inv = nullptr;
+    fe9 *d_ninv = nullptr;        // 1/n (fe9 Montgomery)
+    uint32_t *d_err = nullptr;
+    hipEvent_t ev[4];
+    double last_ms[3] = {0, 0, 0};
+};
+
+// build w2k powers on host and launch k_gen_twiddles: tw[j] = w^j, j < count
+static int gen_tw_table(fe9 *d_out, size_t count, int bits, bool inverse,
+                        int log_size /* transform size 2^log_size */) {
+    fe9 w = fe9_load(inverse ? bn254::FR9_W28_INV : bn254::FR9_W28);
+    for (int k = bn254::FR_TWO_ADICITY; k > log_size; k--)
+        w = mont_sqr9<Fr9T>(w);
+    fe9 w2k[32];
+    w2k[0] = log_size == 0 ? fe9_load(bn254::FR9_ONE) : w;
+    for (int k = 1; k < bits && k < 32; k++) w2k[k] = mont_sqr9<Fr9T>(w2k[k - 1]);
+    fe9 *d_w2k;
+    HIP_TRY(hipMalloc(&d_w2k, sizeof(w2k)));
+    HIP_TRY(hipMemcpy(d_w2k, w2k, sizeof(w2k), hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_gen_twiddles, dim3(blocks_for(count, 256)), dim3(256),
+                       0, 0, d_out, count, d_w2k, bits);
+    HIP_TRY(hipDeviceSynchronize());
+    hipError_t e2 = hipFree(d_w2k);
+    (void)e2;
+    return EM_OK;
+}
+
+// reordered streamed table TW2[c*M + k] = w^(k*c) (fe4m), c < rows:
+// P1's per-row twiddle read becomes a coalesced stream instead of a
+// stride-c gather over an n-sized table (a full cache line per element).
+static int gen_tw2_table(fe4 *d_out, uint32_t M, size_t rows, bool inverse,
+                         int log_size /* transform size 2^log_size */) {
+    fe9 w = fe9_load(inverse ? bn254::FR9_W28_INV : bn254::FR9_W28);
+    for (int k = bn254::FR_TWO_ADICITY; k > log_size; k--)
+        w = mont_sqr9<Fr9T>(w);
+    fe9 w2k[32];
+    w2k[0] = log_size == 0 ? fe9_load(bn254::FR9_ONE) : w;
+    for (int k = 1; k < log_size && k < 32; k++)
+        w2k[k] = mont_sqr9<Fr9T>(w2k[k - 1]);
+    fe9 *d_w2k;
+    HIP_TRY(hipMalloc(&d_w2k, sizeof(w2k)));
+    HIP_TRY(hipMemcpy(d_w2k, w2k, sizeof(w2k), hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_gen_tw2, dim3(blocks_for((size_t)M * rows, 256)),
+                       dim3(256), 0, 0, d_out, M, rows, d_w2k, log_size);
+    HIP_TRY(hipDeviceSynchronize());
+    (void)hipFree(d_w2k);
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
+    if (!plan || n == 0 || (n & (n - 1))) return EM_ERR_INPUT;
+    int logn = 0;
+    while (((size_t)1 << logn) < n) logn++;
+    if (logn > bn254::FR_TWO_ADICITY) return EM_ERR_INPUT;
+    int rc = require_gpu();
+    if (rc) return rc;
+    em_ntt_plan *p = new em_ntt_plan();
+    p->n = n;
+    p->logn = logn;
+    p->fused = (logn > 12 && logn <= 24);
+    p->fused2 = (logn >= 25 && logn <= 26);
+    size_t half = n > 1 ? n / 2 : 1;
+    hipError_t e = hipSuccess;
+    auto mal = [&](void **ptr, size_t bytes) {
+        if (e == hipSuccess) e = hipMalloc(ptr, bytes);
+    };
+    mal((void **)&p->d_data, n * sizeof(fe4));
+    mal((void **)&p->d_bytes, n * 32);
+    mal((void **)&p->d_err, 4);
+    if (p->fused || p->fused2) {
+        if (p->fused) {
+            p->logN1 = (logn + 1) / 2;
+            p->logN2 = logn / 2;
+        } else {
+            // two-level: outer P1 rows of 2^12; inner four-step over the
+            // 2^13/2^14-long rows (M1 x M2, both <= 2^7)
+            p->logN1 = 12;
+            p->logN2 = logn - 12;
+            p->logM1 = (p->logN2 + 1) / 2;
+            p->logM2 = p->logN2 / 2;
+        }
+        mal((void **)&p->d_work, n * sizeof(fe4));
+        mal((void **)&p->d_twfull, n * sizeof(fe4));
+        mal((void **)&p->d_twfull_inv, n * sizeof(fe4));
+        mal((void **)&p->d_twrow1, ((size_t)1 << (p->logN1 - 1)) * sizeof(fe9));
+        mal((void **)&p->d_twrow1_inv, ((size_t)1 << (p->logN1 - 1)) * sizeof(fe9));
+        mal((void **)&p->d_twrow2,
+            ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe9));
+        mal((void **)&p->d_twrow2_inv,
+            ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe9));
+        mal((void **)&p->d_ninv, sizeof(fe9));
+        if (p->fused2) {
+            mal((void **)&p->d_twfull2, ((size_t)1 << p->logN2) * sizeof(fe4));
+            mal((void **)&p->d_twfull2_inv,
+                ((size_t)1 << p->logN2) * sizeof(fe4));
+            mal((void **)&p->d_twrowA,
+                ((size_t)1 << (p->logM1 - 1)) * sizeof(fe9));
+            mal((void **)&p->d_twrowA_inv,
+                ((size_t)1 << (p->logM1 - 1)) * sizeof(fe9));
+            mal((void **)&p->d_twrowB,
+                ((size_t)1 << (p->logM2 - 1)) * sizeof(fe9));
+            mal((void **)&p->d_twrowB_inv,
+                ((size_t)1 << (p->logM2 - 1)) * sizeof(fe9));
+        }
+    } else {
+        mal((void **)&p->d_tw, half * sizeof(fe9));
+        mal((void **)&p->d_tw_inv, half * sizeof(fe9));
+    }
+    for (int i = 0; i < 4 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
+    if (e != hipSuccess) {
+        ethrex_mi355_ntt_plan_destroy(p);
+        return hip_fail(e, "ntt_plan_create");
+    }
+    if (p->fused || p->fused2) {
+        int rc2;
+        uint32_t N1g = 1u << p->logN1;
+        if ((rc2 = gen_tw2_table(p->d_twfull, N1g, n / N1g, false, logn)))
+            return rc2;
+        if ((rc2 = gen_tw2_table(p->d_twfull_inv, N1g, n / N1g, true, logn)))
+            return rc2;
+        if ((rc2 = gen_tw_table(p->d_twrow1, (size_t)1 << (p->logN1 - 1),
+                                p->logN1, false, p->logN1))) return rc2;
+        if ((rc2 = gen_tw_table(p->d_twrow1_inv, (size_t)1 << (p->logN1 - 1),
+                                p->logN1, true, p->logN1))) return rc2;
+        if ((rc2 = gen_tw_table(p->d_twrow2, (size_t)1 << (p->logN2 - 1),
+                                p->logN2, false, p->logN2))) return rc2;
+        if ((rc2 = gen_tw_table(p->d_twrow2_inv, (size_t)1 << (p->logN2 - 1),
+                                p->logN2, true, p->logN2))) return rc2;
+        if (p->fused2) {
+            size_t n2 = (size_t)1 << p->logN2;
+            uint32_t M1g = 1u << p->logM1;
+            if ((rc2 = gen_tw2_table(p->d_twfull2, M1g, n2 / M1g, false,
+                                     p->logN2))) return rc2;
+            if ((rc2 = gen_tw2_table(p->d_twfull2_inv, M1g, n2 / M1g, true,
+                                     p->logN2))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twrowA, (size_t)1 << (p->logM1 - 1),
+                                    p->logM1, false, p->logM1))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twrowA_inv,
+                                    (size_t)1 << (p->logM1 - 1), p->logM1,
+                                    true, p->logM1))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twrowB, (size_t)1 << (p->logM2 - 1),
+                                    p->logM2, false, p->logM2))) return rc2;
+            if ((rc2 = gen_tw_table(p->d_twrowB_inv,
+                                    (size_t)1 << (p->logM2 - 1), p->logM2,
+                                    true, p->logM2))) return rc2;
+        }
+        fe9 ninv = fe9_load(bn254::FR9_INV_POW2[logn]);
+        HIP_TRY(hipMemcpy(p->d_ninv, &ninv, sizeof(fe9), hipMemcpyHostToDevice));
+    } else {
+        int rc2;
+        if ((rc2 = gen_tw_table(p->d_tw, half, logn ? logn : 1, false, logn)))
+            return rc2;
+        if ((rc2 = gen_tw_table(p->d_tw_inv, half, logn ? logn : 1, true, logn)))
+            return rc2;
+    }
+    *plan = p;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_plan_destroy(em_ntt_plan *p) {
+    if (!p) return EM_ERR_INPUT;
+    (void)hipFree(p->d_data);
+    (void)hipFree(p->d_work);
+    (void)hipFree(p->d_bytes);
+    (void)hipFree(p->d_tw);
+    (void)hipFree(p->d_tw_inv);
+    (void)hipFree(p->d_twfull);
+    (void)hipFree(p->d_twfull2);
+    (void)hipFree(p->d_twfull2_inv);
+    (void)hipFree(p->d_twrowA);
+    (void)hipFree(p->d_twrowA_inv);
+    (void)hipFree(p->d_twrowB);
+    (void)hipFree(p->d_twrowB_inv);
+    (void)hipFree(p->d_twfull_inv);
+    (void)hipFree(p->d_twrow1);
+    (void)hipFree(p->d_twrow1_inv);
+    (void)hipFree(p->d_twrow2);
+    (void)hipFree(p->d_twrow2_inv);
+    (void)hipFree(p->d_ninv);
+    (void)hipFree(p->d_err);
+    delete p;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_upload(em_ntt_plan *p, const uint8_t *elems32) {
+    if (!p || !elems32) return EM_ERR_INPUT;
+    HIP_TRY(hipMemset(p->d_err, 0, 4));
+    HIP_TRY(hipMemcpy(p->d_bytes, elems32, p->n * 32, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_fr_from_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       p->d_bytes, p->d_data, p->n, p->d_err);
+    p->cur = 0;
+    uint32_t err;
+    HIP_TRY(hipMemcpy(&err, p->d_err, 4, hipMemcpyDeviceToHost));
+    return err ? EM_ERR_INPUT : EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
+    if (!p) return EM_ERR_INPUT;
+    size_t n = p->n;
+    fe4 *cur = p->cur ? p->d_work : p->d_data;
+    fe4 *oth = p->cur ? p->d_data : p->d_work;
+    HIP_TRY(hipEventRecord(p->ev[0], 0));
+    if (p->fused) {
+        uint32_t N1 = 1u << p->logN1, N2 = 1u << p->logN2;
+        // T0: A[r][c] -> A1[c][r]
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256), 0,
+                           0, cur, oth, N1, N2);
+        HIP_TRY(hipEventRecord(p->ev[1], 0));
+        // P1: row NTT_N1 over each of the N2 rows + w^(k1*c) twiddle
+        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512), 0, 0, oth,
+                           p->logN1, inverse ? p->d_twrow1_inv : p->d_twrow1,
+                           inverse ? p->d_twfull_inv : p->d_twfull,
+                           (const fe9 *)nullptr);
+        // T1
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256), 0,
+                           0, oth, cur, N2, N1);
+        // P2: row NTT_N2 (+ 1/n scale on iNTT)
+        hipLaunchKernelGGL(k_ntt_row, dim3(N1), dim3(512), 0, 0, cur,
+                           p->logN2, inverse ? p->d_twrow2_inv : p->d_twrow2,
+                           (const fe4 *)nullptr,
+                           inverse ? p->d_ninv : (const fe9 *)nullptr);
+        // T2: natural order
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256), 0,
+                           0, cur, oth, N1, N2);
+        p->cur ^= 1;
+    } else if (p->fused2) {
+        // two-level four-step (25 <= logn <= 26): the outer P2 row length
+        // (2^13/2^14) exceeds the 4096-element LDS row kernel, so each of
+        // the N1 outer rows is itself transformed by a BATCHED inner
+        // four-step (M1 x M2, both <= 2^7).  The inner transform is
+        // natural-in / natural-out, so it composes exactly where the
+        // one-level P2 sat.  8 full-array passes instead of logn.
+        uint32_t N1 = 1u << p->logN1, N2 = 1u << p->logN2;
+        uint32_t M1 = 1u << p->logM1, M2 = 1u << p->logM2;
+        const fe9 *tr1 = inverse ? p->d_twrow1_inv : p->d_twrow1;
+        const fe4 *tf = inverse ? p->d_twfull_inv : p->d_twfull;
+        const fe4 *tf2 = inverse ? p->d_twfull2_inv : p->d_twfull2;
+        const fe9 *trA = inverse ? p->d_twrowA_inv : p->d_twrowA;
+        const fe9 *trB = inverse ? p->d_twrowB_inv : p->d_twrowB;
+        // T0 + P1 + T1: outer column NTTs (length N1) + w_n^(k1 c)
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256),
+                           0, 0, cur, oth, N1, N2);
+        HIP_TRY(hipEventRecord(p->ev[1], 0));
+        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512), 0, 0, oth,
+                           p->logN1, tr1, tf, (const fe9 *)nullptr);
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256),
+                           0, 0, oth, cur, N2, N1);
+        // inner batched four-step over the N1 rows of length N2 = M1*M2
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(M2 / 32, M1 / 32, N1),
+                           dim3(256), 0, 0, cur, oth, M1, M2);
+        hipLaunchKernelGGL(k_ntt_row_small, dim3(N1 * M2 * M1 / 1024),
+                           dim3(512), 0, 0, oth, p->logM1, trA, tf2,
+                           (const fe9 *)nullptr, M2 - 1);
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(M1 / 32, M2 / 32, N1),
+                           dim3(256), 0, 0, oth, cur, M2, M1);
+        hipLaunchKernelGGL(k_ntt_row_small, dim3(N1 * M1 * M2 / 1024),
+                           dim3(512), 0, 0, cur, p->logM2, trB,
+                           (const fe4 *)nullptr,
+                           inverse ? p->d_ninv : (const fe9 *)nullptr,
+                           0xffffffffu);
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(M2 / 32, M1 / 32, N1),
+                           dim3(256), 0, 0, cur, oth, M1, M2);
+        // outer T2 -> natural order
+        hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256),
+                           0, 0, oth, cur, N1, N2);
+        // data ends in `cur` (8 passes): no buffer flip
+    } else {
+        if (n > 1) {
+            hipLaunchKernelGGL(k_bit_reverse, dim3(blocks_for(n, 256)), dim3(256),
+                               0, 0, cur, n, p->logn);
+        }
+        HIP_TRY(hipEventRecord(p->ev[1], 0));
+        const fe9 *tw = inverse ? p->d_tw_inv : p->d_tw;
+        for (int s = 1; s <= p->logn; s++) {
+            hipLaunchKernelGGL(k_ntt_stage, dim3(blocks_for(n / 2, 256)),
+                               dim3(256), 0, 0, cur, tw, n, p->logn, s);
+        }
+        if (inverse) {
+            hipLaunchKernelGGL(k_ntt_scale, dim3(blocks_for(n, 256)), dim3(256),
+                               0, 0, cur, n, p->logn);
+        }
+    }
+    HIP_TRY(hipEventRecord(p->ev[2], 0));
+    HIP_TRY(hipDeviceSynchronize());
+    float ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[1]));
+    p->last_ms[0] = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[1], p->ev[2]));
+    p->last_ms[1] = ms;
+    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[2]));
+    p->last_ms[2] = ms;
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_download(em_ntt_plan *p, uint8_t *elems32) {
+    if (!p || !elems32) return EM_ERR_INPUT;
+    fe4 *cur = p->cur ? p->d_work : p->d_data;
+    hipLaunchKernelGGL(k_fr_to_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
+                       cur, p->d_bytes, p->n);
+    HIP_TRY(hipMemcpy(elems32, p->d_bytes, p->n * 32, hipMemcpyDeviceToHost));
+    return EM_OK;
+}
+
+extern "C" int ethrex_mi355_ntt_last_times(em_ntt_plan *p, double times_ms[3]) {
+    if (!p || !times_ms) return EM_ERR_INPUT;
+    memcpy(times_ms, p->last_ms, sizeof p->last_ms);
+    return EM_OK;
+}
+
+// ============================ one-shot NTT ============================
+
+extern "C" int ethrex_mi355_bn254_fr_ntt(uint8_t *elems32, size_t n, int inverse) {
+    if (!elems32 || n == 0) return EM_ERR_INPUT;
+    em_ntt_plan *p = nullptr;
+    int rc = ethrex_mi355_ntt_plan_create(n, &p);
+    if (rc) return rc;
+    rc = ethrex_mi355_ntt_upload(p, elems32);
+    if (!rc) rc = ethrex_mi355_ntt_run(p, inverse);
+    if (!rc) rc = ethrex_mi355_ntt_download(p, elems32);
+    ethrex_mi355_ntt_plan_destroy(p);
+    return rc;
+}
+
+// device pointer to the CURRENT data buffer (packed fe4m Montgomery) — the
+// wrap-pipeline handoff: the MSM plan converts these in place on device
+// (ethrex_mi355_msm_scalars_from_ntt), no PCIe round trip (sp1.rs:122-134
+// wrap flow: the witness NTT output feeds the proving MSM).
+extern "C" int ethrex_mi355_ntt_device_data(em_ntt_plan *p, const void **ptr,
+                                            size_t *n) {
+    if (!p || !ptr) return EM_ERR_INPUT;
+    *ptr = p->cur ? p->d_work : p->d_data;
+    if (n) *n = p->n;
+    return EM_OK;
+}

@@ -10,13 +10,11 @@
 // NO CPU FALLBACK: every compute entry point requires a visible GPU and
 // returns EM_ERR_HIP otherwise.
 // ============================================================================
-#include <hip/hip_runtime.h>
-
-#include <cstdio>
-#include <cstdlib>
-#include <cstring>
-#include <string>
-#include <type_traits>
+// msm_api_impl.h — the curve-templated MSM plan machinery.  Included by
+// one TU per curve (api_msm_bn254/bls/g2.hip); every function is static so
+// each TU instantiates only its own curve.
+#pragma once
+#include "em_api_common.h"  // before rocprim: it needs <cstring> in scope
 
 #include <rocprim/rocprim.hpp>
 
@@ -24,56 +22,8 @@
 #include "gpu_field.h"
 #include "gpu_g1_9.h"
 #include "msm_kernels.h"
-#include "ntt_kernels.h"
-#include "keccak_kernels.h"
 
 using namespace em;
-
-static thread_local std::string g_last_err;
-
-static int hip_fail(hipError_t e, const char *where) {
-    g_last_err = std::string(where) + ": " + hipGetErrorString(e);
-    return EM_ERR_HIP;
-}
-
-#define HIP_TRY(call)                                    \
-    do {                                                 \
-        hipError_t _e = (call);                          \
-        if (_e != hipSuccess) return hip_fail(_e, #call); \
-    } while (0)
-
-extern "C" const char *ethrex_mi355_version(void) { return "0.1.0-gfx950"; }
-
-extern "C" const char *ethrex_mi355_last_error(void) { return g_last_err.c_str(); }
-
-extern "C" int ethrex_mi355_device_count(int *count) {
-    if (!count) return EM_ERR_INPUT;
-    hipError_t e = hipGetDeviceCount(count);
-    if (e != hipSuccess) {
-        *count = 0;
-        return hip_fail(e, "hipGetDeviceCount");
-    }
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_set_device(int device) {
-    HIP_TRY(hipSetDevice(device));
-    return EM_OK;
-}
-
-static int require_gpu() {
-    int n = 0;
-    hipError_t e = hipGetDeviceCount(&n);
-    if (e != hipSuccess || n == 0) {
-        g_last_err = "no HIP device visible (MI355X required; no CPU fallback)";
-        return EM_ERR_HIP;
-    }
-    return EM_OK;
-}
-
-static inline uint32_t blocks_for(size_t n, int bs) {
-    return (uint32_t)((n + bs - 1) / bs);
-}
 
 // ============================ MSM plans ============================
 // One templated plan implementation serves both curves:
@@ -831,309 +781,6 @@ struct em_msm_plan : msm_plan_t<Bn254G1> {};
 struct em_bls_msm_plan : msm_plan_t<BlsG1> {};
 struct em_bls_g2_msm_plan : msm_plan_t<BlsG2> {};
 
-extern "C" int ethrex_mi355_msm_plan_create(size_t n, em_msm_plan **plan) {
-    return msm_create_t(n, (msm_plan_t<Bn254G1> **)plan);
-}
-extern "C" int ethrex_mi355_msm_plan_destroy(em_msm_plan *p) {
-    return msm_destroy_t((msm_plan_t<Bn254G1> *)p);
-}
-extern "C" int ethrex_mi355_msm_upload_points(em_msm_plan *p,
-                                              const uint8_t *points64) {
-    return msm_upload_points_t((msm_plan_t<Bn254G1> *)p, points64);
-}
-extern "C" int ethrex_mi355_msm_gen_points(em_msm_plan *p, uint64_t start) {
-    return msm_gen_points_t((msm_plan_t<Bn254G1> *)p, start);
-}
-extern "C" int ethrex_mi355_msm_download_points(em_msm_plan *p, uint8_t *out64) {
-    return msm_download_points_t((msm_plan_t<Bn254G1> *)p, out64);
-}
-extern "C" int ethrex_mi355_msm_upload_scalars(em_msm_plan *p,
-                                               const uint8_t *scalars32) {
-    return msm_upload_scalars_t((msm_plan_t<Bn254G1> *)p, scalars32);
-}
-extern "C" int ethrex_mi355_msm_run(em_msm_plan *p, uint8_t out[64]) {
-    return msm_run_inner_t((msm_plan_t<Bn254G1> *)p, out, 0);
-}
-extern "C" int ethrex_mi355_msm_run_async(em_msm_plan *p, uint8_t out[64]) {
-    return msm_run_async_t((msm_plan_t<Bn254G1> *)p, out);
-}
-extern "C" int ethrex_mi355_msm_sync(em_msm_plan *p) {
-    return msm_sync_t((msm_plan_t<Bn254G1> *)p);
-}
-/* deliver the OLDEST pending pipelined step without draining the pipeline
- * (N>1 exchange overlap: AllGather step k while the GPU computes k+1) */
-extern "C" int ethrex_mi355_msm_wait_one(em_msm_plan *p) {
-    return msm_wait_one_t((msm_plan_t<Bn254G1> *)p);
-}
-/* pipelined shard step: like run_async but delivers the 96-B Jacobian
- * partial (multi-GPU: the NEXT step's sort chain overlaps this step's
- * compute even across the AllGather + combine exchange). */
-extern "C" int ethrex_mi355_msm_run_partial_async(em_msm_plan *p,
-                                                  uint8_t out[96]) {
-    return msm_run_async_t((msm_plan_t<Bn254G1> *)p, out, 1);
-}
-extern "C" int ethrex_mi355_msm_run_partial(em_msm_plan *p, uint8_t out[96]) {
-    return msm_run_inner_t((msm_plan_t<Bn254G1> *)p, out, 1);
-}
-extern "C" int ethrex_mi355_msm_last_times(em_msm_plan *p, double times_ms[5]) {
-    if (!p || !times_ms) return EM_ERR_INPUT;
-    memcpy(times_ms, p->last_ms, sizeof p->last_ms);
-    return EM_OK;
-}
-
-// combine Jacobian partials reusing the plan's buffers (no per-call
-// hipMalloc: the N>1 exchange runs this every step)
-extern "C" int ethrex_mi355_msm_combine(em_msm_plan *p,
-                                        const uint8_t *jacobians96,
-                                        size_t count, uint8_t out[64]) {
-    if (!p || !jacobians96 || !out || count == 0 || count * 96 > p->n * 64)
-        return EM_ERR_INPUT;
-    HIP_TRY(hipMemcpy(p->d_scratch, jacobians96, 96 * count,
-                      hipMemcpyHostToDevice));
-    hipLaunchKernelGGL((k_g1_combine<Bn254G1>), dim3(1), dim3(64), 0, 0,
-                       p->d_scratch, count, p->d_out);
-    HIP_TRY(hipMemcpy(out, p->d_out, 64, hipMemcpyDeviceToHost));
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_bls_msm_plan_create(size_t n, em_bls_msm_plan **plan) {
-    return msm_create_t(n, (msm_plan_t<BlsG1> **)plan);
-}
-extern "C" int ethrex_mi355_bls_msm_plan_destroy(em_bls_msm_plan *p) {
-    return msm_destroy_t((msm_plan_t<BlsG1> *)p);
-}
-extern "C" int ethrex_mi355_bls_msm_upload_points(em_bls_msm_plan *p,
-                                                  const uint8_t *points96) {
-    return msm_upload_points_t((msm_plan_t<BlsG1> *)p, points96);
-}
-extern "C" int ethrex_mi355_bls_msm_gen_points(em_bls_msm_plan *p,
-                                               uint64_t start) {
-    return msm_gen_points_t((msm_plan_t<BlsG1> *)p, start);
-}
-extern "C" int ethrex_mi355_bls_msm_download_points(em_bls_msm_plan *p,
-                                                    uint8_t *out96) {
-    return msm_download_points_t((msm_plan_t<BlsG1> *)p, out96);
-}
-extern "C" int ethrex_mi355_bls_msm_upload_scalars(em_bls_msm_plan *p,
-                                                   const uint8_t *scalars32) {
-    return msm_upload_scalars_t((msm_plan_t<BlsG1> *)p, scalars32);
-}
-extern "C" int ethrex_mi355_bls_msm_run(em_bls_msm_plan *p, uint8_t out[96]) {
-    return msm_run_inner_t((msm_plan_t<BlsG1> *)p, out, 0);
-}
-extern "C" int ethrex_mi355_bls_msm_run_async(em_bls_msm_plan *p,
-                                              uint8_t out[96]) {
-    return msm_run_async_t((msm_plan_t<BlsG1> *)p, out);
-}
-extern "C" int ethrex_mi355_bls_msm_sync(em_bls_msm_plan *p) {
-    return msm_sync_t((msm_plan_t<BlsG1> *)p);
-}
-extern "C" int ethrex_mi355_bls_msm_run_partial(em_bls_msm_plan *p,
-                                                uint8_t out[144]) {
-    return msm_run_inner_t((msm_plan_t<BlsG1> *)p, out, 1);
-}
-extern "C" int ethrex_mi355_bls_msm_precompute(em_bls_msm_plan *p) {
-    return msm_precompute_t((msm_plan_t<BlsG1> *)p);
-}
-extern "C" int ethrex_mi355_bls_msm_last_times(em_bls_msm_plan *p,
-                                               double times_ms[5]) {
-    if (!p || !times_ms) return EM_ERR_INPUT;
-    memcpy(times_ms, p->last_ms, sizeof p->last_ms);
-    return EM_OK;
-}
-
-// ============================ one-shot MSMs ============================
-
-extern "C" int ethrex_mi355_bn254_g1_msm(const uint8_t *points64,
-                                         const uint8_t *scalars32, size_t n,
-                                         uint8_t out[64]) {
-    if (!points64 || !scalars32 || !out || n == 0) return EM_ERR_INPUT;
-    em_msm_plan *p = nullptr;
-    int rc = ethrex_mi355_msm_plan_create(n, &p);
-    if (rc) return rc;
-    rc = ethrex_mi355_msm_upload_points(p, points64);
-    if (!rc) rc = ethrex_mi355_msm_upload_scalars(p, scalars32);
-    if (!rc) rc = ethrex_mi355_msm_run(p, out);
-    ethrex_mi355_msm_plan_destroy(p);
-    return rc;
-}
-
-extern "C" int ethrex_mi355_bls12381_g1_msm(const uint8_t *points96,
-                                            const uint8_t *scalars32, size_t n,
-                                            uint8_t out[96]) {
-    if (!points96 || !scalars32 || !out || n == 0) return EM_ERR_INPUT;
-    em_bls_msm_plan *p = nullptr;
-    int rc = ethrex_mi355_bls_msm_plan_create(n, &p);
-    if (rc) return rc;
-    rc = ethrex_mi355_bls_msm_upload_points(p, points96);
-    if (!rc) rc = ethrex_mi355_bls_msm_upload_scalars(p, scalars32);
-    if (!rc) rc = ethrex_mi355_bls_msm_run(p, out);
-    ethrex_mi355_bls_msm_plan_destroy(p);
-    return rc;
-}
-
-// ---- BLS12-381 G2 plan + one-shot ABI (EIP-2537 192-byte points) ----
-extern "C" int ethrex_mi355_bls_g2_msm_plan_create(size_t n,
-                                                   em_bls_g2_msm_plan **plan) {
-    return msm_create_t(n, (msm_plan_t<BlsG2> **)plan);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_plan_destroy(em_bls_g2_msm_plan *p) {
-    return msm_destroy_t((msm_plan_t<BlsG2> *)p);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_upload_points(em_bls_g2_msm_plan *p,
-                                                     const uint8_t *pts192) {
-    return msm_upload_points_t((msm_plan_t<BlsG2> *)p, pts192);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_gen_points(em_bls_g2_msm_plan *p,
-                                                  uint64_t start) {
-    return msm_gen_points_t((msm_plan_t<BlsG2> *)p, start);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_download_points(em_bls_g2_msm_plan *p,
-                                                       uint8_t *out192) {
-    return msm_download_points_t((msm_plan_t<BlsG2> *)p, out192);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_upload_scalars(em_bls_g2_msm_plan *p,
-                                                      const uint8_t *s32) {
-    return msm_upload_scalars_t((msm_plan_t<BlsG2> *)p, s32);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_run(em_bls_g2_msm_plan *p,
-                                           uint8_t out[192]) {
-    return msm_run_inner_t((msm_plan_t<BlsG2> *)p, out, 0);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_run_async(em_bls_g2_msm_plan *p,
-                                                 uint8_t out[192]) {
-    return msm_run_async_t((msm_plan_t<BlsG2> *)p, out);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_sync(em_bls_g2_msm_plan *p) {
-    return msm_sync_t((msm_plan_t<BlsG2> *)p);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_run_partial(em_bls_g2_msm_plan *p,
-                                                   uint8_t out[288]) {
-    return msm_run_inner_t((msm_plan_t<BlsG2> *)p, out, 1);
-}
-extern "C" int ethrex_mi355_bls_g2_msm_last_times(em_bls_g2_msm_plan *p,
-                                                  double times_ms[5]) {
-    if (!p) return EM_ERR_INPUT;
-    for (int i = 0; i < 5; i++) times_ms[i] = ((msm_plan_t<BlsG2> *)p)->last_ms[i];
-    return EM_OK;
-}
-extern "C" int ethrex_mi355_bls12381_g2_msm(const uint8_t *points192,
-                                            const uint8_t *scalars32, size_t n,
-                                            uint8_t out[192]) {
-    if (!points192 || !scalars32 || !out || n == 0) return EM_ERR_INPUT;
-    em_bls_g2_msm_plan *p = nullptr;
-    int rc = ethrex_mi355_bls_g2_msm_plan_create(n, &p);
-    if (rc) return rc;
-    rc = ethrex_mi355_bls_g2_msm_upload_points(p, points192);
-    if (!rc) rc = ethrex_mi355_bls_g2_msm_upload_scalars(p, scalars32);
-    if (!rc) rc = ethrex_mi355_bls_g2_msm_run(p, out);
-    ethrex_mi355_bls_g2_msm_plan_destroy(p);
-    return rc;
-}
-
-// ======================= batched keccak-256 =======================
-// witness/statement hashing (the reference's
-// crates/common/crypto/keccak asm path, called per-node in trie hashing).
-
-struct em_keccak_plan {
-    size_t max_bytes = 0, max_n = 0, n = 0;
-    uint8_t *d_msgs = nullptr;
-    uint64_t *d_offs = nullptr;
-    uint8_t *d_out = nullptr;
-    hipEvent_t ev[2];
-    double last_ms = 0;
-};
-
-extern "C" int ethrex_mi355_keccak_plan_create(size_t max_bytes, size_t max_n,
-                                               em_keccak_plan **plan) {
-    if (!plan || max_n == 0) return EM_ERR_INPUT;
-    int rc = require_gpu();
-    if (rc) return rc;
-    auto *p = new em_keccak_plan();
-    p->max_bytes = max_bytes;
-    p->max_n = max_n;
-    hipError_t e = hipMalloc((void **)&p->d_msgs, max_bytes ? max_bytes : 1);
-    if (e == hipSuccess) e = hipMalloc((void **)&p->d_offs, (max_n + 1) * 8);
-    if (e == hipSuccess) e = hipMalloc((void **)&p->d_out, max_n * 32);
-    for (int i = 0; i < 2 && e == hipSuccess; i++)
-        e = hipEventCreate(&p->ev[i]);
-    if (e != hipSuccess) {
-        (void)hipFree(p->d_msgs);
-        (void)hipFree(p->d_offs);
-        (void)hipFree(p->d_out);
-        delete p;
-        return hip_fail(e, "keccak_plan_create");
-    }
-    *plan = p;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_keccak_plan_destroy(em_keccak_plan *p) {
-    if (!p) return EM_ERR_INPUT;
-    (void)hipFree(p->d_msgs);
-    (void)hipFree(p->d_offs);
-    (void)hipFree(p->d_out);
-    delete p;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_keccak_upload(em_keccak_plan *p,
-                                          const uint8_t *msgs,
-                                          const uint64_t *offsets, size_t n) {
-    if (!p || !offsets || n == 0 || n > p->max_n) return EM_ERR_INPUT;
-    if (offsets[n] > p->max_bytes) return EM_ERR_INPUT;
-    if (offsets[n] > 0 && !msgs) return EM_ERR_INPUT;
-    if (offsets[n] > 0)
-        HIP_TRY(hipMemcpy(p->d_msgs, msgs, offsets[n],
-                          hipMemcpyHostToDevice));
-    HIP_TRY(hipMemcpy(p->d_offs, offsets, (n + 1) * 8,
-                      hipMemcpyHostToDevice));
-    p->n = n;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_keccak_run(em_keccak_plan *p) {
-    if (!p || p->n == 0) return EM_ERR_INPUT;
-    HIP_TRY(hipEventRecord(p->ev[0], 0));
-    hipLaunchKernelGGL(k_keccak256_batch, dim3(blocks_for(p->n, 256)),
-                       dim3(256), 0, 0, p->d_msgs, p->d_offs, p->n, p->d_out);
-    HIP_TRY(hipEventRecord(p->ev[1], 0));
-    HIP_TRY(hipDeviceSynchronize());
-    float ms;
-    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[1]));
-    p->last_ms = ms;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_keccak_download(em_keccak_plan *p, uint8_t *out) {
-    if (!p || !out || p->n == 0) return EM_ERR_INPUT;
-    HIP_TRY(hipMemcpy(out, p->d_out, p->n * 32, hipMemcpyDeviceToHost));
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_keccak_last_ms(em_keccak_plan *p, double *ms) {
-    if (!p || !ms) return EM_ERR_INPUT;
-    *ms = p->last_ms;
-    return EM_OK;
-}
-
-/* one-shot convenience (PCIe-inclusive) */
-extern "C" int ethrex_mi355_keccak256_batch(const uint8_t *msgs,
-                                            const uint64_t *offsets, size_t n,
-                                            uint8_t *out32) {
-    if (!offsets || !out32 || n == 0) return EM_ERR_INPUT;
-    em_keccak_plan *p = nullptr;
-    int rc = ethrex_mi355_keccak_plan_create(offsets[n], n, &p);
-    if (rc) return rc;
-    rc = ethrex_mi355_keccak_upload(p, msgs, offsets, n);
-    if (!rc) rc = ethrex_mi355_keccak_run(p);
-    if (!rc) rc = ethrex_mi355_keccak_download(p, out32);
-    ethrex_mi355_keccak_plan_destroy(p);
-    return rc;
-}
-
-// ============================ single ops ============================
-
 // generic 1-thread op runner: in_bytes staged, out_bytes copied back
 template <typename K>
 static int run_single(K kern, const uint8_t *a, size_t la, const uint8_t *b,
@@ -1157,489 +804,4 @@ static int run_single(K kern, const uint8_t *a, size_t la, const uint8_t *b,
     (void)hipFree(d_err);
     if (err & 2u) return EM_ERR_INPUT;
     return err ? EM_ERR_POINT : EM_OK;
-}
-
-extern "C" int ethrex_mi355_bn254_g1_add(const uint8_t p1[64],
-                                         const uint8_t p2[64], uint8_t out[64]) {
-    if (!p1 || !p2 || !out) return EM_ERR_INPUT;
-    return run_single(k_g1_add_single, p1, 64, p2, 64, out, 64);
-}
-
-extern "C" int ethrex_mi355_bn254_g1_mul(const uint8_t point[64],
-                                         const uint8_t scalar[32],
-                                         uint8_t out[64]) {
-    if (!point || !scalar || !out) return EM_ERR_INPUT;
-    return run_single(k_g1_mul_single, point, 64, scalar, 32, out, 64);
-}
-
-extern "C" int ethrex_mi355_bls12381_g1_add(const uint8_t p1[96],
-                                            const uint8_t p2[96],
-                                            uint8_t out[96]) {
-    if (!p1 || !p2 || !out) return EM_ERR_INPUT;
-    return run_single(k_bls_g1_add_single, p1, 96, p2, 96, out, 96);
-}
-
-extern "C" int ethrex_mi355_bls12381_g1_mul(const uint8_t point[96],
-                                            const uint8_t scalar[32],
-                                            uint8_t out[96]) {
-    if (!point || !scalar || !out) return EM_ERR_INPUT;
-    return run_single(k_bls_g1_mul_single, point, 96, scalar, 32, out, 96);
-}
-
-extern "C" int ethrex_mi355_bls12381_g2_add(const uint8_t p1[192],
-                                            const uint8_t p2[192],
-                                            uint8_t out[192]) {
-    return run_single(k_bls_g2_add_single, p1, 192, p2, 192, out, 192);
-}
-
-extern "C" int ethrex_mi355_bls12381_g2_mul(const uint8_t point[192],
-                                            const uint8_t scalar[32],
-                                            uint8_t out[192]) {
-    return run_single(k_bls_g2_mul_single, point, 192, scalar, 32, out, 192);
-}
-
-extern "C" int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96,
-                                             size_t count, uint8_t out[64]) {
-    if (!jacobians96 || !out || count == 0) return EM_ERR_INPUT;
-    int rc = require_gpu();
-    if (rc) return rc;
-    uint8_t *d_in, *d_out;
-    HIP_TRY(hipMalloc(&d_in, 96 * count));
-    HIP_TRY(hipMalloc(&d_out, 64));
-    HIP_TRY(hipMemcpy(d_in, jacobians96, 96 * count, hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_g1_combine<Bn254G1>, dim3(1), dim3(64), 0, 0, d_in,
-                       count, d_out);
-    HIP_TRY(hipMemcpy(out, d_out, 64, hipMemcpyDeviceToHost));
-    (void)hipFree(d_in);
-    (void)hipFree(d_out);
-    return EM_OK;
-}
-
-// HOST-side combine of the N>1 exchange payload: `count` = world size (a
-// handful of 96-B Jacobian partials, one per rank).  This is boundary glue
-// like the delivery-time Horner — the MSM compute stays on the GPU; doing
-// these few adds on the host avoids interposing a kernel + 2 copies on the
-// default stream while pipelined steps are in flight.
-extern "C" int ethrex_mi355_bn254_g1_combine_cpu(const uint8_t *jacobians96,
-                                                 size_t count,
-                                                 uint8_t out[64]) {
-    if (!jacobians96 || !out || count == 0) return EM_ERR_INPUT;
-    g1jT<Bn254G1> acc = g1_inf9<Bn254G1>();
-    for (size_t i = 0; i < count; i++) {
-        g1jT<Bn254G1> t;
-        if (!g1_jac_from_be9<Bn254G1>(t, jacobians96 + 96 * i)) continue;
-        acc = g1_add9<Bn254G1>(acc, t);
-    }
-    g1_to_affine_be9<Bn254G1>(out, acc);
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_bls12381_g1_combine(const uint8_t *jacobians144,
-                                                size_t count, uint8_t out[96]) {
-    if (!jacobians144 || !out || count == 0) return EM_ERR_INPUT;
-    int rc = require_gpu();
-    if (rc) return rc;
-    uint8_t *d_in, *d_out;
-    HIP_TRY(hipMalloc(&d_in, 144 * count));
-    HIP_TRY(hipMalloc(&d_out, 96));
-    HIP_TRY(hipMemcpy(d_in, jacobians144, 144 * count, hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_g1_combine<BlsG1>, dim3(1), dim3(64), 0, 0, d_in,
-                       count, d_out);
-    HIP_TRY(hipMemcpy(out, d_out, 96, hipMemcpyDeviceToHost));
-    (void)hipFree(d_in);
-    (void)hipFree(d_out);
-    return EM_OK;
-}
-
-// ============================ NTT plan ============================
-
-struct em_ntt_plan {
-    size_t n;
-    int logn;
-    bool fused = false;       // four-step path (13 <= logn <= 24)
-    bool fused2 = false;      // two-level four-step (25 <= logn <= 26)
-    int logN1 = 0, logN2 = 0;
-    int logM1 = 0, logM2 = 0; // inner split of N2 (fused2)
-    int cur = 0;              // which buffer holds the data: 0=d_data 1=d_work
-    fe9 *d_data = nullptr;
-    fe9 *d_work = nullptr;    // fused: transpose ping-pong buffer
-    uint8_t *d_bytes = nullptr;
-    fe9 *d_tw = nullptr;      // fallback: forward twiddles, n/2
-    fe9 *d_tw_inv = nullptr;
-    fe9 *d_twfull = nullptr;      // fused: w^j, j<n
-    fe9 *d_twfull_inv = nullptr;
-    fe9 *d_twrow1 = nullptr;      // fused: N1/2 row twiddles (+inv)
-    fe9 *d_twrow1_inv = nullptr;
-    fe9 *d_twrow2 = nullptr;      // fused: N2/2 (+inv)
-    fe9 *d_twrow2_inv = nullptr;
-    fe9 *d_twfull2 = nullptr;     // fused2: inner w_{N2}^j, j<N2 (+inv)
-    fe9 *d_twfull2_inv = nullptr;
-    fe9 *d_twrowA = nullptr;      // fused2: inner M1/2 row twiddles (+inv)
-    fe9 *d_twrowA_inv = nullptr;
-    fe9 *d_twrowB = nullptr;      // fused2: inner M2/2 (+inv)
-    fe9 *d_twrowB_inv = nullptr;
-    fe9 *d_ninv = nullptr;        // 1/n (fe9 Montgomery)
-    uint32_t *d_err = nullptr;
-    hipEvent_t ev[4];
-    double last_ms[3] = {0, 0, 0};
-};
-
-// build w2k powers on host and launch k_gen_twiddles: tw[j] = w^j, j < count
-static int gen_tw_table(fe9 *d_out, size_t count, int bits, bool inverse,
-                        int log_size /* transform size 2^log_size */) {
-    fe9 w = fe9_load(inverse ? bn254::FR9_W28_INV : bn254::FR9_W28);
-    for (int k = bn254::FR_TWO_ADICITY; k > log_size; k--)
-        w = mont_sqr9<Fr9T>(w);
-    fe9 w2k[32];
-    w2k[0] = log_size == 0 ? fe9_load(bn254::FR9_ONE) : w;
-    for (int k = 1; k < bits && k < 32; k++) w2k[k] = mont_sqr9<Fr9T>(w2k[k - 1]);
-    fe9 *d_w2k;
-    HIP_TRY(hipMalloc(&d_w2k, sizeof(w2k)));
-    HIP_TRY(hipMemcpy(d_w2k, w2k, sizeof(w2k), hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_gen_twiddles, dim3(blocks_for(count, 256)), dim3(256),
-                       0, 0, d_out, count, d_w2k, bits);
-    HIP_TRY(hipDeviceSynchronize());
-    hipError_t e2 = hipFree(d_w2k);
-    (void)e2;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
-    if (!plan || n == 0 || (n & (n - 1))) return EM_ERR_INPUT;
-    int logn = 0;
-    while (((size_t)1 << logn) < n) logn++;
-    if (logn > bn254::FR_TWO_ADICITY) return EM_ERR_INPUT;
-    int rc = require_gpu();
-    if (rc) return rc;
-    em_ntt_plan *p = new em_ntt_plan();
-    p->n = n;
-    p->logn = logn;
-    p->fused = (logn > 12 && logn <= 24);
-    p->fused2 = (logn >= 25 && logn <= 26);
-    size_t half = n > 1 ? n / 2 : 1;
-    hipError_t e = hipSuccess;
-    auto mal = [&](void **ptr, size_t bytes) {
-        if (e == hipSuccess) e = hipMalloc(ptr, bytes);
-    };
-    mal((void **)&p->d_data, n * sizeof(fe9));
-    mal((void **)&p->d_bytes, n * 32);
-    mal((void **)&p->d_err, 4);
-    if (p->fused || p->fused2) {
-        if (p->fused) {
-            p->logN1 = (logn + 1) / 2;
-            p->logN2 = logn / 2;
-        } else {
-            // two-level: outer P1 rows of 2^12; inner four-step over the
-            // 2^13/2^14-long rows (M1 x M2, both <= 2^7)
-            p->logN1 = 12;
-            p->logN2 = logn - 12;
-            p->logM1 = (p->logN2 + 1) / 2;
-            p->logM2 = p->logN2 / 2;
-        }
-        mal((void **)&p->d_work, n * sizeof(fe9));
-        mal((void **)&p->d_twfull, n * sizeof(fe9));
-        mal((void **)&p->d_twfull_inv, n * sizeof(fe9));
-        mal((void **)&p->d_twrow1, ((size_t)1 << (p->logN1 - 1)) * sizeof(fe9));
-        mal((void **)&p->d_twrow1_inv, ((size_t)1 << (p->logN1 - 1)) * sizeof(fe9));
-        mal((void **)&p->d_twrow2,
-            ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe9));
-        mal((void **)&p->d_twrow2_inv,
-            ((size_t)1 << (p->logN2 > 0 ? p->logN2 - 1 : 0)) * sizeof(fe9));
-        mal((void **)&p->d_ninv, sizeof(fe9));
-        if (p->fused2) {
-            mal((void **)&p->d_twfull2, ((size_t)1 << p->logN2) * sizeof(fe9));
-            mal((void **)&p->d_twfull2_inv,
-                ((size_t)1 << p->logN2) * sizeof(fe9));
-            mal((void **)&p->d_twrowA,
-                ((size_t)1 << (p->logM1 - 1)) * sizeof(fe9));
-            mal((void **)&p->d_twrowA_inv,
-                ((size_t)1 << (p->logM1 - 1)) * sizeof(fe9));
-            mal((void **)&p->d_twrowB,
-                ((size_t)1 << (p->logM2 - 1)) * sizeof(fe9));
-            mal((void **)&p->d_twrowB_inv,
-                ((size_t)1 << (p->logM2 - 1)) * sizeof(fe9));
-        }
-    } else {
-        mal((void **)&p->d_tw, half * sizeof(fe9));
-        mal((void **)&p->d_tw_inv, half * sizeof(fe9));
-    }
-    for (int i = 0; i < 4 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
-    if (e != hipSuccess) {
-        ethrex_mi355_ntt_plan_destroy(p);
-        return hip_fail(e, "ntt_plan_create");
-    }
-    if (p->fused || p->fused2) {
-        int rc2;
-        if ((rc2 = gen_tw_table(p->d_twfull, n, logn, false, logn))) return rc2;
-        if ((rc2 = gen_tw_table(p->d_twfull_inv, n, logn, true, logn))) return rc2;
-        if ((rc2 = gen_tw_table(p->d_twrow1, (size_t)1 << (p->logN1 - 1),
-                                p->logN1, false, p->logN1))) return rc2;
-        if ((rc2 = gen_tw_table(p->d_twrow1_inv, (size_t)1 << (p->logN1 - 1),
-                                p->logN1, true, p->logN1))) return rc2;
-        if ((rc2 = gen_tw_table(p->d_twrow2, (size_t)1 << (p->logN2 - 1),
-                                p->logN2, false, p->logN2))) return rc2;
-        if ((rc2 = gen_tw_table(p->d_twrow2_inv, (size_t)1 << (p->logN2 - 1),
-                                p->logN2, true, p->logN2))) return rc2;
-        if (p->fused2) {
-            size_t n2 = (size_t)1 << p->logN2;
-            if ((rc2 = gen_tw_table(p->d_twfull2, n2, p->logN2, false,
-                                    p->logN2))) return rc2;
-            if ((rc2 = gen_tw_table(p->d_twfull2_inv, n2, p->logN2, true,
-                                    p->logN2))) return rc2;
-            if ((rc2 = gen_tw_table(p->d_twrowA, (size_t)1 << (p->logM1 - 1),
-                                    p->logM1, false, p->logM1))) return rc2;
-            if ((rc2 = gen_tw_table(p->d_twrowA_inv,
-                                    (size_t)1 << (p->logM1 - 1), p->logM1,
-                                    true, p->logM1))) return rc2;
-            if ((rc2 = gen_tw_table(p->d_twrowB, (size_t)1 << (p->logM2 - 1),
-                                    p->logM2, false, p->logM2))) return rc2;
-            if ((rc2 = gen_tw_table(p->d_twrowB_inv,
-                                    (size_t)1 << (p->logM2 - 1), p->logM2,
-                                    true, p->logM2))) return rc2;
-        }
-        fe9 ninv = fe9_load(bn254::FR9_INV_POW2[logn]);
-        HIP_TRY(hipMemcpy(p->d_ninv, &ninv, sizeof(fe9), hipMemcpyHostToDevice));
-    } else {
-        int rc2;
-        if ((rc2 = gen_tw_table(p->d_tw, half, logn ? logn : 1, false, logn)))
-            return rc2;
-        if ((rc2 = gen_tw_table(p->d_tw_inv, half, logn ? logn : 1, true, logn)))
-            return rc2;
-    }
-    *plan = p;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_ntt_plan_destroy(em_ntt_plan *p) {
-    if (!p) return EM_ERR_INPUT;
-    (void)hipFree(p->d_data);
-    (void)hipFree(p->d_work);
-    (void)hipFree(p->d_bytes);
-    (void)hipFree(p->d_tw);
-    (void)hipFree(p->d_tw_inv);
-    (void)hipFree(p->d_twfull);
-    (void)hipFree(p->d_twfull2);
-    (void)hipFree(p->d_twfull2_inv);
-    (void)hipFree(p->d_twrowA);
-    (void)hipFree(p->d_twrowA_inv);
-    (void)hipFree(p->d_twrowB);
-    (void)hipFree(p->d_twrowB_inv);
-    (void)hipFree(p->d_twfull_inv);
-    (void)hipFree(p->d_twrow1);
-    (void)hipFree(p->d_twrow1_inv);
-    (void)hipFree(p->d_twrow2);
-    (void)hipFree(p->d_twrow2_inv);
-    (void)hipFree(p->d_ninv);
-    (void)hipFree(p->d_err);
-    delete p;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_ntt_upload(em_ntt_plan *p, const uint8_t *elems32) {
-    if (!p || !elems32) return EM_ERR_INPUT;
-    HIP_TRY(hipMemset(p->d_err, 0, 4));
-    HIP_TRY(hipMemcpy(p->d_bytes, elems32, p->n * 32, hipMemcpyHostToDevice));
-    hipLaunchKernelGGL(k_fr_from_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       p->d_bytes, p->d_data, p->n, p->d_err);
-    p->cur = 0;
-    uint32_t err;
-    HIP_TRY(hipMemcpy(&err, p->d_err, 4, hipMemcpyDeviceToHost));
-    return err ? EM_ERR_INPUT : EM_OK;
-}
-
-extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
-    if (!p) return EM_ERR_INPUT;
-    size_t n = p->n;
-    fe9 *cur = p->cur ? p->d_work : p->d_data;
-    fe9 *oth = p->cur ? p->d_data : p->d_work;
-    HIP_TRY(hipEventRecord(p->ev[0], 0));
-    if (p->fused) {
-        uint32_t N1 = 1u << p->logN1, N2 = 1u << p->logN2;
-        // T0: A[r][c] -> A1[c][r]
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256), 0,
-                           0, cur, oth, N1, N2);
-        HIP_TRY(hipEventRecord(p->ev[1], 0));
-        // P1: row NTT_N1 over each of the N2 rows + w^(k1*c) twiddle
-        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(1024), 0, 0, oth,
-                           p->logN1, inverse ? p->d_twrow1_inv : p->d_twrow1,
-                           inverse ? p->d_twfull_inv : p->d_twfull,
-                           (const fe9 *)nullptr);
-        // T1
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(N1 / 32, N2 / 32), dim3(256), 0,
-                           0, oth, cur, N2, N1);
-        // P2: row NTT_N2 (+ 1/n scale on iNTT)
-        hipLaunchKernelGGL(k_ntt_row, dim3(N1), dim3(1024), 0, 0, cur,
-                           p->logN2, inverse ? p->d_twrow2_inv : p->d_twrow2,
-                           (const fe9 *)nullptr,
-                           inverse ? p->d_ninv : (const fe9 *)nullptr);
-        // T2: natural order
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256), 0,
-                           0, cur, oth, N1, N2);
-        p->cur ^= 1;
-    } else if (p->fused2) {
-        // two-level four-step (25 <= logn <= 26): the outer P2 row length
-        // (2^13/2^14) exceeds the 4096-element LDS row kernel, so each of
-        // the N1 outer rows is itself transformed by a BATCHED inner
-        // four-step (M1 x M2, both <= 2^7).  The inner transform is
-        // natural-in / natural-out, so it composes exactly where the
-        // one-level P2 sat.  8 full-array passes instead of logn.
-        uint32_t N1 = 1u << p->logN1, N2 = 1u << p->logN2;
-        uint32_t M1 = 1u << p->logM1, M2 = 1u << p->logM2;
-        const fe9 *tr1 = inverse ? p->d_twrow1_inv : p->d_twrow1;
-        const fe9 *tf = inverse ? p->d_twfull_inv : p->d_twfull;
-        const fe9 *tf2 = inverse ? p->d_twfull2_inv : p->d_twfull2;
-        const fe9 *trA = inverse ? p->d_twrowA_inv : p->d_twrowA;
-        const fe9 *trB = inverse ? p->d_twrowB_inv : p->d_twrowB;
-        // T0 + P1 + T1: outer column NTTs (length N1) + w_n^(k1 c)
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256),
-                           0, 0, cur, oth, N1, N2);
-        HIP_TRY(hipEventRecord(p->ev[1], 0));
-        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(1024), 0, 0, oth,
-                           p->logN1, tr1, tf, (const fe9 *)nullptr);
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(N1 / 32, N2 / 32), dim3(256),
-                           0, 0, oth, cur, N2, N1);
-        // inner batched four-step over the N1 rows of length N2 = M1*M2
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(M2 / 32, M1 / 32, N1),
-                           dim3(256), 0, 0, cur, oth, M1, M2);
-        hipLaunchKernelGGL(k_ntt_row_small, dim3(N1 * M2 * M1 / 1024),
-                           dim3(512), 0, 0, oth, p->logM1, trA, tf2,
-                           (const fe9 *)nullptr, M2 - 1);
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(M1 / 32, M2 / 32, N1),
-                           dim3(256), 0, 0, oth, cur, M2, M1);
-        hipLaunchKernelGGL(k_ntt_row_small, dim3(N1 * M1 * M2 / 1024),
-                           dim3(512), 0, 0, cur, p->logM2, trB,
-                           (const fe9 *)nullptr,
-                           inverse ? p->d_ninv : (const fe9 *)nullptr,
-                           0xffffffffu);
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(M2 / 32, M1 / 32, N1),
-                           dim3(256), 0, 0, cur, oth, M1, M2);
-        // outer T2 -> natural order
-        hipLaunchKernelGGL(k_transpose_fe9, dim3(N2 / 32, N1 / 32), dim3(256),
-                           0, 0, oth, cur, N1, N2);
-        // data ends in `cur` (8 passes): no buffer flip
-    } else {
-        if (n > 1) {
-            hipLaunchKernelGGL(k_bit_reverse, dim3(blocks_for(n, 256)), dim3(256),
-                               0, 0, cur, n, p->logn);
-        }
-        HIP_TRY(hipEventRecord(p->ev[1], 0));
-        const fe9 *tw = inverse ? p->d_tw_inv : p->d_tw;
-        for (int s = 1; s <= p->logn; s++) {
-            hipLaunchKernelGGL(k_ntt_stage, dim3(blocks_for(n / 2, 256)),
-                               dim3(256), 0, 0, cur, tw, n, p->logn, s);
-        }
-        if (inverse) {
-            hipLaunchKernelGGL(k_ntt_scale, dim3(blocks_for(n, 256)), dim3(256),
-                               0, 0, cur, n, p->logn);
-        }
-    }
-    HIP_TRY(hipEventRecord(p->ev[2], 0));
-    HIP_TRY(hipDeviceSynchronize());
-    float ms;
-    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[1]));
-    p->last_ms[0] = ms;
-    HIP_TRY(hipEventElapsedTime(&ms, p->ev[1], p->ev[2]));
-    p->last_ms[1] = ms;
-    HIP_TRY(hipEventElapsedTime(&ms, p->ev[0], p->ev[2]));
-    p->last_ms[2] = ms;
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_ntt_download(em_ntt_plan *p, uint8_t *elems32) {
-    if (!p || !elems32) return EM_ERR_INPUT;
-    fe9 *cur = p->cur ? p->d_work : p->d_data;
-    hipLaunchKernelGGL(k_fr_to_be, dim3(blocks_for(p->n, 256)), dim3(256), 0, 0,
-                       cur, p->d_bytes, p->n);
-    HIP_TRY(hipMemcpy(elems32, p->d_bytes, p->n * 32, hipMemcpyDeviceToHost));
-    return EM_OK;
-}
-
-extern "C" int ethrex_mi355_ntt_last_times(em_ntt_plan *p, double times_ms[3]) {
-    if (!p || !times_ms) return EM_ERR_INPUT;
-    memcpy(times_ms, p->last_ms, sizeof p->last_ms);
-    return EM_OK;
-}
-
-// ============================ one-shot NTT ============================
-
-extern "C" int ethrex_mi355_bn254_fr_ntt(uint8_t *elems32, size_t n, int inverse) {
-    if (!elems32 || n == 0) return EM_ERR_INPUT;
-    em_ntt_plan *p = nullptr;
-    int rc = ethrex_mi355_ntt_plan_create(n, &p);
-    if (rc) return rc;
-    rc = ethrex_mi355_ntt_upload(p, elems32);
-    if (!rc) rc = ethrex_mi355_ntt_run(p, inverse);
-    if (!rc) rc = ethrex_mi355_ntt_download(p, elems32);
-    ethrex_mi355_ntt_plan_destroy(p);
-    return rc;
-}
-
-// ============================ input generation ============================
-// Product-side restatement of BASELINE.md's deterministic input scheme
-// (splitmix64-seeded xoshiro256++, rejection to [0, r), 254-bit mask).
-// Parity-tested against the oracle's independent restatement.
-
-namespace {
-
-struct Xosh {
-    uint64_t s[4];
-};
-
-uint64_t splitmix64_next(uint64_t &x) {
-    uint64_t z = (x += 0x9e3779b97f4a7c15ull);
-    z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
-    z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
-    return z ^ (z >> 31);
-}
-
-uint64_t rotl64(uint64_t x, int k) { return (x << k) | (x >> (64 - k)); }
-
-uint64_t xosh_next(Xosh &g) {
-    uint64_t r = rotl64(g.s[0] + g.s[3], 23) + g.s[0];
-    uint64_t t = g.s[1] << 17;
-    g.s[2] ^= g.s[0];
-    g.s[3] ^= g.s[1];
-    g.s[1] ^= g.s[2];
-    g.s[0] ^= g.s[3];
-    g.s[2] ^= t;
-    g.s[3] = rotl64(g.s[3], 45);
-    return r;
-}
-
-}  // namespace
-
-extern "C" void ethrex_mi355_gen_fr(uint64_t seed, size_t n, uint8_t *out) {
-    Xosh g;
-    uint64_t sm = seed;
-    for (int i = 0; i < 4; i++) g.s[i] = splitmix64_next(sm);
-    const fe4 rmod{{Fr::MOD[0], Fr::MOD[1], Fr::MOD[2], Fr::MOD[3]}};
-    for (size_t i = 0; i < n; i++) {
-        fe4 s;
-        do {
-            s.v[0] = xosh_next(g);
-            s.v[1] = xosh_next(g);
-            s.v[2] = xosh_next(g);
-            s.v[3] = xosh_next(g) & 0x3fffffffffffffffull;
-        } while (fe_geq(s, rmod));
-        fe_to_be(out + 32 * i, s);
-    }
-}
-
-// same scheme for BLS12-381 Fr (255-bit mask, reject >= r_bls)
-extern "C" void ethrex_mi355_bls_gen_fr(uint64_t seed, size_t n, uint8_t *out) {
-    Xosh g;
-    uint64_t sm = seed;
-    for (int i = 0; i < 4; i++) g.s[i] = splitmix64_next(sm);
-    const fe4 rmod{{bn254::FrB::MOD[0], bn254::FrB::MOD[1], bn254::FrB::MOD[2],
-                    bn254::FrB::MOD[3]}};
-    for (size_t i = 0; i < n; i++) {
-        fe4 s;
-        do {
-            s.v[0] = xosh_next(g);
-            s.v[1] = xosh_next(g);
-            s.v[2] = xosh_next(g);
-            s.v[3] = xosh_next(g) & 0x7fffffffffffffffull;
-        } while (fe_geq(s, rmod));
-        fe_to_be(out + 32 * i, s);
-    }
 }

@@ -89,7 +89,7 @@ __device__ __forceinline__ uint32_t msm_digit(const fe4 &k, int w) {
 
 // 64-byte BE affine -> Montgomery(2^261) g1a9 + infinity flag; off-curve ->
 // err.  Coordinates reduced mod p, (0,0) = identity (provider.rs:252-268).
-__global__ void k_parse_points(const uint8_t *__restrict__ in,
+static __global__ void k_parse_points(const uint8_t *__restrict__ in,
                                g1a9 *__restrict__ pts, uint8_t *__restrict__ inf,
                                size_t n, uint32_t *__restrict__ err) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -109,7 +109,7 @@ __global__ void k_parse_points(const uint8_t *__restrict__ in,
 }
 
 // P_i = (start+i+1)*G directly in HBM; per-thread affine conversion.
-__global__ void k_gen_points(g1a9 *__restrict__ pts, uint8_t *__restrict__ inf,
+static __global__ void k_gen_points(g1a9 *__restrict__ pts, uint8_t *__restrict__ inf,
                              size_t n, uint64_t start) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
@@ -126,7 +126,7 @@ __global__ void k_gen_points(g1a9 *__restrict__ pts, uint8_t *__restrict__ inf,
 }
 
 // download points as BE affine bytes (for parity tests)
-__global__ void k_points_to_be(const g1a9 *__restrict__ pts,
+static __global__ void k_points_to_be(const g1a9 *__restrict__ pts,
                                const uint8_t *__restrict__ inf,
                                uint8_t *__restrict__ out, size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -140,7 +140,7 @@ __global__ void k_points_to_be(const g1a9 *__restrict__ pts,
 }
 
 // 32-byte BE scalars -> canonical fe4 reduced mod r (from_be_bytes_mod_order)
-__global__ void k_parse_scalars(const uint8_t *__restrict__ in,
+static __global__ void k_parse_scalars(const uint8_t *__restrict__ in,
                                 fe4 *__restrict__ out, size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
@@ -153,7 +153,7 @@ __global__ void k_parse_scalars(const uint8_t *__restrict__ in,
 // t = raw + carry_in; t <= 2^(C-1) keeps d = +t, else d = t - 2^C (carry 1);
 // bucket id = |d| - 1 (zero digits / infinities park at id 0 with SGN_SKIP).
 template <typename CFG>
-__global__ void k_digits(const fe4 *__restrict__ scalars,
+static __global__ void k_digits(const fe4 *__restrict__ scalars,
                          const uint8_t *__restrict__ inf,
                          uint32_t *__restrict__ keys, uint32_t *__restrict__ vals,
                          size_t n) {
@@ -195,7 +195,7 @@ __global__ void k_digits(const fe4 *__restrict__ scalars,
 
 // ---- bucket segment offsets: lower_bound of each bucket id ----
 template <typename CFG>
-__global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total,
+static __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total,
                           uint32_t *__restrict__ offsets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b > CFG::NBUCKETS) return;
@@ -218,7 +218,7 @@ __global__ void k_offsets(const uint32_t *__restrict__ sorted_keys, size_t total
 // processes similar-length runs (a wave pays max-of-64 Poisson run lengths
 // otherwise: measured ~1.2-1.5x divergence loss) ----
 template <typename CFG>
-__global__ void k_bucket_lengths(const uint32_t *__restrict__ offsets,
+static __global__ void k_bucket_lengths(const uint32_t *__restrict__ offsets,
                                  uint32_t *__restrict__ len,
                                  uint32_t *__restrict__ ids) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
@@ -299,7 +299,7 @@ constexpr int PAIR_K = EM_PAIR_K;  // pair slots per thread per level
 
 // per-bucket pair counts for the next level: ceil(len/2); digit-0 buckets
 // contribute nothing (skip_d0 set at level 0; empty thereafter).
-__global__ void k_pair_counts(const uint32_t *__restrict__ off_in,
+static __global__ void k_pair_counts(const uint32_t *__restrict__ off_in,
                               uint32_t *__restrict__ cnt, uint32_t nbuckets,
                               uint32_t dmask, int skip_d0) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
@@ -599,7 +599,7 @@ k_emit_windows(const g1jT<C> *__restrict__ windows, uint8_t *__restrict__ out) {
 
 // ---- single-op kernels (zisk-mirror ABI + Jacobian combine) ----
 
-__global__ void k_g1_add_single(const uint8_t *in /* 128 B: p1||p2 */,
+static __global__ void k_g1_add_single(const uint8_t *in /* 128 B: p1||p2 */,
                                 uint8_t *out, uint32_t *err) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     g1j9 acc = g1_inf9();
@@ -618,7 +618,7 @@ __global__ void k_g1_add_single(const uint8_t *in /* 128 B: p1||p2 */,
     g1_to_affine_be9(out, acc);
 }
 
-__global__ void k_g1_mul_single(const uint8_t *in /* 96 B: point||scalar */,
+static __global__ void k_g1_mul_single(const uint8_t *in /* 96 B: point||scalar */,
                                 uint8_t *out, uint32_t *err) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     fe9 x = to_mont9(fe9_from_be(in));
@@ -642,7 +642,7 @@ __global__ void k_g1_mul_single(const uint8_t *in /* 96 B: point||scalar */,
 
 // combine count Jacobian partials (96-B BE canonical each) -> affine
 template <typename C>
-__global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
+static __global__ void k_g1_combine(const uint8_t *__restrict__ in, size_t count,
                              uint8_t *__restrict__ out) {
     using F = typename C::F;
     constexpr int NB = F::W64 * 8;
@@ -671,7 +671,7 @@ using CfgFB = msm_cfg<FB_C, FB_C>;  // NWIN=1: single merged window space
 
 // P_ext[w*n + i] = 2^(12w) * P_i (affine); inf entries follow the base flag
 template <typename C>
-__global__ void k_fb_precompute(const g1aT<C> *__restrict__ pts,
+static __global__ void k_fb_precompute(const g1aT<C> *__restrict__ pts,
                                 const uint8_t *__restrict__ inf, size_t n,
                                 g1aT<C> *__restrict__ ext) {
     size_t e = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -693,7 +693,7 @@ __global__ void k_fb_precompute(const g1aT<C> *__restrict__ pts,
 
 // digits for fixed-base: entry e = (w, i); key = digit only (single window
 // space), value = index into the precomputed table
-__global__ void k_fb_digits(const fe4 *__restrict__ scalars,
+static __global__ void k_fb_digits(const fe4 *__restrict__ scalars,
                             const uint8_t *__restrict__ inf,
                             uint32_t *__restrict__ keys,
                             uint32_t *__restrict__ vals, size_t n) {
@@ -714,7 +714,7 @@ __global__ void k_fb_digits(const fe4 *__restrict__ scalars,
 
 // 96-byte BE affine -> Montgomery(2^406) g1aB + infinity flag.
 // err bits: 1 = non-canonical/off-curve/subgroup (details via last_error)
-__global__ void k_bls_parse_points(const uint8_t *__restrict__ in,
+static __global__ void k_bls_parse_points(const uint8_t *__restrict__ in,
                                    g1aB *__restrict__ pts,
                                    uint8_t *__restrict__ inf, size_t n,
                                    uint32_t *__restrict__ err) {
@@ -745,7 +745,7 @@ __global__ void k_bls_parse_points(const uint8_t *__restrict__ in,
 }
 
 // P_i = (start+i+1)*G
-__global__ void k_bls_gen_points(g1aB *__restrict__ pts,
+static __global__ void k_bls_gen_points(g1aB *__restrict__ pts,
                                  uint8_t *__restrict__ inf, size_t n,
                                  uint64_t start) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -761,7 +761,7 @@ __global__ void k_bls_gen_points(g1aB *__restrict__ pts,
     inf[i] = 0;
 }
 
-__global__ void k_bls_points_to_be(const g1aB *__restrict__ pts,
+static __global__ void k_bls_points_to_be(const g1aB *__restrict__ pts,
                                    const uint8_t *__restrict__ inf,
                                    uint8_t *__restrict__ out, size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -776,7 +776,7 @@ __global__ void k_bls_points_to_be(const g1aB *__restrict__ pts,
 
 // scalars: raw 256-bit big-endian -> u64[4] LE words (NO reduction:
 // blst SCALAR_BITS = 256); windows cover all 256 bits (16 x 16)
-__global__ void k_bls_parse_scalars(const uint8_t *__restrict__ in,
+static __global__ void k_bls_parse_scalars(const uint8_t *__restrict__ in,
                                     fe4 *__restrict__ out, size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
@@ -788,7 +788,7 @@ __global__ void k_bls_parse_scalars(const uint8_t *__restrict__ in,
 }
 
 // single ops (parity probes, bls_blst.rs g1_add / p1_mult semantics)
-__global__ void k_bls_g1_add_single(const uint8_t *in /* 192 B */, uint8_t *out,
+static __global__ void k_bls_g1_add_single(const uint8_t *in /* 192 B */, uint8_t *out,
                                     uint32_t *err) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     g1jB acc = g1_inf9<BlsG1>();
@@ -812,7 +812,7 @@ __global__ void k_bls_g1_add_single(const uint8_t *in /* 192 B */, uint8_t *out,
     g1_to_affine_be9(out, acc);
 }
 
-__global__ void k_bls_g1_mul_single(const uint8_t *in /* 96+32 B */,
+static __global__ void k_bls_g1_mul_single(const uint8_t *in /* 96+32 B */,
                                     uint8_t *out, uint32_t *err) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     fe14 xr = feT_from_be<FpB14T>(in);
@@ -844,7 +844,7 @@ using g1jG2 = g1jT<BlsG2>;
 
 // parse + validate: canonical coords, (0,0,0,0) identity, on-curve, and
 // r-subgroup (bls_blst.rs read_g2_subgroup) when check_subgroup is set
-__global__ void k_bls_g2_parse_points(const uint8_t *__restrict__ in,
+static __global__ void k_bls_g2_parse_points(const uint8_t *__restrict__ in,
                                       g1aG2 *__restrict__ pts,
                                       uint8_t *__restrict__ inf, size_t n,
                                       uint32_t *__restrict__ err) {
@@ -882,7 +882,7 @@ __global__ void k_bls_g2_parse_points(const uint8_t *__restrict__ in,
 }
 
 // P_i = (start+i+1) * G2gen
-__global__ void k_bls_g2_gen_points(g1aG2 *__restrict__ pts,
+static __global__ void k_bls_g2_gen_points(g1aG2 *__restrict__ pts,
                                     uint8_t *__restrict__ inf, size_t n,
                                     uint64_t start) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -898,7 +898,7 @@ __global__ void k_bls_g2_gen_points(g1aG2 *__restrict__ pts,
     inf[i] = 0;
 }
 
-__global__ void k_bls_g2_points_to_be(const g1aG2 *__restrict__ pts,
+static __global__ void k_bls_g2_points_to_be(const g1aG2 *__restrict__ pts,
                                       const uint8_t *__restrict__ inf,
                                       uint8_t *__restrict__ out, size_t n) {
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -933,7 +933,7 @@ __device__ __forceinline__ int bls_g2_parse_one(const uint8_t *b, g1aG2 &p,
     return 0;
 }
 
-__global__ void k_bls_g2_add_single(const uint8_t *in /* 384 B */,
+static __global__ void k_bls_g2_add_single(const uint8_t *in /* 384 B */,
                                     uint8_t *out, uint32_t *err) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     g1jG2 acc = g1_inf9<BlsG2>();
@@ -951,7 +951,7 @@ __global__ void k_bls_g2_add_single(const uint8_t *in /* 384 B */,
     g1_to_affine_be9<BlsG2>(out, acc);
 }
 
-__global__ void k_bls_g2_mul_single(const uint8_t *in /* 192 + 32 B */,
+static __global__ void k_bls_g2_mul_single(const uint8_t *in /* 192 + 32 B */,
                                     uint8_t *out, uint32_t *err) {
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
     g1aG2 p;

@@ -30,7 +30,8 @@ for _f in ("device_count", "set_device", "bn254_g1_add", "bn254_g1_mul",
            "msm_gen_points", "msm_download_points", "msm_upload_scalars",
            "msm_run", "msm_run_partial", "msm_run_async", "msm_sync",
            "msm_run_partial_async", "msm_wait_one", "bn254_g1_combine_cpu",
-           "msm_last_times", "msm_combine",
+           "msm_last_times", "msm_combine", "msm_scalars_from_ntt",
+           "ntt_device_data",
            "ntt_plan_create", "ntt_plan_destroy", "ntt_upload", "ntt_run",
            "ntt_download", "ntt_last_times"):
     getattr(_lib, f"ethrex_mi355_{_f}").restype = ctypes.c_int
@@ -196,6 +197,18 @@ class MsmPlan:
             return b""
         _check(_lib.ethrex_mi355_msm_wait_one(self._p), "msm_wait_one")
         return bytes(self._pending.pop(0))
+
+    def scalars_from_ntt(self, ntt_plan, offset: int = 0):
+        """wrap-pipeline handoff: take this plan's n scalars from the NTT
+        plan's device-resident output at element `offset` (on-device, no
+        PCIe; the sp1.rs:122-134 wrap flow feeds the witness NTT output
+        into the proving MSM)."""
+        ptr = ctypes.c_void_p()
+        _check(_lib.ethrex_mi355_ntt_device_data(ntt_plan._p,
+                                                 ctypes.byref(ptr), None),
+               "ntt_device_data")
+        _check(_lib.ethrex_mi355_msm_scalars_from_ntt(
+            self._p, ptr, ctypes.c_uint64(offset)), "msm_scalars_from_ntt")
 
     def combine(self, jacobians: bytes, count: int) -> bytes:
         """combine Jacobian partials reusing this plan's device buffers"""

@@ -35,9 +35,27 @@ def proof_output(prover_type: str, proof_bytes: bytes) -> dict:
                       "proof": list(proof_bytes)}}
 
 
+def _witness_of(input_data):
+    """ProgramInput carrying an execution witness: {"witness": {"state":
+    [hex], "headers": [hex]}, "first_block_number": N} (the RpcExecution-
+    Witness shape of the reference's prover cache fixtures,
+    tooling/zkevm_bench/src/cache.rs).  Returns (state, headers, fbn) or
+    None for witness-less plumbing inputs."""
+    if not (isinstance(input_data, dict) and "witness" in input_data):
+        return None
+    w = input_data["witness"]
+
+    def _hx(s):
+        return bytes.fromhex(s[2:] if s.startswith("0x") else s)
+    return ([_hx(s) for s in w["state"]], [_hx(s) for s in w["headers"]],
+            int(input_data["first_block_number"]))
+
+
 class ExecBackend:
-    """Mirror of ExecBackend (backend/exec.rs): statement execution only,
-    sentinel proof bytes; the protocol-level mock prover."""
+    """Mirror of ExecBackend (backend/exec.rs): statement execution on the
+    CPU (the reference's Exec path IS a CPU re-execution), sentinel proof
+    bytes; the protocol-level mock prover.  Witness-carrying inputs run
+    the witness-validation statement (ethrex_amd/witness.py)."""
 
     def prover_type(self) -> str:
         return "Exec"
@@ -45,13 +63,21 @@ class ExecBackend:
     def serialize_input(self, input_data) -> bytes:
         return json.dumps(input_data, sort_keys=True).encode()
 
-    def execute(self, input_data) -> None:
-        self.serialize_input(input_data)  # statement re-execution stand-in
+    def execute(self, input_data):
+        wit = _witness_of(input_data)
+        if wit is None:
+            self.serialize_input(input_data)  # witness-less plumbing input
+            return None
+        from . import witness as W
+        return W.validate_witness(wit[0], wit[1], wit[2], W.cpu_hash_batch)
 
     def prove(self, input_data, proof_format: str):
-        self.execute(input_data)
-        return {"output": hashlib.sha256(
+        statement = self.execute(input_data)
+        out = {"output": hashlib.sha256(
             self.serialize_input(input_data)).hexdigest()}
+        if statement is not None:
+            out["statement"] = statement
+        return out
 
     def verify(self, proof) -> None:
         if "output" not in proof:
@@ -87,15 +113,53 @@ class Mi355Backend:
             hashlib.sha256(self.serialize_input(input_data)).digest()[:8],
             "little")
 
-    def execute(self, input_data) -> None:
-        self.serialize_input(input_data)
+    def _gpu_hash_batch(self, msgs):
+        """batched keccak over the device KeccakPlan (the witness nodes'
+        bulk hashing — §8f row 4)."""
+        import ethrex_amd as ea
+        offs = [0]
+        buf = bytearray()
+        for m in msgs:
+            buf += m
+            offs.append(len(buf))
+        kp = ea.KeccakPlan(max(len(buf), 1), len(msgs))
+        try:
+            kp.upload(bytes(buf), offs)
+            kp.run()
+            out = kp.download()
+        finally:
+            kp.destroy()
+        return [out[32 * i:32 * i + 32] for i in range(len(msgs))]
+
+    def execute(self, input_data):
+        """The witness-validation statement, with the node hashing on the
+        GPU (block_execution_witness.rs rebuild semantics); witness-less
+        plumbing inputs keep the serialize-only behavior."""
+        wit = _witness_of(input_data)
+        if wit is None:
+            self.serialize_input(input_data)
+            return None
+        import ethrex_amd as ea
+        if ea.device_count() < 1:
+            raise BackendError("mi355: no GPU visible (no CPU fallback)")
+        ea.set_device(self.device)
+        from . import witness as W
+        return W.validate_witness(wit[0], wit[1], wit[2],
+                                  self._gpu_hash_batch)
 
     def prove(self, input_data, proof_format: str):
         import ethrex_amd as ea
         if ea.device_count() < 1:
             raise BackendError("mi355: no GPU visible (no CPU fallback)")
         ea.set_device(self.device)
-        seed = self._seed(input_data)
+        statement = self.execute(input_data)
+        if statement is not None:
+            # the proof binds to the STATEMENT (initial state root +
+            # linked-witness commitment), not to a hash of the input JSON
+            seed = int.from_bytes(
+                bytes.fromhex(statement["commitment"])[:8], "little")
+        else:
+            seed = self._seed(input_data)
         n = 1 << self.msm_log2
         plan = ea.MsmPlan(n)
         try:
@@ -110,15 +174,21 @@ class Mi355Backend:
         rc, ntt_out = ea.fr_ntt(elems, m, False)
         if rc != 0:
             raise BackendError(f"mi355: ntt rc={rc}")
-        return {"msm": msm_out, "ntt_digest": hashlib.sha256(ntt_out).digest()}
+        proof = {"msm": msm_out,
+                 "ntt_digest": hashlib.sha256(ntt_out).digest()}
+        if statement is not None:
+            proof["statement"] = statement
+        return proof
 
     def verify(self, proof) -> None:
         if len(proof.get("msm", b"")) != 64:
             raise BackendError("mi355: bad proof")
 
     def to_proof_bytes(self, proof, proof_format: str) -> dict:
-        return proof_output(self.prover_type(),
-                            proof["msm"] + proof["ntt_digest"])
+        pb = proof["msm"] + proof["ntt_digest"]
+        if "statement" in proof:
+            pb += bytes.fromhex(proof["statement"]["commitment"])
+        return proof_output(self.prover_type(), pb)
 
 
 BACKENDS = {"exec": ExecBackend, "mi355": Mi355Backend}

@@ -102,6 +102,11 @@ int ethrex_mi355_bn254_g1_combine(const uint8_t *jacobians96, size_t count,
  * the HOST — boundary glue; avoids touching GPU streams mid-pipeline */
 int ethrex_mi355_bn254_g1_combine_cpu(const uint8_t *jacobians96,
                                       size_t count, uint8_t out[64]);
+/* wrap-pipeline handoff (sp1.rs:122-134 flow): take this plan's n scalars
+ * from an NTT plan's device-resident output (ethrex_mi355_ntt_device_data)
+ * at element `offset` — the composed MSM+NTT step stays on-device */
+int ethrex_mi355_msm_scalars_from_ntt(em_msm_plan *plan, const void *ntt_data,
+                                      uint64_t offset);
 /* per-phase HIP-event timings of the last run, milliseconds:
  * [0]=digits+sort, [1]=bucket accumulation, [2]=bucket reduction,
  * [3]=window combine + affine, [4]=total */
@@ -132,6 +137,10 @@ int ethrex_mi355_ntt_run(em_ntt_plan *plan, int inverse);
 int ethrex_mi355_ntt_download(em_ntt_plan *plan, uint8_t *elems32);
 /* [0]=bit-reverse, [1]=butterfly stages total, [2]=total */
 int ethrex_mi355_ntt_last_times(em_ntt_plan *plan, double times_ms[3]);
+/* device pointer to the current transform output (packed 4x64 Montgomery)
+ * for the on-device MSM handoff (ethrex_mi355_msm_scalars_from_ntt) */
+int ethrex_mi355_ntt_device_data(em_ntt_plan *plan, const void **ptr,
+                                 size_t *n);
 
 /* ---- deterministic input generation (host-side; the product restatement
  *      of BASELINE.md's xoshiro256++/splitmix64 scheme; parity-tested

@@ -92,3 +92,44 @@ def test_proof_with_public_values_variant():
     data = encode_verify_batches_calldata(1, [(1, {"SP1": out})])
     _, (_r, sp1, _t) = decode_verify_batches_calldata(data)
     assert sp1 == [b"zz"]
+
+
+def test_calldata_matches_hand_derived_golden():
+    """Byte-for-byte check against an INDEPENDENTLY hand-derived calldata
+    vector (the round-trip test could not catch an offset-
+    layout misconception shared by encoder and decoder).  The expected
+    bytes below are constructed word-by-word from the Solidity ABI spec
+    (head/tail encoding, element offsets relative to the start of each
+    dynamic array's data area), mirroring send_verify_batches_tx
+    (l1_proof_sender.rs:514-556): two batches, SP1 proofs only.
+    """
+    p1, p2 = b"\xaa\xbb\xcc", b"\x11" * 33
+    batches = [
+        (5, {"SP1": {"Proof": {"proof": list(p1)}}}),
+        (6, {"SP1": {"Proof": {"proof": list(p2)}}}),
+    ]
+    got = encode_verify_batches_calldata(5, batches)
+
+    def w(v):  # one 32-byte big-endian word
+        return v.to_bytes(32, "big")
+
+    # keccak("verifyBatches(uint256,bytes[],bytes[],bytes[])")[:4] — the
+    # keccak itself is pinned by canonical vectors in test_keccak256_*
+    want = bytes.fromhex("9711f750")
+    want += w(5)            # uint256 firstBatchNumber
+    # three bytes[] head offsets, relative to start of args (head = 4 words)
+    want += w(128)          # risc0Proofs: right after the head
+    want += w(128 + 160)    # sp1Proofs: after risc0 tail (160 B, below)
+    want += w(288 + 256)    # tdxProofs: after sp1 tail (256 B, below)
+    # risc0Proofs = ["", ""]: count, two element offsets relative to the
+    # word after the count (2 offset words = 64 B), two empty bytes
+    # elements (a zero length word each) => 5 words = 160 B
+    want += w(2) + w(64) + w(96) + w(0) + w(0)
+    # sp1Proofs = [p1 (3 B -> 2 words), p2 (33 B -> 3 words)]:
+    # count, offsets 64 and 64+64, then len+data+pad for each => 256 B
+    want += w(2) + w(64) + w(128)
+    want += w(3) + p1 + b"\x00" * 29
+    want += w(33) + p2 + b"\x00" * 31
+    # tdxProofs = ["", ""]
+    want += w(2) + w(64) + w(96) + w(0) + w(0)
+    assert got == want

@@ -1,0 +1,128 @@
+"""Witness validation (statement generation) + MPT builder parity — CPU.
+
+The anchor is REAL Ethereum data: the committed hoodi witness fixture
+(extracted from the reference's fixtures/cache/rpc_prover cache, see
+tests/golden/witness_hoodi_1265656.json.gz _provenance).  If our keccak,
+RLP codec, node layout or trie linking deviated from the reference's
+(crates/common/types/block_execution_witness.rs + crates/common/trie),
+the witness would not link to the parent header's state root.
+"""
+import os
+import random
+
+import pytest
+
+from ethrex_amd import rlp, trie, witness
+from ethrex_amd.keccak import keccak256
+
+FIXTURE = os.path.join(os.path.dirname(__file__), "golden",
+                       "witness_hoodi_1265656.json.gz")
+
+
+@pytest.fixture(scope="module")
+def wfix():
+    state, headers, fbn = witness.load_witness_fixture(FIXTURE)
+    node_map, skipped = witness.build_node_map(state,
+                                               witness.cpu_hash_batch)
+    return state, headers, fbn, node_map, skipped
+
+
+def test_witness_links_to_parent_state_root(wfix):
+    state, headers, fbn, _, _ = wfix
+    st = witness.validate_witness(state, headers, fbn,
+                                  witness.cpu_hash_batch)
+    # regression pin: the exact linked shape + commitment of the fixture
+    assert st["initial_state_root"] == (
+        "4bec425c34f89aeb56c78586d586c76044bcc428e7c34ab7c72d389c39ff3eaf")
+    assert st["n_witness_nodes"] == 1705
+    assert st["n_skipped"] == 1  # the 0x80 Null sentinel
+    assert st["n_state_nodes_linked"] == 285
+    assert st["n_accounts"] == 48
+    assert st["n_storage_tries"] == 26
+    assert st["commitment"] == (
+        "8a13cb1b77ffd524c615c0189ef878a107091c0b933834553980b4e05c6e0a51")
+
+
+def test_witness_nodes_reencode_byte_exact(wfix):
+    """Every decodable witness node re-encodes to its original bytes —
+    pins the RLP codec + node layout against 1704 real trie nodes."""
+    state = wfix[0]
+    n = 0
+    for b in state:
+        if b == b"\x80":
+            continue
+        assert rlp.encode(rlp.decode(b)) == b
+        n += 1
+    assert n == 1704
+
+
+def test_rlp_decode_rejects_noncanonical():
+    for bad in (b"\x81\x05", b"\xb8\x01x", b"\x80\x00"):
+        with pytest.raises(ValueError):
+            rlp.decode(bad)
+
+
+def test_empty_trie_hash():
+    # trie.rs:78 constant == keccak256(rlp(b"")) == keccak256(0x80)
+    assert keccak256(b"\x80") == witness.EMPTY_TRIE_HASH
+    assert trie.trie_root({}, witness.cpu_hash_batch) == \
+        witness.EMPTY_TRIE_HASH
+
+
+def test_trie_builder_rebuilds_real_storage_tries(wfix):
+    """Extract every FULLY-present storage trie from the hoodi witness and
+    rebuild it with the product builder: the recomputed root must equal
+    the account's storage_root — the builder pinned end-to-end against
+    real Ethereum tries."""
+    state, headers, fbn, node_map, _ = wfix
+    root = witness.find_parent_state_root(headers, fbn)
+    st = witness.link_trie(root, node_map)
+    rebuilt = 0
+    for v in st.leaf_values:
+        acct = rlp.decode(v)
+        if not (isinstance(acct, list) and len(acct) == 4):
+            continue
+        sroot = acct[2]
+        if sroot == witness.EMPTY_TRIE_HASH or sroot not in node_map:
+            continue
+        leaves, missing = witness.collect_leaves(sroot, node_map)
+        if missing or not leaves:
+            continue  # partial trie: cannot rebuild
+        got = trie.trie_root(leaves, witness.cpu_hash_batch)
+        assert got == sroot
+        rebuilt += 1
+    assert rebuilt >= 3  # the fixture carries several complete tries
+
+
+def test_trie_builder_matches_recursive_reference():
+    from mpt_reference import mpt_root
+    rng = random.Random(7)
+    for case in range(6):
+        n = [1, 2, 3, 17, 100, 400][case]
+        pairs = {}
+        for _ in range(n):
+            k = bytes(rng.randrange(256) for _ in range(32))
+            v = bytes(rng.randrange(256) for _ in range(rng.randrange(1, 40)))
+            pairs[k] = v
+        assert trie.trie_root(pairs, witness.cpu_hash_batch) == \
+            mpt_root(pairs)
+    # shared-prefix + tiny-value shapes (inline < 32 B nodes)
+    pairs = {bytes([0xAB, i]) + b"\x00" * 6: bytes([i]) for i in range(5)}
+    assert trie.trie_root(pairs, witness.cpu_hash_batch) == mpt_root(pairs)
+
+
+def test_account_leaf_matches_witness_accounts(wfix):
+    """account_leaf's RLP matches real account leaf values byte-exactly."""
+    state, headers, fbn, node_map, _ = wfix
+    root = witness.find_parent_state_root(headers, fbn)
+    st = witness.link_trie(root, node_map)
+    checked = 0
+    for v in st.leaf_values:
+        acct = rlp.decode(v)
+        if not (isinstance(acct, list) and len(acct) == 4):
+            continue
+        nonce = int.from_bytes(acct[0], "big")
+        balance = int.from_bytes(acct[1], "big")
+        assert trie.account_leaf(nonce, balance, acct[2], acct[3]) == v
+        checked += 1
+    assert checked >= 40
