@@ -438,6 +438,7 @@ def main():
                     "batched GPU keccak; the reference hashes per node "
                     "on CPU (crates/common/trie)",
         }
+    cpu_baseline = None
     if rank == 0 and not args.no_cpu_baseline:
         cpu_baseline = cpu_baseline_leg()
 
