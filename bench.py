@@ -233,12 +233,12 @@ def main():
             nplan.run(False)
             total_ms.append(nplan.last_times()["total_ms"])
         ntt_dt = (time.perf_counter() - t1) / args.steps
-        # whole-transform algorithmic bytes (fe9 elements are 36 B resident):
-        #  four-step fused path (13<=logn<=24): T0+T1+T2 transposes 72 B/elem
-        #  each, P1 72+36 (tw_full gather), P2 72 => 396 B/elem over 5 passes;
-        #  fallback radix-2: logn stage launches x 72 B/elem.
+        # whole-transform algorithmic bytes (fe4m resident: 32 B/element):
+        #  four-step fused path (13<=logn<=24): T0+T1+T2 transposes 64 B/elem
+        #  each, P1 64+32 (TW2 stream), P2 64 => 352 B/elem over 5 passes;
+        #  fallback radix-2: logn stage launches x 64 B/elem.
         fused = 12 < args.ntt_log2 <= 24
-        ntt_alg_bytes = (396 if fused else 72 * args.ntt_log2) * m
+        ntt_alg_bytes = (352 if fused else 64 * args.ntt_log2) * m
         avg_total_ms = sum(total_ms) / len(total_ms)
         ntt = {
             "metric": "bn254_ntt_elems_per_s",
@@ -249,7 +249,7 @@ def main():
             "path": "four-step-fused" if fused else "radix2-stages",
             "roofline": {
                 "bound": "hbm",
-                "kernel": "whole transform (k_transpose_fe9 + k_ntt_row)"
+                "kernel": "whole transform (k_transpose_fe4 + k_ntt_row)"
                           if fused else "k_ntt_stage passes",
                 "achieved": ntt_alg_bytes / (avg_total_ms / 1000.0),
                 "peak": hbm_peak,

@@ -253,9 +253,45 @@ __device__ __host__ __forceinline__ feL<T::L> mont_mul9(const feL<T::L> &A,
     return r;
 }
 
+// Montgomery squaring: the 81 product mads shrink to 45 (36 cross terms
+// with pre-doubled multiplicands + 9 diagonals) — ~11% fewer instructions
+// per square; 2 of the ~10 muls in a mixed point add are squares.  Column
+// values are IDENTICAL to mont_mul9(a,a) (partial sums are positive and
+// bounded by the final column value), so the overflow budget is unchanged;
+// d[i] = 2*a[i] <= 2^31 fits u32 for both the L=9 (inputs <= 2^30) and
+// L=14 (inputs < 2^29) rules.
 template <typename T = Fq9T>
 __device__ __host__ __forceinline__ feL<T::L> mont_sqr9(const feL<T::L> &a) {
-    return mont_mul9<T>(a, a);
+    constexpr int LN = T::L;
+    u64 t[2 * LN - 1];
+#pragma unroll
+    for (int k = 0; k < 2 * LN - 1; k++) t[k] = 0;
+    u32 d[LN];
+#pragma unroll
+    for (int i = 0; i < LN; i++) d[i] = a.v[i] << 1;
+#pragma unroll
+    for (int i = 0; i < LN; i++) {
+        t[2 * i] += (u64)a.v[i] * a.v[i];
+#pragma unroll
+        for (int j = i + 1; j < LN; j++) t[i + j] += (u64)d[i] * a.v[j];
+    }
+#pragma unroll
+    for (int k = 0; k < LN; k++) {
+        u32 m = ((u32)t[k] * T::N0INV) & bn254::FQ9_MASK;
+#pragma unroll
+        for (int j = 0; j < LN; j++) t[k + j] += (u64)m * T::P[j];
+        t[k + 1] += t[k] >> 29;
+    }
+    feL<LN> r;
+    u64 c = 0;
+#pragma unroll
+    for (int k = LN; k < 2 * LN - 1; k++) {
+        c += t[k];
+        r.v[k - LN] = (u32)c & bn254::FQ9_MASK;
+        c >>= 29;
+    }
+    r.v[LN - 1] = (u32)c;
+    return r;
 }
 
 // ---- conversions canonical u64 words <-> 29-bit limbs ----

@@ -293,10 +293,17 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     mal((void **)&p->d_offsets2, ((size_t)nbuckets + 1) * 4);
     mal((void **)&p->d_sched2, (size_t)nbuckets * 4);
     mal((void **)&p->d_out2, (size_t)NWIN_MAX * msm_plan_t<C>::JB);
+    // the compute chain gets the higher stream priority: the VALU-bound
+    // bucket walk should not lose workgroup-dispatch slots to the next
+    // step's (HBM-bound, latency-tolerant) sort chain
+    int prio_lo = 0, prio_hi = 0;
+    (void)hipDeviceGetStreamPriorityRange(&prio_lo, &prio_hi);
     if (e == hipSuccess)
-        e = hipStreamCreateWithFlags(&p->s_sort, hipStreamNonBlocking);
+        e = hipStreamCreateWithPriority(&p->s_sort, hipStreamNonBlocking,
+                                        prio_lo);
     if (e == hipSuccess)
-        e = hipStreamCreateWithFlags(&p->s_comp, hipStreamNonBlocking);
+        e = hipStreamCreateWithPriority(&p->s_comp, hipStreamNonBlocking,
+                                        prio_hi);
     for (int i = 0; i < 2 && e == hipSuccess; i++) {
         if (e == hipSuccess)
             e = hipHostMalloc((void **)&p->h_out[i],

@@ -159,16 +159,22 @@ k_transpose_fe4(const fe4 *__restrict__ src, fe4 *__restrict__ dst,
 // registers per LDS round trip), radix-2^2 / radix-2 tail for logM % 3.
 // tw2 (fe4m, may be null): output k scaled by TW2[c*M + k], c = blockIdx
 // & cmask; scale (fe9, may be null): iNTT 1/n factor.
+// skewed LDS indexing: physical slot = i + (i >> 6).  The bit-reversed
+// load scatter otherwise lands a whole wave on one bank group (rev of
+// consecutive i strides 2048 elements; 9*2048 dwords = 0 mod 64 banks
+// = a 64-way conflict); with the skew, lane l's slot moves by
+// rev6(l)*65 -> 9*rev6(l) mod 64, a permutation of the banks.
+#define EM_SK(i) ((i) + ((i) >> 6))
 __global__ void __launch_bounds__(512)
 k_ntt_row(fe4 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
           const fe4 *__restrict__ tw2, const fe9 *__restrict__ scale,
           uint32_t cmask = 0xffffffffu) {
-    __shared__ fe9 smem[4096];
+    __shared__ fe9 smem[4096 + 64];
     const uint32_t M = 1u << logM;
     fe4 *row = data + (size_t)blockIdx.x * M;
     for (uint32_t i = threadIdx.x; i < M; i += blockDim.x) {
         uint32_t j = __brev(i) >> (32 - logM);
-        smem[j] = fe4m_unpack(row[i]);
+        smem[EM_SK(j)] = fe4m_unpack(row[i]);
     }
     __syncthreads();
     int s = 1;
@@ -178,14 +184,14 @@ k_ntt_row(fe4 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
         for (uint32_t t = threadIdx.x; t < (M >> 3); t += blockDim.x) {
             uint32_t j = t & (q - 1);
             uint32_t base = ((t >> (s - 1)) << (s + 2)) + j;
-            fe9 e0 = smem[base];
-            fe9 e1 = smem[base + q];
-            fe9 e2 = smem[base + 2 * q];
-            fe9 e3 = smem[base + 3 * q];
-            fe9 e4 = smem[base + 4 * q];
-            fe9 e5 = smem[base + 5 * q];
-            fe9 e6 = smem[base + 6 * q];
-            fe9 e7 = smem[base + 7 * q];
+            fe9 e0 = smem[EM_SK(base)];
+            fe9 e1 = smem[EM_SK(base + q)];
+            fe9 e2 = smem[EM_SK(base + 2 * q)];
+            fe9 e3 = smem[EM_SK(base + 3 * q)];
+            fe9 e4 = smem[EM_SK(base + 4 * q)];
+            fe9 e5 = smem[EM_SK(base + 5 * q)];
+            fe9 e6 = smem[EM_SK(base + 6 * q)];
+            fe9 e7 = smem[EM_SK(base + 7 * q)];
             // stage s: stride-q pairs, shared twiddle
             fe9 w0 = tw_row[j << (logM - s)];
             fe9 v;
@@ -233,14 +239,14 @@ k_ntt_row(fe4 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
             v = mont_mul9<Fr9T>(e7, w4d);
             e7 = subm9<Fr9T>(e3, v);
             e3 = add9_n<Fr9T>(e3, v);
-            smem[base] = e0;
-            smem[base + q] = e1;
-            smem[base + 2 * q] = e2;
-            smem[base + 3 * q] = e3;
-            smem[base + 4 * q] = e4;
-            smem[base + 5 * q] = e5;
-            smem[base + 6 * q] = e6;
-            smem[base + 7 * q] = e7;
+            smem[EM_SK(base)] = e0;
+            smem[EM_SK(base + q)] = e1;
+            smem[EM_SK(base + 2 * q)] = e2;
+            smem[EM_SK(base + 3 * q)] = e3;
+            smem[EM_SK(base + 4 * q)] = e4;
+            smem[EM_SK(base + 5 * q)] = e5;
+            smem[EM_SK(base + 6 * q)] = e6;
+            smem[EM_SK(base + 7 * q)] = e7;
         }
         __syncthreads();
     }
@@ -251,20 +257,20 @@ k_ntt_row(fe4 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
             uint32_t j = t & (q - 1);
             uint32_t idx = ((t >> (s - 1)) << (s + 1)) + j;
             fe9 w1 = tw_row[j << (logM - s)];
-            fe9 a = smem[idx];
-            fe9 b = mont_mul9<Fr9T>(smem[idx + q], w1);
-            fe9 c = smem[idx + 2 * q];
-            fe9 d = mont_mul9<Fr9T>(smem[idx + 3 * q], w1);
+            fe9 a = smem[EM_SK(idx)];
+            fe9 b = mont_mul9<Fr9T>(smem[EM_SK(idx + q)], w1);
+            fe9 c = smem[EM_SK(idx + 2 * q)];
+            fe9 d = mont_mul9<Fr9T>(smem[EM_SK(idx + 3 * q)], w1);
             fe9 t0 = add9_n<Fr9T>(a, b);
             fe9 t1 = subm9<Fr9T>(a, b);
             fe9 t2 = add9_n<Fr9T>(c, d);
             fe9 t3 = subm9<Fr9T>(c, d);
             fe9 u2 = mont_mul9<Fr9T>(t2, tw_row[j << (logM - s - 1)]);
             fe9 u3 = mont_mul9<Fr9T>(t3, tw_row[(j + q) << (logM - s - 1)]);
-            smem[idx] = add9_n<Fr9T>(t0, u2);
-            smem[idx + 2 * q] = subm9<Fr9T>(t0, u2);
-            smem[idx + q] = add9_n<Fr9T>(t1, u3);
-            smem[idx + 3 * q] = subm9<Fr9T>(t1, u3);
+            smem[EM_SK(idx)] = add9_n<Fr9T>(t0, u2);
+            smem[EM_SK(idx + 2 * q)] = subm9<Fr9T>(t0, u2);
+            smem[EM_SK(idx + q)] = add9_n<Fr9T>(t1, u3);
+            smem[EM_SK(idx + 3 * q)] = subm9<Fr9T>(t1, u3);
         }
         __syncthreads();
     }
@@ -274,17 +280,17 @@ k_ntt_row(fe4 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
         for (uint32_t t = threadIdx.x; t < (M >> 1); t += blockDim.x) {
             uint32_t j = t & (half - 1);
             uint32_t idx = ((t >> (s - 1)) << s) + j;
-            fe9 u = smem[idx];
-            fe9 v = mont_mul9<Fr9T>(smem[idx + half], tw_row[j << (logM - s)]);
-            smem[idx] = add9_n<Fr9T>(u, v);
-            smem[idx + half] = subm9<Fr9T>(u, v);
+            fe9 u = smem[EM_SK(idx)];
+            fe9 v = mont_mul9<Fr9T>(smem[EM_SK(idx + half)], tw_row[j << (logM - s)]);
+            smem[EM_SK(idx)] = add9_n<Fr9T>(u, v);
+            smem[EM_SK(idx + half)] = subm9<Fr9T>(u, v);
         }
         __syncthreads();
     }
     uint64_t c = blockIdx.x & cmask;
     const fe4 *t2row = tw2 ? tw2 + (size_t)c * M : nullptr;
     for (uint32_t k = threadIdx.x; k < M; k += blockDim.x) {
-        fe9 x = smem[k];
+        fe9 x = smem[EM_SK(k)];
         if (t2row) x = mont_mul9<Fr9T>(x, fe4m_unpack(t2row[k]));
         if (scale) x = mont_mul9<Fr9T>(x, *scale);
         row[k] = fe4m_pack(x);
