@@ -5,10 +5,11 @@ Workload: the metric's configuration — BN254 G1 MSM at 2^24 points
 (BASELINE.json `metric`; fits one GPU) — plus the NTT-at-2^24 secondary leg
 on rank 0.  A "step" is ONE full MSM pass over the resident inputs: scalar
 digit decomposition + sort + bucket accumulation + reduction + window
-combine (+ at N>1 the RCCL AllGather of the 96-B Jacobian partials over
-xGMI and the on-GPU combine).  Inputs (points + scalars) are resident in
-HBM before the timed region; nothing inside the timed region is cached or
-skipped.
+combine (+ at N>1 the AllGather of the 96-B host-delivered Jacobian
+partials over a gloo subgroup and the host combine, overlapped with the
+next step's GPU compute via msm_wait_one).  Inputs (points + scalars) are
+resident in HBM before the timed region; nothing inside the timed region
+is cached or skipped.
 
 N>1: one process per GPU (torch.distributed over RCCL), points sharded by
 index range, scalars seeded per rank (BASELINE.md scheme) — total work is
@@ -91,6 +92,7 @@ def main():
     import ethrex_amd
 
     dist = None
+    gloo_pg = None
     if world > 1:
         import torch
         import torch.distributed as tdist
@@ -98,6 +100,10 @@ def main():
         tdist.init_process_group(backend="nccl")
         torch.cuda.set_device(local_rank)
         dist = tdist
+        # the 96-B partial exchange runs over a gloo subgroup: the payload
+        # is host-side (msm_wait_one delivery) and a CPU gather adds no
+        # H2D/D2H hops and never touches the GPU streams mid-pipeline
+        gloo_pg = tdist.new_group(backend="gloo")
 
     ethrex_amd.set_device(local_rank)
 
@@ -119,7 +125,8 @@ def main():
         # sync-path step (warmup/reference): shard partial -> AllGather of
         # the 96-B Jacobian payloads over RCCL/xGMI -> host combine
         part = plan.run_partial()
-        allparts = allgather_partials(part, dist, device="cuda")
+        allparts = allgather_partials(part, dist, device="cpu",
+                                      group=gloo_pg)
         return ethrex_amd.g1_combine_cpu(allparts, world)
     # warmup
     first = None
@@ -156,10 +163,12 @@ def main():
         for _ in range(1, args.steps):
             plan.run_partial_async()
             part = plan.wait_one()
-            allparts = allgather_partials(part, dist, device="cuda")
+            allparts = allgather_partials(part, dist, device="cpu",
+                                      group=gloo_pg)
             pipelined_last = ethrex_amd.g1_combine_cpu(allparts, world)
         part = plan.wait_one()
-        allparts = allgather_partials(part, dist, device="cuda")
+        allparts = allgather_partials(part, dist, device="cpu",
+                                      group=gloo_pg)
         pipelined_last = ethrex_amd.g1_combine_cpu(allparts, world)
     if world > 1:
         dist.barrier()
@@ -286,7 +295,8 @@ def main():
             if world == 1:
                 return mplan2.run()
             part = mplan2.run_partial()
-            allp = allgather_partials(part, dist, device="cuda")
+            allp = allgather_partials(part, dist, device="cpu",
+                                      group=gloo_pg)
             return ethrex_amd.g1_combine_cpu(allp, world)
 
         for _ in range(max(args.warmup, 1)):
@@ -465,8 +475,8 @@ def main():
                 "windows": MSM_WINDOWS,
                 "signed_digits": True,
                 "point_adds_per_step": point_adds(n_total),
-                "parallelism": f"point-index sharding x{n_gpus}, RCCL "
-                               "allgather of 96B G1 partials" if n_gpus > 1
+                "parallelism": f"point-index sharding x{n_gpus}, pipelined "
+                               "96B partial allgather + host combine" if n_gpus > 1
                                else "single GPU",
             },
             "roofline": roofline,

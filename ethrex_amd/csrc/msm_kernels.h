@@ -54,8 +54,11 @@ struct msm_cfg {
     static constexpr int SORT_BITS = IB + 6;   // + window bits (global sort)
     // SEG tuned per window size (GPU-measured): 16 for the 64K-bucket c=16
     // windows (SEG=8 doubles weighted-reduce work there, reduce 1.8->2.5 ms);
-    // 8 for the smaller FB/c=8 windows (blob reduce 0.98->0.75 ms)
-    static constexpr int SEG = (1 << IB) >= 65536 ? 16 : 8;
+    // 2 for the smaller FB/c=8 windows — their reduce is LATENCY-bound on
+    // few waves (G2: 1024 threads total), so short segment scans + 4x the
+    // threads beat longer per-thread runs (r01 had 8; r02 host Horner
+    // removed the window-scale chains that dominated before)
+    static constexpr int SEG = (1 << IB) >= 65536 ? 16 : 2;
     static constexpr int NSEG = (1 << IB) / SEG;        // segments per window
     static constexpr int RED_BLOCK = 256;
     // when a 256-thread block spans multiple windows the LDS tree is skipped

@@ -17,13 +17,18 @@ def shard_range(n_total: int, world: int, rank: int):
     return lo, hi
 
 
-def allgather_partials(partial96: bytes, tdist, device="cpu"):
+def allgather_partials(partial96: bytes, tdist, device="cpu", group=None):
     """AllGather each rank's 96-B Jacobian partial; returns concatenated
-    world*96 bytes in rank order."""
+    world*96 bytes in rank order.
+
+    The payload is delivered to the HOST by msm_wait_one, so the N>1
+    bench path gathers it over a gloo subgroup (device="cpu"): 96 B over
+    loopback/shm costs ~0.1 ms and — unlike a cuda-tensor gather — adds
+    no H2D/D2H hops and never touches the GPU streams mid-pipeline."""
     import torch
     assert len(partial96) == 96
-    world = tdist.get_world_size()
+    world = tdist.get_world_size(group) if group else tdist.get_world_size()
     t_in = torch.frombuffer(bytearray(partial96), dtype=torch.uint8).to(device)
     t_out = torch.empty(world * 96, dtype=torch.uint8, device=device)
-    tdist.all_gather_into_tensor(t_out, t_in)
+    tdist.all_gather_into_tensor(t_out, t_in, group=group)
     return bytes(t_out.cpu().numpy().tobytes())
