@@ -97,6 +97,8 @@ def main():
         import torch
         import torch.distributed as tdist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")  # container
+        # hostnames may not resolve on the GPU boxes; loopback always does
         tdist.init_process_group(backend="nccl")
         torch.cuda.set_device(local_rank)
         dist = tdist
