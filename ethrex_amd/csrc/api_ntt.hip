@@ -102,8 +102,8 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
     em_ntt_plan *p = new em_ntt_plan();
     p->n = n;
     p->logn = logn;
-    p->fused = (logn > 12 && logn <= 24);
-    p->fused2 = (logn >= 25 && logn <= 26);
+    p->fused = (logn > 12 && logn < 22);
+    p->fused2 = (logn >= 22 && logn <= 26);
     size_t half = n > 1 ? n / 2 : 1;
     hipError_t e = hipSuccess;
     auto mal = [&](void **ptr, size_t bytes) {
@@ -117,12 +117,13 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
             p->logN1 = (logn + 1) / 2;
             p->logN2 = logn / 2;
         } else {
-            // two-level: outer P1 rows of 2^12; inner four-step over the
-            // 2^13/2^14-long rows (M1 x M2, both <= 2^7)
-            p->logN1 = 12;
-            p->logN2 = logn - 12;
-            p->logM1 = (p->logN2 + 1) / 2;
-            p->logM2 = p->logN2 / 2;
+            // two-level: outer P1 rows of 2^(logn-16) (6..10 — through the
+            // batched small-row kernel, 4+ blocks/CU); inner four-step over
+            // the 2^16-long rows (M1 = M2 = 2^8)
+            p->logN1 = logn - 16;
+            p->logN2 = 16;
+            p->logM1 = 8;
+            p->logM2 = 8;
         }
         mal((void **)&p->d_work, n * sizeof(fe4));
         mal((void **)&p->d_twfull, n * sizeof(fe4));
@@ -286,8 +287,9 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256),
                            0, 0, cur, oth, N1, N2);
         HIP_TRY(hipEventRecord(p->ev[1], 0));
-        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512), 0, 0, oth,
-                           p->logN1, tr1, tf, (const fe9 *)nullptr);
+        hipLaunchKernelGGL(k_ntt_row_small, dim3(N2 * N1 / 1024), dim3(512),
+                           0, 0, oth, p->logN1, tr1, tf,
+                           (const fe9 *)nullptr, 0xffffffffu);
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256),
                            0, 0, oth, cur, N2, N1);
         // inner batched four-step over the N1 rows of length N2 = M1*M2
@@ -295,7 +297,7 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
                            dim3(256), 0, 0, cur, oth, M1, M2);
         hipLaunchKernelGGL(k_ntt_row_small, dim3(N1 * M2 * M1 / 1024),
                            dim3(512), 0, 0, oth, p->logM1, trA, tf2,
-                           (const fe9 *)nullptr, M2 - 1);
+                           (const fe9 *)nullptr, M2 - 1);  // (see P1 above)
         hipLaunchKernelGGL(k_transpose_fe4, dim3(M1 / 32, M2 / 32, N1),
                            dim3(256), 0, 0, oth, cur, M2, M1);
         hipLaunchKernelGGL(k_ntt_row_small, dim3(N1 * M1 * M2 / 1024),

@@ -474,7 +474,7 @@ static int msm_run_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     hipError_t e = hipSuccess;
     constexpr int DBITS = CFG::DBITS;  // per-window sort key bits
     const int nwin = FB ? FB_NWIN : CFG::NWIN;
-    if (!FB && p->n >= (1u << 18)) {
+    if (!FB && p->n >= (1u << 23)) {
         for (int w = 0; w < nwin && e == hipSuccess; w++) {
             size_t tmp = p->sort_tmp_bytes;
             size_t off = (size_t)w * p->n;
@@ -688,7 +688,7 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
                            p->d_keys, p->d_vals, p->n);
     }
     hipError_t e = hipSuccess;
-    if (!FB && p->n >= (1u << 18)) {
+    if (!FB && p->n >= (1u << 23)) {
         for (int w = 0; w < CFG::NWIN && e == hipSuccess; w++) {
             size_t tmp = p->sort_tmp_bytes;
             size_t off = (size_t)w * p->n;
