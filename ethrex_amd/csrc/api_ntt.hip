@@ -104,6 +104,10 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
     p->logn = logn;
     p->fused = (logn > 12 && logn < 22);
     p->fused2 = (logn >= 22 && logn <= 26);
+    if (std::getenv("EM_NTT_ONELEVEL") && logn <= 24 && p->fused2) {
+        p->fused = true;
+        p->fused2 = false;
+    }
     size_t half = n > 1 ? n / 2 : 1;
     hipError_t e = hipSuccess;
     auto mal = [&](void **ptr, size_t bytes) {
@@ -117,13 +121,19 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
             p->logN1 = (logn + 1) / 2;
             p->logN2 = logn / 2;
         } else {
-            // two-level: outer P1 rows of 2^(logn-16) (6..10 — through the
-            // batched small-row kernel, 4+ blocks/CU); inner four-step over
-            // the 2^16-long rows (M1 = M2 = 2^8)
+            // two-level: outer P1 rows of 2^logN1 (through the batched
+            // small-row kernel when logN1 <= 10, the dynamic-LDS row
+            // kernel for 11-12); inner four-step over the 2^logN2 rows.
+            // EM_NTT_SPLIT overrides logN1 for A/B measurement.
             p->logN1 = logn - 16;
-            p->logN2 = 16;
-            p->logM1 = 8;
-            p->logM2 = 8;
+            if (const char *e = std::getenv("EM_NTT_SPLIT")) {
+                int v = atoi(e);
+                if (v >= logn - 16 && v <= 12 && logn - v >= 10)
+                    p->logN1 = v;
+            }
+            p->logN2 = logn - p->logN1;
+            p->logM1 = (p->logN2 + 1) / 2;
+            p->logM2 = p->logN2 / 2;
         }
         mal((void **)&p->d_work, n * sizeof(fe4));
         mal((void **)&p->d_twfull, n * sizeof(fe4));
@@ -152,6 +162,12 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
         mal((void **)&p->d_tw, half * sizeof(fe9));
         mal((void **)&p->d_tw_inv, half * sizeof(fe9));
     }
+    // the 4096-element rows need 146 KiB of dynamic LDS (opt-in above 64K)
+    if (e == hipSuccess)
+        e = hipFuncSetAttribute(
+            (const void *)&k_ntt_row,
+            hipFuncAttributeMaxDynamicSharedMemorySize,
+            (int)((4096 + 64) * sizeof(fe9)));
     for (int i = 0; i < 4 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
     if (e != hipSuccess) {
         ethrex_mi355_ntt_plan_destroy(p);
@@ -253,7 +269,8 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
                            0, cur, oth, N1, N2);
         HIP_TRY(hipEventRecord(p->ev[1], 0));
         // P1: row NTT_N1 over each of the N2 rows + w^(k1*c) twiddle
-        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512), 0, 0, oth,
+        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512),
+                           (((size_t)1 << p->logN1) + (((size_t)1 << p->logN1) >> 6)) * sizeof(fe9), 0, oth,
                            p->logN1, inverse ? p->d_twrow1_inv : p->d_twrow1,
                            inverse ? p->d_twfull_inv : p->d_twfull,
                            (const fe9 *)nullptr);
@@ -261,7 +278,8 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256), 0,
                            0, oth, cur, N2, N1);
         // P2: row NTT_N2 (+ 1/n scale on iNTT)
-        hipLaunchKernelGGL(k_ntt_row, dim3(N1), dim3(512), 0, 0, cur,
+        hipLaunchKernelGGL(k_ntt_row, dim3(N1), dim3(512),
+                           (((size_t)1 << p->logN2) + (((size_t)1 << p->logN2) >> 6)) * sizeof(fe9), 0, cur,
                            p->logN2, inverse ? p->d_twrow2_inv : p->d_twrow2,
                            (const fe4 *)nullptr,
                            inverse ? p->d_ninv : (const fe9 *)nullptr);
@@ -287,9 +305,15 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256),
                            0, 0, cur, oth, N1, N2);
         HIP_TRY(hipEventRecord(p->ev[1], 0));
-        hipLaunchKernelGGL(k_ntt_row_small, dim3(N2 * N1 / 1024), dim3(512),
-                           0, 0, oth, p->logN1, tr1, tf,
-                           (const fe9 *)nullptr, 0xffffffffu);
+        if (p->logN1 <= 10) {
+            hipLaunchKernelGGL(k_ntt_row_small, dim3(N2 * N1 / 1024),
+                               dim3(512), 0, 0, oth, p->logN1, tr1, tf,
+                               (const fe9 *)nullptr, 0xffffffffu);
+        } else {
+            hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512),
+                               (((size_t)1 << p->logN1) + (((size_t)1 << p->logN1) >> 6)) * sizeof(fe9), 0, oth, p->logN1, tr1, tf,
+                               (const fe9 *)nullptr);
+        }
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256),
                            0, 0, oth, cur, N2, N1);
         // inner batched four-step over the N1 rows of length N2 = M1*M2

@@ -169,7 +169,9 @@ __global__ void __launch_bounds__(512)
 k_ntt_row(fe4 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
           const fe4 *__restrict__ tw2, const fe9 *__restrict__ scale,
           uint32_t cmask = 0xffffffffu) {
-    __shared__ fe9 smem[4096 + 64];
+    // dynamic LDS, (M + M/64) fe9 slots: 2048-element rows then run 2
+    // blocks/CU (73 KiB) instead of being pinned at the 4096-row footprint
+    extern __shared__ fe9 smem[];
     const uint32_t M = 1u << logM;
     fe4 *row = data + (size_t)blockIdx.x * M;
     for (uint32_t i = threadIdx.x; i < M; i += blockDim.x) {
