@@ -70,6 +70,25 @@ static int gen_tw_table(fe9 *d_out, size_t count, int bits, bool inverse,
     return EM_OK;
 }
 
+// row-kernel dispatch for the one-level path: default radix-2^3 at 512
+// threads; EM_NTT_R4=1 selects the radix-2^2 variant, EM_NTT_TD sets its
+// block size (512/768/1024) — A/B occupancy experiment.
+static void launch_row(uint32_t nrows, int logM, fe4 *data,
+                       const fe9 *tw_row, const fe4 *tw2, const fe9 *scale) {
+    size_t lds = (((size_t)1 << logM) + (((size_t)1 << logM) >> 6)) *
+                 sizeof(fe9);
+    static int r4 = std::getenv("EM_NTT_R4") ? 1 : 0;
+    static int td = std::getenv("EM_NTT_TD") ? atoi(std::getenv("EM_NTT_TD"))
+                                             : 1024;
+    if (r4) {
+        hipLaunchKernelGGL(k_ntt_row4, dim3(nrows), dim3(td), lds, 0, data,
+                           logM, tw_row, tw2, scale, 0xffffffffu);
+    } else {
+        hipLaunchKernelGGL(k_ntt_row, dim3(nrows), dim3(512), lds, 0, data,
+                           logM, tw_row, tw2, scale, 0xffffffffu);
+    }
+}
+
 // reordered streamed table TW2[c*M + k] = w^(k*c) (fe4m), c < rows:
 // P1's per-row twiddle read becomes a coalesced stream instead of a
 // stride-c gather over an n-sized table (a full cache line per element).
@@ -102,11 +121,15 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
     em_ntt_plan *p = new em_ntt_plan();
     p->n = n;
     p->logn = logn;
-    p->fused = (logn > 12 && logn < 22);
-    p->fused2 = (logn >= 22 && logn <= 26);
-    if (std::getenv("EM_NTT_ONELEVEL") && logn <= 24 && p->fused2) {
-        p->fused = true;
-        p->fused2 = false;
+    // one-level four-step up to 2^24 (A/B-measured best: 3.37 ms vs 3.76
+    // for the two-level split at 2^24); two-level only where the row
+    // length exceeds the LDS row kernel (2^25/26).  EM_NTT_TWOLEVEL
+    // forces the two-level path for measurement.
+    p->fused = (logn > 12 && logn <= 24);
+    p->fused2 = (logn >= 25 && logn <= 26);
+    if (std::getenv("EM_NTT_TWOLEVEL") && logn >= 22 && p->fused) {
+        p->fused = false;
+        p->fused2 = true;
     }
     size_t half = n > 1 ? n / 2 : 1;
     hipError_t e = hipSuccess;
@@ -125,7 +148,7 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
             // small-row kernel when logN1 <= 10, the dynamic-LDS row
             // kernel for 11-12); inner four-step over the 2^logN2 rows.
             // EM_NTT_SPLIT overrides logN1 for A/B measurement.
-            p->logN1 = logn - 16;
+            p->logN1 = logn >= 25 ? 12 : logn - 16;
             if (const char *e = std::getenv("EM_NTT_SPLIT")) {
                 int v = atoi(e);
                 if (v >= logn - 16 && v <= 12 && logn - v >= 10)
@@ -166,6 +189,11 @@ extern "C" int ethrex_mi355_ntt_plan_create(size_t n, em_ntt_plan **plan) {
     if (e == hipSuccess)
         e = hipFuncSetAttribute(
             (const void *)&k_ntt_row,
+            hipFuncAttributeMaxDynamicSharedMemorySize,
+            (int)((4096 + 64) * sizeof(fe9)));
+    if (e == hipSuccess)
+        e = hipFuncSetAttribute(
+            (const void *)&k_ntt_row4,
             hipFuncAttributeMaxDynamicSharedMemorySize,
             (int)((4096 + 64) * sizeof(fe9)));
     for (int i = 0; i < 4 && e == hipSuccess; i++) e = hipEventCreate(&p->ev[i]);
@@ -269,20 +297,18 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
                            0, cur, oth, N1, N2);
         HIP_TRY(hipEventRecord(p->ev[1], 0));
         // P1: row NTT_N1 over each of the N2 rows + w^(k1*c) twiddle
-        hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512),
-                           (((size_t)1 << p->logN1) + (((size_t)1 << p->logN1) >> 6)) * sizeof(fe9), 0, oth,
-                           p->logN1, inverse ? p->d_twrow1_inv : p->d_twrow1,
-                           inverse ? p->d_twfull_inv : p->d_twfull,
-                           (const fe9 *)nullptr);
+        launch_row(N2, p->logN1,
+                   oth, inverse ? p->d_twrow1_inv : p->d_twrow1,
+                   inverse ? p->d_twfull_inv : p->d_twfull,
+                   (const fe9 *)nullptr);
         // T1
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256), 0,
                            0, oth, cur, N2, N1);
         // P2: row NTT_N2 (+ 1/n scale on iNTT)
-        hipLaunchKernelGGL(k_ntt_row, dim3(N1), dim3(512),
-                           (((size_t)1 << p->logN2) + (((size_t)1 << p->logN2) >> 6)) * sizeof(fe9), 0, cur,
-                           p->logN2, inverse ? p->d_twrow2_inv : p->d_twrow2,
-                           (const fe4 *)nullptr,
-                           inverse ? p->d_ninv : (const fe9 *)nullptr);
+        launch_row(N1, p->logN2,
+                   cur, inverse ? p->d_twrow2_inv : p->d_twrow2,
+                   (const fe4 *)nullptr,
+                   inverse ? p->d_ninv : (const fe9 *)nullptr);
         // T2: natural order
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N2 / 32, N1 / 32), dim3(256), 0,
                            0, cur, oth, N1, N2);

@@ -299,6 +299,70 @@ k_ntt_row(fe4 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
     }
 }
 
+
+// radix-2^2 row variant (A/B occupancy experiment, EM_NTT_R4): 4 elements
+// per thread per round trip, launchable at 512/768/1024 threads — more
+// waves/SIMD than the radix-2^3 shape at the cost of 1.5x the LDS round
+// trips.  Same skewed layout and IO as k_ntt_row.
+__global__ void __launch_bounds__(1024)
+k_ntt_row4(fe4 *__restrict__ data, int logM, const fe9 *__restrict__ tw_row,
+           const fe4 *__restrict__ tw2, const fe9 *__restrict__ scale,
+           uint32_t cmask) {
+    extern __shared__ fe9 smem[];
+    const uint32_t M = 1u << logM;
+    fe4 *row = data + (size_t)blockIdx.x * M;
+    for (uint32_t i = threadIdx.x; i < M; i += blockDim.x) {
+        uint32_t j = __brev(i) >> (32 - logM);
+        smem[EM_SK(j)] = fe4m_unpack(row[i]);
+    }
+    __syncthreads();
+    int s = 1;
+    for (; s + 1 <= logM; s += 2) {
+        uint32_t q = 1u << (s - 1);
+        for (uint32_t t = threadIdx.x; t < (M >> 2); t += blockDim.x) {
+            uint32_t j = t & (q - 1);
+            uint32_t idx = ((t >> (s - 1)) << (s + 1)) + j;
+            fe9 w1 = tw_row[j << (logM - s)];
+            fe9 a = smem[EM_SK(idx)];
+            fe9 b = mont_mul9<Fr9T>(smem[EM_SK(idx + q)], w1);
+            fe9 c = smem[EM_SK(idx + 2 * q)];
+            fe9 d = mont_mul9<Fr9T>(smem[EM_SK(idx + 3 * q)], w1);
+            fe9 t0 = add9_n<Fr9T>(a, b);
+            fe9 t1 = subm9<Fr9T>(a, b);
+            fe9 t2 = add9_n<Fr9T>(c, d);
+            fe9 t3 = subm9<Fr9T>(c, d);
+            fe9 u2 = mont_mul9<Fr9T>(t2, tw_row[j << (logM - s - 1)]);
+            fe9 u3 = mont_mul9<Fr9T>(t3, tw_row[(j + q) << (logM - s - 1)]);
+            smem[EM_SK(idx)] = add9_n<Fr9T>(t0, u2);
+            smem[EM_SK(idx + 2 * q)] = subm9<Fr9T>(t0, u2);
+            smem[EM_SK(idx + q)] = add9_n<Fr9T>(t1, u3);
+            smem[EM_SK(idx + 3 * q)] = subm9<Fr9T>(t1, u3);
+        }
+        __syncthreads();
+    }
+    for (; s <= logM; s++) {
+        uint32_t half = 1u << (s - 1);
+        for (uint32_t t = threadIdx.x; t < (M >> 1); t += blockDim.x) {
+            uint32_t j = t & (half - 1);
+            uint32_t idx = ((t >> (s - 1)) << s) + j;
+            fe9 u = smem[EM_SK(idx)];
+            fe9 v = mont_mul9<Fr9T>(smem[EM_SK(idx + half)],
+                                    tw_row[j << (logM - s)]);
+            smem[EM_SK(idx)] = add9_n<Fr9T>(u, v);
+            smem[EM_SK(idx + half)] = subm9<Fr9T>(u, v);
+        }
+        __syncthreads();
+    }
+    uint64_t c = blockIdx.x & cmask;
+    const fe4 *t2row = tw2 ? tw2 + (size_t)c * M : nullptr;
+    for (uint32_t k = threadIdx.x; k < M; k += blockDim.x) {
+        fe9 x = smem[EM_SK(k)];
+        if (t2row) x = mont_mul9<Fr9T>(x, fe4m_unpack(t2row[k]));
+        if (scale) x = mont_mul9<Fr9T>(x, *scale);
+        row[k] = fe4m_pack(x);
+    }
+}
+
 // packed small-row NTT: 1024/M rows of length M <= 512 per block (the
 // 144-KB k_ntt_row runs one block/CU and would idle 1-(M/4096) of its
 // threads on the two-level path's 64/128-point inner rows).  One butterfly
