@@ -77,7 +77,11 @@ static void launch_row(uint32_t nrows, int logM, fe4 *data,
                        const fe9 *tw_row, const fe4 *tw2, const fe9 *scale) {
     size_t lds = (((size_t)1 << logM) + (((size_t)1 << logM) >> 6)) *
                  sizeof(fe9);
-    static int r4 = std::getenv("EM_NTT_R4") ? 1 : 0;
+    // default: radix-2^2 at 1024 threads (A/B best: 3.28 ms vs 3.38 for
+    // the radix-2^3/512 shape at 2^24 — 4 waves/SIMD hides the LDS+mul
+    // latency better than fewer round trips at 2 waves).  EM_NTT_R8
+    // selects the radix-2^3 variant.
+    static int r4 = std::getenv("EM_NTT_R8") ? 0 : 1;
     static int td = std::getenv("EM_NTT_TD") ? atoi(std::getenv("EM_NTT_TD"))
                                              : 1024;
     if (r4) {
