@@ -216,10 +216,12 @@ def main():
         "unit": "point_adds/s",
         "frac": kernel_adds_per_s / valu_peak_adds,
         "inst_per_add": INST_PER_ADD,
-        # measured per-launch HBM bytes (rocprofv3 FETCH_SIZE+WRITE_SIZE)
-        # for the default 2^24 1-GPU config; null for other shapes until
-        # measured.  See profiles/ for the PMC passes.
-        "traffic": None,
+        # measured per-launch HBM bytes (rocprofv3 FETCH_SIZE+WRITE_SIZE,
+        # profiles/r02_summary.md PMC pass; 1.5x gather overfetch from
+        # random 72-B point reads vs 128-B lines) for the default 2^24
+        # 1-GPU config; null for other shapes
+        "traffic": (28.95e9 if (args.msm_log2 == 24 and shard == n_total)
+                    else None),
         "hbm_secondary": {
             "achieved": alg_bytes / (avg_bucket_ms / 1000.0),
             "peak": hbm_peak,
@@ -266,7 +268,11 @@ def main():
                 "peak": hbm_peak,
                 "unit": "B/s",
                 "frac": (ntt_alg_bytes / (avg_total_ms / 1000.0)) / hbm_peak,
-                "traffic": None,
+                # measured per-transform bytes at 2^24 (PMC: rows and
+                # transposes move exactly the algorithmic bytes — zero
+                # overfetch; the ~42% gap to wall time is the row kernels'
+                # compute/latency, see profiles/r02_summary.md)
+                "traffic": 5.54e9 if args.ntt_log2 == 24 else None,
             },
         }
         nplan.destroy()

@@ -340,9 +340,7 @@ extern "C" int ethrex_mi355_ntt_run(em_ntt_plan *p, int inverse) {
                                dim3(512), 0, 0, oth, p->logN1, tr1, tf,
                                (const fe9 *)nullptr, 0xffffffffu);
         } else {
-            hipLaunchKernelGGL(k_ntt_row, dim3(N2), dim3(512),
-                               (((size_t)1 << p->logN1) + (((size_t)1 << p->logN1) >> 6)) * sizeof(fe9), 0, oth, p->logN1, tr1, tf,
-                               (const fe9 *)nullptr);
+            launch_row(N2, p->logN1, oth, tr1, tf, (const fe9 *)nullptr);
         }
         hipLaunchKernelGGL(k_transpose_fe4, dim3(N1 / 32, N2 / 32), dim3(256),
                            0, 0, oth, cur, N2, N1);

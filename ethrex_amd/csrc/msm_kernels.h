@@ -178,8 +178,9 @@ static __global__ void k_digits(const fe4 *__restrict__ scalars,
                 carry = 0;
             }
             bool sk = skip || mag == 0;
+            uint32_t id = sk ? 0u : mag - 1;
             keys[(size_t)w * n + i] =
-                ((uint32_t)w << CFG::IB) | (sk ? 0u : mag - 1);
+                (KT)(K16 ? id : (((uint32_t)w << CFG::IB) | id));
             vals[(size_t)w * n + i] =
                 (uint32_t)i | (sign << 31) | (sk ? SGN_SKIP : 0u);
         }
@@ -190,7 +191,8 @@ static __global__ void k_digits(const fe4 *__restrict__ scalars,
         for (int w = 0; w < CFG::NWIN; w++) {
             uint32_t d = msm_digit<CFG::C>(k, w);
             if (skip) d = 0;  // identity points: park in bucket 0
-            keys[(size_t)w * n + i] = ((uint32_t)w << CFG::C) | d;
+            keys[(size_t)w * n + i] =
+                (KT)(K16 ? d : (((uint32_t)w << CFG::C) | d));
             vals[(size_t)w * n + i] = (uint32_t)i;
         }
     }
