@@ -155,11 +155,14 @@ static __global__ void k_parse_scalars(const uint8_t *__restrict__ in,
 // (skipped by the walk).  SIGNED: balanced recode with carry — window value
 // t = raw + carry_in; t <= 2^(C-1) keeps d = +t, else d = t - 2^C (carry 1);
 // bucket id = |d| - 1 (zero digits / infinities park at id 0 with SGN_SKIP).
-template <typename CFG>
+template <typename CFG, typename KT = uint32_t>
 static __global__ void k_digits(const fe4 *__restrict__ scalars,
                          const uint8_t *__restrict__ inf,
-                         uint32_t *__restrict__ keys, uint32_t *__restrict__ vals,
+                         KT *__restrict__ keys, uint32_t *__restrict__ vals,
                          size_t n) {
+    // u16 keys: in-window id only (window implicit by segment; 25% less
+    // sort traffic).  u32 keys: (w << IB/C) | id, globally sortable.
+    constexpr bool K16 = sizeof(KT) == 2;
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     fe4 k = scalars[i];
@@ -196,6 +199,32 @@ static __global__ void k_digits(const fe4 *__restrict__ scalars,
             vals[(size_t)w * n + i] = (uint32_t)i;
         }
     }
+}
+
+// segment-local offsets for u16 per-window keys: bucket b = (w, id); its
+// run is the lower_bound of id within the window's segment [w*n, (w+1)*n)
+template <typename CFG>
+static __global__ void k_offsets_seg(const uint16_t *__restrict__ sorted_keys,
+                                     size_t n /* per window */,
+                                     uint32_t *__restrict__ offsets) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b > CFG::NBUCKETS) return;
+    if (b == CFG::NBUCKETS) {
+        offsets[b] = (uint32_t)(n * CFG::NWIN);
+        return;
+    }
+    uint32_t w = b >> CFG::IB;
+    uint16_t id = (uint16_t)(b & CFG::DMASK);
+    const uint16_t *seg = sorted_keys + (size_t)w * n;
+    size_t lo = 0, hi = n;
+    while (lo < hi) {
+        size_t mid = (lo + hi) >> 1;
+        if (seg[mid] < id)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    offsets[b] = (uint32_t)((size_t)w * n + lo);
 }
 
 // ---- bucket segment offsets: lower_bound of each bucket id ----
