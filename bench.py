@@ -276,6 +276,20 @@ def main():
             },
         }
         nplan.destroy()
+        # secondary named config (BASELINE configs[2]): 2^22-element NTT
+        if args.ntt_log2 == 24:
+            m2 = 1 << 22
+            np2 = ethrex_amd.NttPlan(m2)
+            np2.upload(ethrex_amd.gen_fr(44, m2))
+            for _ in range(max(args.warmup, 1)):
+                np2.run(False)
+            t2 = time.perf_counter()
+            for _ in range(args.steps):
+                np2.run(False)
+            dt2 = (time.perf_counter() - t2) / args.steps
+            ntt["ntt_2_22"] = {"value": m2 / dt2, "ms": dt2 * 1000.0,
+                               "n": m2}
+            np2.destroy()
 
     phase_ms = {k: round(v, 3) for k, v in plan.last_times().items()}
 
