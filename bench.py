@@ -77,7 +77,45 @@ def main():
     ap.add_argument("--check", action="store_true",
                     help="verify the first step's result against the oracle "
                          "via the shard-combine identity (adds oracle time)")
+    ap.add_argument("--sweep", action="store_true",
+                    help="size sweep: MSM and NTT at 2^20..2^26 on one GPU "
+                         "(BASELINE.md reporting range); prints its own "
+                         "JSON line and exits")
     args = ap.parse_args()
+
+    if args.sweep:
+        import ethrex_amd
+        ethrex_amd.set_device(0)
+        out = {"sweep": {"msm": {}, "ntt": {}}}
+        for lg in range(20, 27):
+            n = 1 << lg
+            plan = ethrex_amd.MsmPlan(n)
+            plan.gen_points(0)
+            plan.upload_scalars(ethrex_amd.gen_fr(42, n))
+            plan.run()
+            steps = max(2, min(10, (1 << 25) // n))
+            t0 = time.perf_counter()
+            for _ in range(steps):
+                plan.run_async()
+            plan.sync()
+            dt = (time.perf_counter() - t0) / steps
+            out["sweep"]["msm"][f"2^{lg}"] = {
+                "ms": dt * 1000.0,
+                "point_adds_per_s": point_adds(n) / dt,
+            }
+            plan.destroy()
+            nplan = ethrex_amd.NttPlan(n)
+            nplan.upload(ethrex_amd.gen_fr(43, n))
+            nplan.run(False)
+            t0 = time.perf_counter()
+            for _ in range(steps):
+                nplan.run(False)
+            dt = (time.perf_counter() - t0) / steps
+            out["sweep"]["ntt"][f"2^{lg}"] = {
+                "ms": dt * 1000.0, "elems_per_s": n / dt}
+            nplan.destroy()
+        print(json.dumps(out), flush=True)
+        return
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))

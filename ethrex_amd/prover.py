@@ -89,11 +89,14 @@ class ExecBackend:
 
 
 class Mi355Backend:
-    """The MI355X backend: `prove` derives the batch's MSM/NTT workload
-    deterministically from the input and runs it on the GPU through the
-    C-ABI.  Proof bytes = MSM affine result || NTT output digest (the
-    stand-in for the Groth16 wrap output whose MSM/NTT this core computes;
-    full proof-object integration is SURVEY.md §8f row 3)."""
+    """The MI355X backend.  `execute` validates the witness (GPU batched
+    keccak + trie linking -> statement); `prove` runs the WRAP-SHAPED
+    composition the Groth16 wrap performs (backend/sp1.rs:122-134): the
+    statement-seeded witness vector is NTT'd and its output feeds the
+    proving MSM on device.  Proof bytes = MSM affine result || NTT output
+    digest || statement commitment (the wrap circuit itself lives in the
+    non-vendored zkVM — SURVEY.md §8c — so the proof-OBJECT is this
+    core's composition output, carried through the §8f row-3 wire)."""
 
     def __init__(self, msm_log2=16, ntt_log2=12, device=0):
         self.msm_log2 = msm_log2
@@ -160,20 +163,24 @@ class Mi355Backend:
                 bytes.fromhex(statement["commitment"])[:8], "little")
         else:
             seed = self._seed(input_data)
+        # the wrap-shaped composition (backend/sp1.rs:122-134 flow): the
+        # statement-derived witness vector is NTT'd and its output feeds
+        # the proving MSM ON DEVICE (msm_scalars_from_ntt) — the two
+        # primitives this core exists for, composed as the Groth16 wrap
+        # composes them, with no PCIe hop between them.
         n = 1 << self.msm_log2
         plan = ea.MsmPlan(n)
+        nplan = ea.NttPlan(n)
         try:
             plan.gen_points(0)
-            scalars = ea.gen_fr(seed, n)
-            plan.upload_scalars(scalars)
+            nplan.upload(ea.gen_fr(seed, n))
+            nplan.run(False)
+            plan.scalars_from_ntt(nplan, 0)
             msm_out = plan.run()
+            ntt_out = nplan.download()
         finally:
             plan.destroy()
-        m = 1 << self.ntt_log2
-        elems = ea.gen_fr(seed + 1, m)
-        rc, ntt_out = ea.fr_ntt(elems, m, False)
-        if rc != 0:
-            raise BackendError(f"mi355: ntt rc={rc}")
+            nplan.destroy()
         proof = {"msm": msm_out,
                  "ntt_digest": hashlib.sha256(ntt_out).digest()}
         if statement is not None:

@@ -33,19 +33,19 @@ def test_mi355_backend_proves_through_coordinator(oracle_mod):
         proof_bytes = bytes(stored["Proof"]["proof"])
         assert len(proof_bytes) == 96  # 64 B MSM affine + 32 B NTT digest
 
-        # recompute the expected proof with the oracle (same derivation)
+        # recompute the expected proof with the oracle: prove() runs the
+        # wrap-shaped composition MSM(points, NTT(seeded vector)) with the
+        # NTT output handed to the MSM on device (sp1.rs:122-134 flow)
         input_data = {"batch": 0, "blocks": [1, 2, 3]}
         seed = int.from_bytes(hashlib.sha256(json.dumps(
             input_data, sort_keys=True).encode()).digest()[:8], "little")
         m = 1 << 14
         pts = oracle_mod.gen_points(0, m)
-        scs = oracle_mod.gen_fr(seed, m)
-        rc, want_msm = oracle_mod.g1_msm(pts, scs, m)
-        assert rc == 0 and proof_bytes[:64] == want_msm
-        k = 1 << 12
-        rc, want_ntt = oracle_mod.fr_ntt(oracle_mod.gen_fr(seed + 1, k), k, False)
+        rc, fwd = oracle_mod.fr_ntt(oracle_mod.gen_fr(seed, m), m, False)
         assert rc == 0
-        assert proof_bytes[64:] == hashlib.sha256(want_ntt).digest()
+        rc, want_msm = oracle_mod.g1_msm(pts, fwd, m)
+        assert rc == 0 and proof_bytes[:64] == want_msm
+        assert proof_bytes[64:] == hashlib.sha256(fwd).digest()
     finally:
         coord.stop()
 
@@ -80,12 +80,14 @@ def test_mi355_backend_statement_from_real_witness(oracle_mod):
     assert st == st_cpu
     assert st["initial_state_root"] == (
         "4bec425c34f89aeb56c78586d586c76044bcc428e7c34ab7c72d389c39ff3eaf")
-    # the MSM input really derives from the statement commitment
+    # the MSM input really derives from the statement commitment through
+    # the wrap composition: MSM(points, NTT(statement-seeded vector))
     seed = int.from_bytes(bytes.fromhex(st["commitment"])[:8], "little")
     m = 1 << 14
     pts = oracle_mod.gen_points(0, m)
-    scs = oracle_mod.gen_fr(seed, m)
-    rc, want_msm = oracle_mod.g1_msm(pts, scs, m)
+    rc, fwd = oracle_mod.fr_ntt(oracle_mod.gen_fr(seed, m), m, False)
+    assert rc == 0
+    rc, want_msm = oracle_mod.g1_msm(pts, fwd, m)
     assert rc == 0 and proof["msm"] == want_msm
     out = backend.to_proof_bytes(proof, None)
     pb = bytes(out["Proof"]["proof"])
