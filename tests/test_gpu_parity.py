@@ -451,3 +451,19 @@ def test_gpu_ntt_2_26_sparse_dft_spot_check(gpu):
         want = sum(v * pow(w, (i * j) % n, R) for i, v in nz.items()) % R
         got = int.from_bytes(fwd[32 * j:32 * (j + 1)], "big")
         assert got == want, f"A[{j}] mismatch"
+
+
+def test_gpu_ntt_parity_2_22_direct(gpu, oracle_mod):
+    """Direct oracle parity at the named BASELINE config (2^22-element
+    NTT, configs[2]) — forward and inverse."""
+    n = 1 << 22
+    elems = gpu.gen_fr(44, n)
+    plan = gpu.NttPlan(n)
+    plan.upload(elems)
+    plan.run(False)
+    fwd = plan.download()
+    rc, want = oracle_mod.fr_ntt(elems, n, False)
+    assert rc == 0 and fwd == want
+    plan.run(True)
+    assert plan.download() == elems
+    plan.destroy()
