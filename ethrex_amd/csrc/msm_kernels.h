@@ -1,21 +1,29 @@
 // ============================================================================
 // Pippenger bucket MSM — gfx950 kernels, curve- and geometry-templated
-// (BN254 G1 and BLS12-381 G1 on the 29-bit-limb field core).
+// (BN254 G1, BLS12-381 G1 and G2-over-Fp2 on the 29-bit-limb field core).
 //
-// Pipeline per run (window bits CB from msm_cfg; c=16 for large MSMs,
-// c=8 for n <= 2^16, FB_C=13 fixed-base for blob-KZG):
+// Pipeline per run (window bits CB from msm_cfg; SIGNED c=17 for large
+// BN254 MSMs, unsigned c=16 for large BLS, c=8 for n <= 2^16, FB_C=13
+// fixed-base for blob-KZG):
 //   1. parse scalars (BN254: ark from_be_bytes_mod_order reduction;
 //      BLS: raw 256-bit per blst — both as packed u64[4] for digits)
-//   2. digit extraction -> (key = window<<CB | digit, value = point index)
-//   3. device radix sort of the NWIN*n pairs (rocPRIM)
+//   2. digit extraction -> (key, value = point index [+ sign/skip bits
+//      for signed configs]); keys are u16 in-window ids for the
+//      per-window-sort path (n >= 2^23), u32 (window<<IB | id) otherwise
+//   3. rocPRIM radix sorts: one per window over just the id bits (large
+//      n), or one global sort (small n — ~60 fewer host launches/step)
 //   4. per-bucket offsets by binary search + LENGTH-SORTED bucket schedule
 //      (waves then process similar-length runs: no Poisson divergence)
 //   5. bucket accumulation: one thread per scheduled bucket walks its run
-//      with XYZZ+affine mixed adds — the VALU-bound hot kernel
-//   6. segment running sums -> weighted reduce (folds the 2^(CB*w) window
-//      factor SIMD-wide) -> per-window LDS trees -> combine + affine.
+//      with XYZZ+affine mixed adds (negating y for negative digits) —
+//      the VALU-issue-bound hot kernel (saturated at 3 waves/SIMD:
+//      profiles/r02_summary.md)
+//   6. segment running sums -> weighted reduce -> per-window LDS trees ->
+//      k_emit_windows.  Windows leave UNSCALED; the host folds the
+//      2^(CB*w) factors (Horner) at delivery, overlapped with the next
+//      pipelined step's GPU work.
 //
-// Work shape: ~NWIN*(n + 2^(CB+1)) mixed adds; HBM traffic is only the
+// Work shape: ~NWIN*(n + 2^(IB+1)) mixed adds; HBM traffic is only the
 // gathered points + sorted pairs => VALU-bound, so there
 // is deliberately no MFMA anywhere here.
 // ============================================================================

@@ -126,3 +126,45 @@ def test_account_leaf_matches_witness_accounts(wfix):
         assert trie.account_leaf(nonce, balance, acct[2], acct[3]) == v
         checked += 1
     assert checked >= 40
+
+
+def test_witness_missing_root_raises(wfix):
+    """A witness whose parent state root is absent from the node map must
+    fail with RootNotFound (get_embedded_root_committed semantics)."""
+    state, headers, fbn, node_map, _ = wfix
+    with pytest.raises(ValueError, match="RootNotFound"):
+        witness.link_trie(b"\x11" * 32, dict(node_map))
+
+
+def test_witness_tolerates_unused_and_malformed_nodes():
+    """EELS test_validation_state_extra_unused_trie_node semantics: extra
+    undecodable entries and the 0x80 sentinel are skipped, not rejected
+    (block_execution_witness.rs:214-231)."""
+    # a tiny self-consistent trie: one leaf node as root
+    leaf = rlp.encode([b"\x20" + b"\xab" * 32, b"\x01" * 40])
+    root = keccak256(leaf)
+    state = [leaf, b"\x80", b"\xff\xff\xff", b"garbage-not-rlp",
+             rlp.encode(b"just-a-string")]
+    node_map, skipped = witness.build_node_map(state,
+                                               witness.cpu_hash_batch)
+    assert skipped == 4  # sentinel + 2 undecodable + non-list RLP
+    st = witness.link_trie(root, node_map)
+    assert st.n_nodes == 1 and st.leaf_values == [b"\x01" * 40]
+
+
+def test_witness_missing_interior_nodes_counted():
+    """A partial trie (the normal witness case) links what is present and
+    counts absent hash references instead of failing."""
+    leaf = rlp.encode([b"\x20" + b"\xcd" * 32, b"\x02" * 40])
+    lh = keccak256(leaf)
+    branch = rlp.encode([lh, b"\x33" * 32] + [b""] * 14 + [b""])
+    root = keccak256(branch)
+    node_map, _ = witness.build_node_map([branch, leaf],
+                                         witness.cpu_hash_batch)
+    st = witness.link_trie(root, node_map)
+    assert st.n_nodes == 2 and st.missing == 1
+
+
+def test_empty_state_trie_root():
+    st = witness.link_trie(witness.EMPTY_TRIE_HASH, {})
+    assert st.n_nodes == 0 and st.missing == 0
