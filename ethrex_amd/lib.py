@@ -204,9 +204,15 @@ class MsmPlan:
         PCIe; the sp1.rs:122-134 wrap flow feeds the witness NTT output
         into the proving MSM)."""
         ptr = ctypes.c_void_p()
+        nn = ctypes.c_size_t(0)
         _check(_lib.ethrex_mi355_ntt_device_data(ntt_plan._p,
-                                                 ctypes.byref(ptr), None),
+                                                 ctypes.byref(ptr),
+                                                 ctypes.byref(nn)),
                "ntt_device_data")
+        if offset + self.n > nn.value:
+            raise HipCoreError(EM_ERR_INPUT,
+                               f"scalars_from_ntt: offset {offset} + n "
+                               f"{self.n} exceeds NTT size {nn.value}")
         _check(_lib.ethrex_mi355_msm_scalars_from_ntt(
             self._p, ptr, ctypes.c_uint64(offset)), "msm_scalars_from_ntt")
 
