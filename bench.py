@@ -147,7 +147,8 @@ def main():
 
     ethrex_amd.set_device(local_rank)
 
-    from ethrex_amd.dist import allgather_partials, shard_range
+    from ethrex_amd.dist import (allgather_partials,
+                                 pipelined_shard_steps, shard_range)
 
     n_total = 1 << args.msm_log2
     lo, hi = shard_range(n_total, n_gpus, rank)
@@ -195,21 +196,11 @@ def main():
             plan.run_async()
         pipelined_last = plan.sync()
     else:
-        # pipelined shard loop: enqueue step k, then deliver step k-1's
-        # Jacobian partial (wait_one leaves step k running), AllGather it
-        # over RCCL/xGMI and combine on the host — the exchange of step k-1
-        # fully overlaps the GPU compute of step k.
-        plan.run_partial_async()
-        for _ in range(1, args.steps):
-            plan.run_partial_async()
-            part = plan.wait_one()
-            allparts = allgather_partials(part, dist, device="cpu",
-                                      group=gloo_pg)
-            pipelined_last = ethrex_amd.g1_combine_cpu(allparts, world)
-        part = plan.wait_one()
-        allparts = allgather_partials(part, dist, device="cpu",
-                                      group=gloo_pg)
-        pipelined_last = ethrex_amd.g1_combine_cpu(allparts, world)
+        # pipelined shard loop (ethrex_amd/dist.py pipelined_shard_steps;
+        # control flow pinned by the world-2 gloo test): the exchange of
+        # step k-1 fully overlaps the GPU compute of step k
+        pipelined_last = pipelined_shard_steps(plan, args.steps, dist,
+                                               world, group=gloo_pg)
     if world > 1:
         dist.barrier()
         torch.cuda.synchronize()

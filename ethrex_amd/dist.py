@@ -17,6 +17,37 @@ def shard_range(n_total: int, world: int, rank: int):
     return lo, hi
 
 
+def pipelined_shard_steps(plan, steps, tdist, world, group=None,
+                          combine=None, collect=None):
+    """The N>1 timed loop (bench.py): enqueue step k, deliver step k-1's
+    Jacobian partial (plan.wait_one leaves step k running on the GPU),
+    AllGather it and combine on the host — the exchange of step k-1
+    fully overlaps the GPU compute of step k.
+
+    `plan` needs run_partial_async() and wait_one(); `combine(bytes,
+    world) -> bytes` defaults to the host combine.  Returns the last
+    step's combined result; appends every step's result to `collect`
+    when given (tests)."""
+    if combine is None:
+        from .lib import g1_combine_cpu
+        combine = g1_combine_cpu
+    last = None
+    plan.run_partial_async()
+    for _ in range(1, steps):
+        plan.run_partial_async()
+        part = plan.wait_one()
+        allp = allgather_partials(part, tdist, device="cpu", group=group)
+        last = combine(allp, world)
+        if collect is not None:
+            collect.append(last)
+    part = plan.wait_one()
+    allp = allgather_partials(part, tdist, device="cpu", group=group)
+    last = combine(allp, world)
+    if collect is not None:
+        collect.append(last)
+    return last
+
+
 def allgather_partials(partial96: bytes, tdist, device="cpu", group=None):
     """AllGather each rank's 96-B Jacobian partial; returns concatenated
     world*96 bytes in rank order.
