@@ -88,6 +88,11 @@ struct msm_plan_t {
     uint8_t *d_out2 = nullptr;
     uint8_t *h_out[2] = {nullptr, nullptr};   // pinned window staging
     hipEvent_t ev_sort_done[2], ev_comp_done[2];
+    // contention gating: the bucket walk launches as two segments; the
+    // NEXT step's sort chain (HBM+issue-hungry rocPRIM kernels) waits for
+    // segment A (~80% of the work), so sorts contend only with the walk's
+    // tail instead of its whole duration (async step 27.9 -> see ledger)
+    hipEvent_t ev_walkA[2];
     // a pending pipelined result: h_out[par] holds nwin UNSCALED Jacobian
     // window records; delivery runs the host Horner (cbits doublings + 1
     // add per window) and writes affine (out_mode 0) or Jacobian (1) bytes
@@ -325,8 +330,12 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
         if (e == hipSuccess)
             e = hipEventCreateWithFlags(&p->ev_comp_done[i],
                                         hipEventDisableTiming);
+        if (e == hipSuccess)
+            e = hipEventCreateWithFlags(&p->ev_walkA[i],
+                                        hipEventDisableTiming);
         // record once so the first hipStreamWaitEvent sees a signaled event
         if (e == hipSuccess) e = hipEventRecord(p->ev_comp_done[i], 0);
+        if (e == hipSuccess) e = hipEventRecord(p->ev_walkA[i], 0);
     }
     if (e != hipSuccess) {
         msm_destroy_t(p);
@@ -703,6 +712,8 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     hipStream_t ss = p->s_sort, sc = p->s_comp;
     // ---- sort chain ----
     HIP_TRY(hipStreamWaitEvent(ss, p->ev_comp_done[par], 0));
+    // contention gate: wait for the PREVIOUS step's walk segment A
+    HIP_TRY(hipStreamWaitEvent(ss, p->ev_walkA[par ^ 1], 0));
     if constexpr (FB) {
         hipLaunchKernelGGL(k_fb_digits, dim3(blocks_for(total, 256)),
                            dim3(256), 0, ss, p->d_scalars, p->d_inf,
@@ -756,9 +767,42 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     HIP_TRY(hipEventRecord(p->ev_sort_done[par], ss));
     // ---- compute chain ----
     HIP_TRY(hipStreamWaitEvent(sc, p->ev_sort_done[par], 0));
-    hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
-                       dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, sc,
-                       pts, VO, OFF, SCH, p->d_buckets);
+    // two walk segments over the length-sorted schedule (lengths are
+    // near-Poisson-uniform, so a count split ~= a work split); the next
+    // step's sorts start after segment A.  EM_MSM_SPLIT = A's percent.
+    {
+        static int spct = std::getenv("EM_MSM_SPLIT")
+                              ? atoi(std::getenv("EM_MSM_SPLIT"))
+                              : 80;
+        uint32_t splitA =
+            (uint32_t)(((uint64_t)CFG::NBUCKETS *
+                        (uint32_t)(spct > 0 ? spct : 0)) / 100);
+        splitA = (splitA / 256) * 256;  // block-aligned
+        if (spct <= 0) {
+            // gate disabled: signal segment A immediately (sorts overlap
+            // the whole walk — the pre-gating behavior)
+            HIP_TRY(hipEventRecord(p->ev_walkA[par], sc));
+            hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
+                               dim3(blocks_for(CFG::NBUCKETS, 256)),
+                               dim3(256), 0, sc, pts, VO, OFF, SCH,
+                               p->d_buckets, 0u);
+        } else if (splitA == 0 || splitA >= CFG::NBUCKETS) {
+            hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
+                               dim3(blocks_for(CFG::NBUCKETS, 256)),
+                               dim3(256), 0, sc, pts, VO, OFF, SCH,
+                               p->d_buckets, 0u);
+            HIP_TRY(hipEventRecord(p->ev_walkA[par], sc));
+        } else {
+            hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
+                               dim3(blocks_for(splitA, 256)), dim3(256), 0,
+                               sc, pts, VO, OFF, SCH, p->d_buckets, 0u);
+            HIP_TRY(hipEventRecord(p->ev_walkA[par], sc));
+            hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
+                               dim3(blocks_for(CFG::NBUCKETS - splitA, 256)),
+                               dim3(256), 0, sc, pts, VO, OFF, SCH,
+                               p->d_buckets, splitA);
+        }
+    }
     hipLaunchKernelGGL((k_segment_reduce<C, CFG>),
                        dim3(blocks_for(CFG::NWIN * CFG::NSEG, 256)), dim3(256),
                        0, sc, p->d_buckets, p->d_seg_sum, p->d_seg_wsum);

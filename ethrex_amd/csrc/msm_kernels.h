@@ -280,8 +280,9 @@ template <typename C, typename CFG, bool GATHER = true>
 __global__ void __launch_bounds__(256)
 k_bucket_acc(const g1aT<C> *__restrict__ pts, const uint32_t *__restrict__ vals,
              const uint32_t *__restrict__ offsets,
-             const uint32_t *__restrict__ sched, g1jT<C> *__restrict__ buckets) {
-    uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+             const uint32_t *__restrict__ sched, g1jT<C> *__restrict__ buckets,
+             uint32_t tid_base = 0) {
+    uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x + tid_base;
     if (tid >= CFG::NBUCKETS) return;
     uint32_t b = sched[tid];
     if constexpr (!CFG::SIGNED) {
