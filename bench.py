@@ -340,23 +340,15 @@ def main():
         mplan2 = ethrex_amd.MsmPlan(phi - plo)
         mplan2.gen_points(plo)
 
-        nplan2.run(False)  # NTT_0: each step consumes the previous
-        # step's transform output and enqueues the NEXT transform so it
-        # overlaps this step's MSM on the GPU (the NTT's device-sync
-        # blocks only the HOST; the MSM streams keep running)
-
         def wrap_step():
-            mplan2.scalars_from_ntt(nplan2, plo)   # drains MSM k-1
+            nplan2.run(False)
+            mplan2.scalars_from_ntt(nplan2, plo)
             if world == 1:
-                mplan2.run_async()                 # MSM_k enqueued
-            else:
-                mplan2.run_partial_async()
-            nplan2.run(False)                      # NTT_{k+1} overlaps MSM_k
-            if world > 1:
-                part = mplan2.wait_one()
-                allp = allgather_partials(part, dist, device="cpu",
-                                          group=gloo_pg)
-                return ethrex_amd.g1_combine_cpu(allp, world)
+                return mplan2.run()
+            part = mplan2.run_partial()
+            allp = allgather_partials(part, dist, device="cpu",
+                                      group=gloo_pg)
+            return ethrex_amd.g1_combine_cpu(allp, world)
 
         for _ in range(max(args.warmup, 1)):
             wrap_step()
@@ -366,7 +358,6 @@ def main():
         tp = time.perf_counter()
         for _ in range(args.steps):
             wrap_step()
-        mplan2.sync()
         if world > 1:
             dist.barrier()
             torch.cuda.synchronize()

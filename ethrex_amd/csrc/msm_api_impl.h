@@ -771,9 +771,13 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     // near-Poisson-uniform, so a count split ~= a work split); the next
     // step's sorts start after segment A.  EM_MSM_SPLIT = A's percent.
     {
+        // A/B-measured OFF by default: gating the next step's sorts on
+        // 70-90% of the walk cost +0.4..+1.4 ms at 2^24 — the async gap
+        // over the compute chain is HOST pacing (launch+delivery), not
+        // sort contention (profiles/r02_summary.md ledger)
         static int spct = std::getenv("EM_MSM_SPLIT")
                               ? atoi(std::getenv("EM_MSM_SPLIT"))
-                              : 80;
+                              : 0;
         uint32_t splitA =
             (uint32_t)(((uint64_t)CFG::NBUCKETS *
                         (uint32_t)(spct > 0 ? spct : 0)) / 100);
