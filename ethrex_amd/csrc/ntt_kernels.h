@@ -21,10 +21,13 @@
 // coalesced stream instead of the old stride-c gather over an n-sized
 // table (which fetched a full cache line per element).
 //
-// Row kernels run radix-2^3 rounds: three DIT stages per LDS round trip
-// (8 elements per thread in registers) — same multiply count as radix-2
-// (0.5 mul/element/stage) but 1/3 the LDS traffic and barriers of the
-// round-1 radix-2^2 shape.
+// Two row-kernel shapes, A/B-measured (profiles/r02_summary.md):
+// radix-2^2 at 1024 threads (4 waves/SIMD; the DEFAULT — occupancy hides
+// the LDS+mul latency best) and radix-2^3 at 512 threads (3 DIT stages
+// per LDS round trip, 1/3 the LDS traffic; EM_NTT_R8 selects it).  Both
+// use a SKEWED LDS layout — the bit-reversed load scatter otherwise hits
+// one bank group wave-wide — and dynamic LDS so 2048-element rows run 2
+// blocks/CU.
 //
 // Two paths (selected in api_ntt.hip):
 //   13 <= logn <= 24: four-step fused (3 tiled transposes + 2 LDS row-NTT
