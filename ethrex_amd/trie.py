@@ -158,3 +158,40 @@ def account_leaf(nonce: int, balance: int, storage_root: bytes,
             return b""
         return v.to_bytes((v.bit_length() + 7) // 8, "big")
     return rlp.encode([_int(nonce), _int(balance), storage_root, code_hash])
+
+
+def trie_root_hashed_keys(pairs, hash_batch) -> bytes:
+    """MPT root over FIXED 32-byte keys through the NATIVE structure
+    builder (C host; ~100x the Python radix build) with the same
+    level-synchronized hashing: every level is one `hash_batch` call
+    (the GPU KeccakPlan on the product path).  Byte-identical to
+    trie_root() — pinned by tests on random sets and the hoodi witness
+    storage tries."""
+    from .lib import MptBuilder
+    if not pairs:
+        return EMPTY_TRIE_HASH
+    keys = sorted(pairs)
+    assert all(len(k) == 32 for k in keys)
+    vals = b"".join(pairs[k] for k in keys)
+    offs = [0]
+    for k in keys:
+        offs.append(offs[-1] + len(pairs[k]))
+    b = MptBuilder(b"".join(keys), vals, offs)
+    try:
+        for d in range(b.max_depth(), -1, -1):
+            msgs, moffs = b.level_encode(d)
+            if len(moffs) > 1:
+                hashes = hash_batch_encoded(msgs, moffs, hash_batch)
+                b.level_set_hashes(d, hashes)
+            else:
+                b.level_set_hashes(d, b"")
+        return b.root()
+    finally:
+        b.destroy()
+
+
+def hash_batch_encoded(msgs: bytes, offs, hash_batch) -> bytes:
+    """adapt a list-of-bytes hash_batch to the packed (buf, offsets) level
+    layout the native builder emits"""
+    parts = [msgs[offs[i]:offs[i + 1]] for i in range(len(offs) - 1)]
+    return b"".join(hash_batch(parts))

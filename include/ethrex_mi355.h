@@ -115,6 +115,24 @@ int ethrex_mi355_msm_last_times(em_msm_plan *plan, double times_ms[5]);
 int ethrex_mi355_msm_combine(em_msm_plan *plan, const uint8_t *jacobians96,
                              size_t count, uint8_t out[64]);
 
+/* ---- native MPT structure builder (host; §8f row 4 witness-generation
+ * speedup).  Fixed 32-byte sorted distinct keys (the hashed-key state /
+ * storage trie shape); per-level >= 32-B node encodings come out in the
+ * batched-keccak layout, are hashed on the GPU by the caller, and the
+ * hashes patch back before the next (shallower) level encodes.  Levels
+ * run from mpt_max_depth down to 0; the root is always hashed. ---- */
+typedef struct em_mpt em_mpt;
+int ethrex_mi355_mpt_create(const uint8_t *keys32, const uint8_t *vals,
+                            const uint64_t *val_offs, size_t n, em_mpt **out);
+int ethrex_mi355_mpt_destroy(em_mpt *t);
+int ethrex_mi355_mpt_max_depth(em_mpt *t, int *depth);
+int ethrex_mi355_mpt_level_encode(em_mpt *t, int depth, uint8_t *buf,
+                                  uint64_t *offs, size_t buf_cap,
+                                  size_t max_n, size_t *n_hash);
+int ethrex_mi355_mpt_level_set_hashes(em_mpt *t, int depth,
+                                      const uint8_t *h32, size_t n_hash);
+int ethrex_mi355_mpt_root(em_mpt *t, uint8_t out[32]);
+
 /* ---- batched Keccak-256 (witness/statement hashing; original Keccak
  * padding, Ethereum keccak256).  offsets[n+1] delimits message i as
  * [offsets[i], offsets[i+1]); out32 receives 32 bytes per message. ---- */
