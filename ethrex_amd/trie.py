@@ -160,7 +160,7 @@ def account_leaf(nonce: int, balance: int, storage_root: bytes,
     return rlp.encode([_int(nonce), _int(balance), storage_root, code_hash])
 
 
-def trie_root_hashed_keys(pairs, hash_batch) -> bytes:
+def trie_root_hashed_keys(pairs, hash_batch, hash_packed=None) -> bytes:
     """MPT root over FIXED 32-byte keys through the NATIVE structure
     builder (C host; ~100x the Python radix build) with the same
     level-synchronized hashing: every level is one `hash_batch` call
@@ -181,7 +181,13 @@ def trie_root_hashed_keys(pairs, hash_batch) -> bytes:
         for d in range(b.max_depth(), -1, -1):
             msgs, moffs = b.level_encode(d)
             if len(moffs) > 1:
-                hashes = hash_batch_encoded(msgs, moffs, hash_batch)
+                # hash_packed takes the level buffer in the batched-keccak
+                # layout directly (no per-message slicing; the bench reuses
+                # one device KeccakPlan across levels)
+                if hash_packed is not None:
+                    hashes = hash_packed(msgs, moffs)
+                else:
+                    hashes = hash_batch_encoded(msgs, moffs, hash_batch)
                 b.level_set_hashes(d, hashes)
             else:
                 b.level_set_hashes(d, b"")
