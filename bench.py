@@ -142,8 +142,14 @@ def main():
         dist = tdist
         # the 96-B partial exchange runs over a gloo subgroup: the payload
         # is host-side (msm_wait_one delivery) and a CPU gather adds no
-        # H2D/D2H hops and never touches the GPU streams mid-pipeline
-        gloo_pg = tdist.new_group(backend="gloo")
+        # H2D/D2H hops and never touches the GPU streams mid-pipeline.
+        # If gloo cannot form on this box, fall back to the RCCL group
+        # with a cuda-tensor gather (the r01-proven shape) — the one-shot
+        # driver SCALE run must not die on a gloo environment quirk.
+        try:
+            gloo_pg = tdist.new_group(backend="gloo")
+        except Exception:
+            gloo_pg = None
 
     ethrex_amd.set_device(local_rank)
 
@@ -160,13 +166,15 @@ def main():
     scalars = ethrex_amd.gen_fr(42 + rank, shard)
     plan.upload_scalars(scalars)
 
+    xdev = "cpu" if (world == 1 or gloo_pg is not None) else "cuda"
+
     def step():
         if world == 1:
             return plan.run()
         # sync-path step (warmup/reference): shard partial -> AllGather of
-        # the 96-B Jacobian payloads over RCCL/xGMI -> host combine
+        # the 96-B Jacobian payloads -> host combine
         part = plan.run_partial()
-        allparts = allgather_partials(part, dist, device="cpu",
+        allparts = allgather_partials(part, dist, device=xdev,
                                       group=gloo_pg)
         return ethrex_amd.g1_combine_cpu(allparts, world)
     # warmup
@@ -200,7 +208,8 @@ def main():
         # control flow pinned by the world-2 gloo test): the exchange of
         # step k-1 fully overlaps the GPU compute of step k
         pipelined_last = pipelined_shard_steps(plan, args.steps, dist,
-                                               world, group=gloo_pg)
+                                               world, group=gloo_pg,
+                                               device=xdev)
     if world > 1:
         dist.barrier()
         torch.cuda.synchronize()
@@ -346,7 +355,7 @@ def main():
             if world == 1:
                 return mplan2.run()
             part = mplan2.run_partial()
-            allp = allgather_partials(part, dist, device="cpu",
+            allp = allgather_partials(part, dist, device=xdev,
                                       group=gloo_pg)
             return ethrex_amd.g1_combine_cpu(allp, world)
 

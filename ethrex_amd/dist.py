@@ -18,7 +18,7 @@ def shard_range(n_total: int, world: int, rank: int):
 
 
 def pipelined_shard_steps(plan, steps, tdist, world, group=None,
-                          combine=None, collect=None):
+                          combine=None, collect=None, device="cpu"):
     """The N>1 timed loop (bench.py): enqueue step k, deliver step k-1's
     Jacobian partial (plan.wait_one leaves step k running on the GPU),
     AllGather it and combine on the host — the exchange of step k-1
@@ -36,12 +36,12 @@ def pipelined_shard_steps(plan, steps, tdist, world, group=None,
     for _ in range(1, steps):
         plan.run_partial_async()
         part = plan.wait_one()
-        allp = allgather_partials(part, tdist, device="cpu", group=group)
+        allp = allgather_partials(part, tdist, device=device, group=group)
         last = combine(allp, world)
         if collect is not None:
             collect.append(last)
     part = plan.wait_one()
-    allp = allgather_partials(part, tdist, device="cpu", group=group)
+    allp = allgather_partials(part, tdist, device=device, group=group)
     last = combine(allp, world)
     if collect is not None:
         collect.append(last)
