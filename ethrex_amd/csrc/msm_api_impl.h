@@ -88,6 +88,14 @@ struct msm_plan_t {
     uint8_t *d_out2 = nullptr;
     uint8_t *h_out[2] = {nullptr, nullptr};   // pinned window staging
     hipEvent_t ev_sort_done[2], ev_comp_done[2];
+    // per-parity hipGraphs of the sort and compute chains: one step costs
+    // ~70 kernel dispatches (15 per-window rocPRIM sorts alone); captured
+    // once (all pointers are parity-fixed) and replayed as 2 graph
+    // launches — the async step's gap over the compute chain is host
+    // dispatch pacing.  Rebuilt if the point table changes (precompute).
+    hipGraphExec_t gx_sort[2] = {nullptr, nullptr};
+    hipGraphExec_t gx_comp[2] = {nullptr, nullptr};
+    const void *gx_pts[2] = {nullptr, nullptr};
     // contention gating: the bucket walk launches as two segments; the
     // NEXT step's sort chain (HBM+issue-hungry rocPRIM kernels) waits for
     // segment A (~80% of the work), so sorts contend only with the walk's
@@ -158,6 +166,10 @@ static int msm_destroy_t(msm_plan_t<C> *p) {
     (void)hipFree(p->d_out2);
     if (p->h_out[0]) (void)hipHostFree(p->h_out[0]);
     if (p->h_out[1]) (void)hipHostFree(p->h_out[1]);
+    for (int i = 0; i < 2; i++) {
+        if (p->gx_sort[i]) (void)hipGraphExecDestroy(p->gx_sort[i]);
+        if (p->gx_comp[i]) (void)hipGraphExecDestroy(p->gx_comp[i]);
+    }
     if (p->s_sort) (void)hipStreamDestroy(p->s_sort);
     if (p->s_comp) (void)hipStreamDestroy(p->s_comp);
     delete p;
@@ -690,30 +702,23 @@ static int msm_sync_t(msm_plan_t<C> *p) {
     return msm_deliver(p, p->apar);
 }
 
-// one pipelined step: sort chain on s_sort (buffers chosen by step parity),
-// compute chain on s_comp ordered behind it by event.  Returns after
-// ENQUEUE; the result lands in `out` by the time msm_sync (or the depth-2
-// backpressure of a later run_async) returns.  The proving loop runs many
-// MSMs back-to-back, so steady-state cost = max(sort chain, compute chain)
-// instead of their sum.
+// ---- pipelined-step chain enqueue helpers (shared by the direct path
+// and the hipGraph capture below; pure kernel/rocPRIM nodes) ----
+#define EM_HT(call)                         \
+    do {                                    \
+        hipError_t _e = (call);             \
+        if (_e != hipSuccess) return _e;    \
+    } while (0)
+
 template <typename C, typename CFG>
-static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
+static hipError_t msm_enqueue_sort_chain(msm_plan_t<C> *p, int par,
+                                         hipStream_t ss) {
     constexpr bool FB = std::is_same_v<CFG, CfgFB>;
     size_t total = FB ? p->n * (size_t)FB_NWIN : p->n * (size_t)CFG::NWIN;
-    const g1aT<C> *pts = FB ? p->d_pts_ext : p->d_pts;
-    int par = p->apar;
-    int rc = msm_deliver(p, par);  // free this parity's slots (step k-2)
-    if (rc) return rc;
     uint32_t *KO = par ? p->d_keys_out2 : p->d_keys_out;
     uint32_t *VO = par ? p->d_vals_out2 : p->d_vals_out;
     uint32_t *OFF = par ? p->d_offsets2 : p->d_offsets;
     uint32_t *SCH = par ? p->d_sched2 : p->d_sched;
-    uint8_t *DOUT = par ? p->d_out2 : p->d_out;
-    hipStream_t ss = p->s_sort, sc = p->s_comp;
-    // ---- sort chain ----
-    HIP_TRY(hipStreamWaitEvent(ss, p->ev_comp_done[par], 0));
-    // contention gate: wait for the PREVIOUS step's walk segment A
-    HIP_TRY(hipStreamWaitEvent(ss, p->ev_walkA[par ^ 1], 0));
     if constexpr (FB) {
         hipLaunchKernelGGL(k_fb_digits, dim3(blocks_for(total, 256)),
                            dim3(256), 0, ss, p->d_scalars, p->d_inf,
@@ -728,23 +733,21 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
                            dim3(256), 0, ss, p->d_scalars, p->d_inf,
                            p->d_keys, p->d_vals, p->n);
     }
-    hipError_t e = hipSuccess;
     if (!FB && p->keys16) {
-        for (int w = 0; w < CFG::NWIN && e == hipSuccess; w++) {
+        for (int w = 0; w < CFG::NWIN; w++) {
             size_t tmp = p->sort_tmp_bytes;
             size_t off = (size_t)w * p->n;
-            e = rocprim::radix_sort_pairs(
+            EM_HT(rocprim::radix_sort_pairs(
                 p->d_sort_tmp, tmp, (const uint16_t *)p->d_keys + off,
                 (uint16_t *)KO + off, p->d_vals + off, VO + off, p->n, 0,
-                CFG::DBITS, ss);
+                CFG::DBITS, ss));
         }
     } else {
         size_t tmp = p->sort_tmp_bytes;
-        e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys, KO,
-                                      p->d_vals, VO, total, 0,
-                                      CFG::SORT_BITS, ss);
+        EM_HT(rocprim::radix_sort_pairs(p->d_sort_tmp, tmp, p->d_keys, KO,
+                                        p->d_vals, VO, total, 0,
+                                        CFG::SORT_BITS, ss));
     }
-    if (e != hipSuccess) return hip_fail(e, "radix_sort_pairs (async)");
     if (!FB && p->keys16) {
         hipLaunchKernelGGL((k_offsets_seg<CFG>),
                            dim3(blocks_for((size_t)CFG::NBUCKETS + 1, 256)),
@@ -757,50 +760,51 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
     hipLaunchKernelGGL((k_bucket_lengths<CFG>),
                        dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256), 0, ss,
                        OFF, p->d_blen, p->d_bids);
-    {
-        size_t tmp2 = p->sort_tmp_bytes;
-        e = rocprim::radix_sort_pairs(p->d_sort_tmp, tmp2, p->d_blen,
-                                      p->d_blen_out, p->d_bids, SCH,
-                                      (size_t)CFG::NBUCKETS, 0, 32, ss);
-        if (e != hipSuccess) return hip_fail(e, "bucket length sort (async)");
-    }
-    HIP_TRY(hipEventRecord(p->ev_sort_done[par], ss));
-    // ---- compute chain ----
-    HIP_TRY(hipStreamWaitEvent(sc, p->ev_sort_done[par], 0));
-    // two walk segments over the length-sorted schedule (lengths are
-    // near-Poisson-uniform, so a count split ~= a work split); the next
-    // step's sorts start after segment A.  EM_MSM_SPLIT = A's percent.
-    {
-        // A/B-measured OFF by default: gating the next step's sorts on
-        // 70-90% of the walk cost +0.4..+1.4 ms at 2^24 — the async gap
-        // over the compute chain is HOST pacing (launch+delivery), not
-        // sort contention (profiles/r02_summary.md ledger)
-        static int spct = std::getenv("EM_MSM_SPLIT")
-                              ? atoi(std::getenv("EM_MSM_SPLIT"))
-                              : 0;
+    size_t tmp2 = p->sort_tmp_bytes;
+    EM_HT(rocprim::radix_sort_pairs(p->d_sort_tmp, tmp2, p->d_blen,
+                                    p->d_blen_out, p->d_bids, SCH,
+                                    (size_t)CFG::NBUCKETS, 0, 32, ss));
+    return hipSuccess;
+}
+
+// compute chain: bucket walk -> reduction -> window emission.  in_graph
+// runs the plain single-launch walk (event records stay OUTSIDE graphs);
+// the direct path keeps the EM_MSM_SPLIT gate experiment.
+template <typename C, typename CFG>
+static hipError_t msm_enqueue_comp_chain(msm_plan_t<C> *p, int par,
+                                         const g1aT<C> *pts, hipStream_t sc,
+                                         bool in_graph) {
+    uint32_t *VO = par ? p->d_vals_out2 : p->d_vals_out;
+    uint32_t *OFF = par ? p->d_offsets2 : p->d_offsets;
+    uint32_t *SCH = par ? p->d_sched2 : p->d_sched;
+    uint8_t *DOUT = par ? p->d_out2 : p->d_out;
+    static int spct = std::getenv("EM_MSM_SPLIT")
+                          ? atoi(std::getenv("EM_MSM_SPLIT"))
+                          : 0;
+    if (in_graph || spct <= 0) {
+        // gate off (A/B-measured default: gating the sorts on part of the
+        // walk only LOST time — the async gap is host pacing, which the
+        // graph path removes instead)
+        if (!in_graph)
+            EM_HT(hipEventRecord(p->ev_walkA[par], sc));
+        hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
+                           dim3(blocks_for(CFG::NBUCKETS, 256)), dim3(256),
+                           0, sc, pts, VO, OFF, SCH, p->d_buckets, 0u);
+    } else {
         uint32_t splitA =
-            (uint32_t)(((uint64_t)CFG::NBUCKETS *
-                        (uint32_t)(spct > 0 ? spct : 0)) / 100);
-        splitA = (splitA / 256) * 256;  // block-aligned
-        if (spct <= 0) {
-            // gate disabled: signal segment A immediately (sorts overlap
-            // the whole walk — the pre-gating behavior)
-            HIP_TRY(hipEventRecord(p->ev_walkA[par], sc));
+            (uint32_t)(((uint64_t)CFG::NBUCKETS * (uint32_t)spct) / 100);
+        splitA = (splitA / 256) * 256;
+        if (splitA == 0 || splitA >= CFG::NBUCKETS) {
             hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
                                dim3(blocks_for(CFG::NBUCKETS, 256)),
                                dim3(256), 0, sc, pts, VO, OFF, SCH,
                                p->d_buckets, 0u);
-        } else if (splitA == 0 || splitA >= CFG::NBUCKETS) {
-            hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
-                               dim3(blocks_for(CFG::NBUCKETS, 256)),
-                               dim3(256), 0, sc, pts, VO, OFF, SCH,
-                               p->d_buckets, 0u);
-            HIP_TRY(hipEventRecord(p->ev_walkA[par], sc));
+            EM_HT(hipEventRecord(p->ev_walkA[par], sc));
         } else {
             hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
                                dim3(blocks_for(splitA, 256)), dim3(256), 0,
                                sc, pts, VO, OFF, SCH, p->d_buckets, 0u);
-            HIP_TRY(hipEventRecord(p->ev_walkA[par], sc));
+            EM_HT(hipEventRecord(p->ev_walkA[par], sc));
             hipLaunchKernelGGL((k_bucket_acc<C, CFG>),
                                dim3(blocks_for(CFG::NBUCKETS - splitA, 256)),
                                dim3(256), 0, sc, pts, VO, OFF, SCH,
@@ -816,12 +820,103 @@ static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
                        p->d_seg_wsum, p->d_partials);
     hipLaunchKernelGGL((k_window_sum<C, CFG>), dim3(CFG::NWIN), dim3(64), 0,
                        sc, p->d_partials, p->d_windows);
-    // emit the UNSCALED Jacobian window records; the Horner window combine
-    // + affine conversion run on the host at delivery time (overlapped with
-    // the next steps' GPU work)
-    constexpr int NW = FB ? 1 : CFG::NWIN;
     hipLaunchKernelGGL((k_emit_windows<C, CFG>), dim3(1), dim3(64), 0, sc,
                        p->d_windows, DOUT);
+    return hipSuccess;
+}
+
+// capture one chain as a hipGraph on `st` and instantiate it
+template <typename F>
+static int msm_capture_graph(hipStream_t st, hipGraphExec_t *exec, F enqueue,
+                             const char *what) {
+    hipGraph_t g = nullptr;
+    HIP_TRY(hipStreamBeginCapture(st, hipStreamCaptureModeRelaxed));
+    hipError_t ec = enqueue();
+    hipError_t e2 = hipStreamEndCapture(st, &g);
+    if (ec != hipSuccess || e2 != hipSuccess) {
+        if (g) (void)hipGraphDestroy(g);
+        return hip_fail(ec != hipSuccess ? ec : e2, what);
+    }
+    hipError_t e3 = hipGraphInstantiate(exec, g, nullptr, nullptr, 0);
+    (void)hipGraphDestroy(g);
+    if (e3 != hipSuccess) return hip_fail(e3, what);
+    return EM_OK;
+}
+
+// one pipelined step: sort chain on s_sort (buffers chosen by step parity),
+// compute chain on s_comp ordered behind it by event.  Returns after
+// ENQUEUE; the result lands in `out` by the time msm_sync (or the depth-2
+// backpressure of a later run_async) returns.  The proving loop runs many
+// MSMs back-to-back, so steady-state cost = max(sort chain, compute chain)
+// instead of their sum.
+//
+// A step is ~70 dispatches (15 per-window rocPRIM sorts alone), so by
+// default both chains are CAPTURED per parity as hipGraphs and replayed
+// as two graph launches — the measured async-over-compute-chain gap was
+// host dispatch pacing.  All pointers are parity-fixed; graphs rebuild if
+// the point table changes (fixed-base precompute).  EM_MSM_GRAPH=0
+// disables; the EM_MSM_SPLIT gate experiment implies the direct path.
+template <typename C, typename CFG>
+static int msm_run_async_cfg(msm_plan_t<C> *p, uint8_t *out, int out_mode) {
+    constexpr bool FB = std::is_same_v<CFG, CfgFB>;
+    const g1aT<C> *pts = FB ? p->d_pts_ext : p->d_pts;
+    int par = p->apar;
+    int rc = msm_deliver(p, par);  // free this parity's slots (step k-2)
+    if (rc) return rc;
+    hipStream_t ss = p->s_sort, sc = p->s_comp;
+    static bool graph_on = std::getenv("EM_MSM_GRAPH")
+                               ? atoi(std::getenv("EM_MSM_GRAPH")) != 0
+                               : true;
+    static bool split_on = std::getenv("EM_MSM_SPLIT")
+                               ? atoi(std::getenv("EM_MSM_SPLIT")) > 0
+                               : false;
+    bool use_graph = graph_on && !split_on && !p->use_tree;
+    HIP_TRY(hipStreamWaitEvent(ss, p->ev_comp_done[par], 0));
+    // contention gate: wait for the PREVIOUS step's walk segment A
+    HIP_TRY(hipStreamWaitEvent(ss, p->ev_walkA[par ^ 1], 0));
+    if (use_graph &&
+        (p->gx_sort[par] == nullptr || p->gx_pts[par] != (const void *)pts)) {
+        if (p->gx_sort[par]) {
+            (void)hipGraphExecDestroy(p->gx_sort[par]);
+            p->gx_sort[par] = nullptr;
+        }
+        if (p->gx_comp[par]) {
+            (void)hipGraphExecDestroy(p->gx_comp[par]);
+            p->gx_comp[par] = nullptr;
+        }
+        rc = msm_capture_graph(
+            ss, &p->gx_sort[par],
+            [&]() { return msm_enqueue_sort_chain<C, CFG>(p, par, ss); },
+            "sort-chain graph capture");
+        if (rc) return rc;
+        rc = msm_capture_graph(
+            sc, &p->gx_comp[par],
+            [&]() {
+                return msm_enqueue_comp_chain<C, CFG>(p, par, pts, sc, true);
+            },
+            "compute-chain graph capture");
+        if (rc) return rc;
+        p->gx_pts[par] = (const void *)pts;
+    }
+    // ---- sort chain ----
+    if (use_graph) {
+        HIP_TRY(hipGraphLaunch(p->gx_sort[par], ss));
+    } else {
+        hipError_t ec = msm_enqueue_sort_chain<C, CFG>(p, par, ss);
+        if (ec != hipSuccess) return hip_fail(ec, "sort chain (async)");
+    }
+    HIP_TRY(hipEventRecord(p->ev_sort_done[par], ss));
+    // ---- compute chain ----
+    HIP_TRY(hipStreamWaitEvent(sc, p->ev_sort_done[par], 0));
+    if (use_graph) {
+        HIP_TRY(hipEventRecord(p->ev_walkA[par], sc));  // gate-off marker
+        HIP_TRY(hipGraphLaunch(p->gx_comp[par], sc));
+    } else {
+        hipError_t ec = msm_enqueue_comp_chain<C, CFG>(p, par, pts, sc, false);
+        if (ec != hipSuccess) return hip_fail(ec, "compute chain (async)");
+    }
+    uint8_t *DOUT = par ? p->d_out2 : p->d_out;
+    constexpr int NW = FB ? 1 : CFG::NWIN;
     HIP_TRY(hipMemcpyAsync(p->h_out[par], DOUT,
                            (size_t)NW * msm_plan_t<C>::JB,
                            hipMemcpyDeviceToHost, sc));
