@@ -321,17 +321,27 @@ static int msm_create_t(size_t n, msm_plan_t<C> **plan) {
     mal((void **)&p->d_offsets2, ((size_t)nbuckets + 1) * 4);
     mal((void **)&p->d_sched2, (size_t)nbuckets * 4);
     mal((void **)&p->d_out2, (size_t)NWIN_MAX * msm_plan_t<C>::JB);
-    // the compute chain gets the higher stream priority: the VALU-bound
-    // bucket walk should not lose workgroup-dispatch slots to the next
-    // step's (HBM-bound, latency-tolerant) sort chain
-    int prio_lo = 0, prio_hi = 0;
-    (void)hipDeviceGetStreamPriorityRange(&prio_lo, &prio_hi);
-    if (e == hipSuccess)
-        e = hipStreamCreateWithPriority(&p->s_sort, hipStreamNonBlocking,
-                                        prio_lo);
-    if (e == hipSuccess)
-        e = hipStreamCreateWithPriority(&p->s_comp, hipStreamNonBlocking,
-                                        prio_hi);
+    // EQUAL-priority streams (default).  The pass-B experiment that gave
+    // s_comp higher priority STARVED the low-priority sort stream: the
+    // kernel timeline showed the sorts serializing into a 4.3 ms gap
+    // AFTER each bucket walk instead of overlapping it (ROCm priority
+    // queues only dispatch low-priority work when the high-priority
+    // queue is idle).  EM_MSM_PRIO=1 re-enables the priority variant.
+    if (std::getenv("EM_MSM_PRIO")) {
+        int prio_lo = 0, prio_hi = 0;
+        (void)hipDeviceGetStreamPriorityRange(&prio_lo, &prio_hi);
+        if (e == hipSuccess)
+            e = hipStreamCreateWithPriority(&p->s_sort,
+                                            hipStreamNonBlocking, prio_lo);
+        if (e == hipSuccess)
+            e = hipStreamCreateWithPriority(&p->s_comp,
+                                            hipStreamNonBlocking, prio_hi);
+    } else {
+        if (e == hipSuccess)
+            e = hipStreamCreateWithFlags(&p->s_sort, hipStreamNonBlocking);
+        if (e == hipSuccess)
+            e = hipStreamCreateWithFlags(&p->s_comp, hipStreamNonBlocking);
+    }
     for (int i = 0; i < 2 && e == hipSuccess; i++) {
         if (e == hipSuccess)
             e = hipHostMalloc((void **)&p->h_out[i],
